@@ -1,0 +1,457 @@
+"""DeepConsensus-AMD model networks (PyTorch-ROCm).
+
+Re-implements the reference architectures from scratch for MI355X:
+
+* ``EncoderOnlyLearnedValuesTransformer`` — the production model
+  (reference deepconsensus/models/networks.py:368-520): per-row learned
+  embeddings of bases/PW/IP/strand/CCS/(ccs_bq)/SN scaled by sqrt(width) with
+  id-0 zero-masking (networks.py:42-63), a no-bias condenser GEMM to
+  transformer_input_size (networks.py:426-434), sinusoidal position encoding,
+  N x (banded MHA + FFN) with ReZero residuals (encoder_stack.py:55-93,
+  attention_layer.py:112-213, ffn_layer.py:50-87), final fp32 LayerNorm
+  (eps=1e-6) and a Dense(5)+softmax head (networks.py:207-214,342-345).
+* ``FullyConnectedNet`` — the FC baseline (networks.py:67-92).
+* ``ConvNet`` — a compact conv baseline standing in for the reference's
+  keras ResNet50V2 backbone (networks.py:121-170); same I/O contract.
+
+The torch modules here are the correctness reference; the HIP/CDNA4 kernels in
+deepconsensus_amd/ops replace the hot path on gfx950 and are validated against
+this implementation.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from deepconsensus_amd.models.config import Params, get_indices
+from deepconsensus_amd.utils import constants
+
+
+def sinusoidal_position_encoding(
+    length: int,
+    hidden_size: int,
+    min_timescale: float = 1.0,
+    max_timescale: float = 1.0e4,
+    device=None,
+    dtype=torch.float32,
+) -> torch.Tensor:
+    """[length, hidden] sinusoidal encoding (tf-models RelativePositionEmbedding)."""
+    position = torch.arange(length, device=device, dtype=torch.float32)
+    num_timescales = hidden_size // 2
+    log_timescale_increment = math.log(max_timescale / min_timescale) / max(
+        num_timescales - 1, 1
+    )
+    inv_timescales = min_timescale * torch.exp(
+        torch.arange(num_timescales, device=device, dtype=torch.float32)
+        * -log_timescale_increment
+    )
+    scaled_time = position[:, None] * inv_timescales[None, :]
+    signal = torch.cat([torch.sin(scaled_time), torch.cos(scaled_time)], dim=1)
+    return signal.to(dtype)
+
+
+class ScaledEmbedding(nn.Module):
+    """Embedding scaled by sqrt(width) with id-0 rows zero-masked.
+
+    Parity: ModifiedOnDeviceEmbedding (networks.py:42-63); init
+    normal(0, width^-0.5) like the reference's EmbeddingSharedWeights.
+    """
+
+    def __init__(self, vocab_size: int, width: int):
+        super().__init__()
+        self.width = width
+        self.table = nn.Parameter(torch.empty(vocab_size, width))
+        nn.init.normal_(self.table, mean=0.0, std=width**-0.5)
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        emb = F.embedding(ids, self.table) * math.sqrt(self.width)
+        return emb * (ids != 0).unsqueeze(-1).to(emb.dtype)
+
+
+class BandedSelfAttention(nn.Module):
+    """Multi-head self-attention with a +/-attn_win_size banded mask.
+
+    Parity: attention_layer.Attention/SelfAttention (attention_layer.py:34-237)
+    — EinsumDense QKV projections without bias, query scaled by head_dim^-0.5,
+    out-of-band logits set to -1e9, softmax, output projection without bias.
+    """
+
+    def __init__(
+        self,
+        hidden_size: int,
+        num_heads: int,
+        dropout: float,
+        attn_win_size: Optional[int],
+        max_length: int,
+    ):
+        super().__init__()
+        if hidden_size % num_heads:
+            raise ValueError(
+                f"Hidden size ({hidden_size}) must be divisible by the number "
+                f"of heads ({num_heads})."
+            )
+        self.hidden_size = hidden_size
+        self.num_heads = num_heads
+        self.head_dim = hidden_size // num_heads
+        self.dropout = dropout
+        self.attn_win_size = attn_win_size
+        self.q_proj = nn.Linear(hidden_size, hidden_size, bias=False)
+        self.k_proj = nn.Linear(hidden_size, hidden_size, bias=False)
+        self.v_proj = nn.Linear(hidden_size, hidden_size, bias=False)
+        self.out_proj = nn.Linear(hidden_size, hidden_size, bias=False)
+        # Glorot-uniform init, matching the reference EinsumDense kernels.
+        for m in (self.q_proj, self.k_proj, self.v_proj, self.out_proj):
+            nn.init.xavier_uniform_(m.weight)
+        band = self._band_mask(max_length)
+        self.register_buffer("band_mask", band, persistent=False)
+
+    def _band_mask(self, length: int) -> torch.Tensor:
+        if self.attn_win_size is None:
+            return torch.ones(length, length, dtype=torch.bool)
+        i = torch.arange(length)
+        return (i[:, None] - i[None, :]).abs() <= self.attn_win_size
+
+    def forward(
+        self, x: torch.Tensor, training: bool, need_weights: bool = False
+    ):
+        b, t, _ = x.shape
+        h, d = self.num_heads, self.head_dim
+        q = self.q_proj(x).view(b, t, h, d).transpose(1, 2)  # [B,H,T,D]
+        k = self.k_proj(x).view(b, t, h, d).transpose(1, 2)
+        v = self.v_proj(x).view(b, t, h, d).transpose(1, 2)
+        q = q * (d**-0.5)
+        logits = torch.matmul(q, k.transpose(-1, -2))  # [B,H,T,T]
+        mask = self.band_mask[:t, :t]
+        logits = torch.where(
+            mask, logits, torch.tensor(-1e9, dtype=logits.dtype,
+                                       device=logits.device)
+        )
+        # Softmax in fp32 for stability (reference attention_layer.py:208-211).
+        weights = torch.softmax(logits.float(), dim=-1).to(x.dtype)
+        if training and self.dropout > 0:
+            weights = F.dropout(weights, p=self.dropout, training=True)
+        ctx = torch.matmul(weights, v)  # [B,H,T,D]
+        ctx = ctx.transpose(1, 2).reshape(b, t, self.hidden_size)
+        out = self.out_proj(ctx)
+        if need_weights:
+            return out, weights
+        return out, None
+
+
+class FeedForward(nn.Module):
+    """Dense(filter, ReLU) -> dropout -> Dense(hidden) (ffn_layer.py:50-87)."""
+
+    def __init__(self, hidden_size: int, filter_size: int, dropout: float):
+        super().__init__()
+        self.filter_layer = nn.Linear(hidden_size, filter_size, bias=True)
+        self.output_layer = nn.Linear(filter_size, hidden_size, bias=True)
+        nn.init.xavier_uniform_(self.filter_layer.weight)
+        nn.init.zeros_(self.filter_layer.bias)
+        nn.init.xavier_uniform_(self.output_layer.weight)
+        nn.init.zeros_(self.output_layer.bias)
+        self.dropout = dropout
+
+    def forward(self, x: torch.Tensor, training: bool) -> torch.Tensor:
+        y = F.relu(self.filter_layer(x))
+        if training and self.dropout > 0:
+            y = F.dropout(y, p=self.dropout, training=True)
+        return self.output_layer(y)
+
+
+class SublayerWrapper(nn.Module):
+    """ReZero (x + alpha*y, alpha init 0) or pre-LayerNorm residual wrapper.
+
+    Parity: encoder_stack.PrePostProcessingWrapper (encoder_stack.py:43-93).
+    """
+
+    def __init__(self, params: Params):
+        super().__init__()
+        self.rezero = bool(params["rezero"])
+        self.post_dropout = params["layer_postprocess_dropout"]
+        if self.rezero:
+            self.alpha = nn.Parameter(torch.zeros(()))
+        else:
+            self.layer_norm = nn.LayerNorm(params["hidden_size"], eps=1e-6)
+
+    def pre(self, x: torch.Tensor) -> torch.Tensor:
+        if self.rezero:
+            return x
+        # LN computed in fp32 (reference pins LN dtype to float32).
+        return self.layer_norm(x.float()).to(x.dtype)
+
+    def post(
+        self, x: torch.Tensor, y: torch.Tensor, training: bool
+    ) -> torch.Tensor:
+        if training and self.post_dropout > 0:
+            y = F.dropout(y, p=self.post_dropout, training=True)
+        if self.rezero:
+            return x + self.alpha * y
+        return x + y
+
+
+class EncoderLayer(nn.Module):
+    def __init__(self, params: Params, max_length: int):
+        super().__init__()
+        self.attn = BandedSelfAttention(
+            params["hidden_size"],
+            params["num_heads"],
+            params["attention_dropout"],
+            params.get("attn_win_size"),
+            max_length,
+        )
+        self.ffn = FeedForward(
+            params["hidden_size"], params["filter_size"], params["relu_dropout"]
+        )
+        self.attn_wrap = SublayerWrapper(params)
+        self.ffn_wrap = SublayerWrapper(params)
+
+    def forward(self, x, training: bool, need_weights: bool = False):
+        y, w = self.attn(self.attn_wrap.pre(x), training, need_weights)
+        x = self.attn_wrap.post(x, y, training)
+        y = self.ffn(self.ffn_wrap.pre(x), training)
+        x = self.ffn_wrap.post(x, y, training)
+        return x, w
+
+
+class EncoderOnlyTransformer(nn.Module):
+    """Encoder stack + Dense(5)+softmax head over per-position features.
+
+    Input: rows tensor [B, total_rows, L] (or [B, total_rows, L, 1]); the base
+    class consumes it as [B, L, hidden] features directly
+    (networks.py:241-284 squeeze/transpose).
+    """
+
+    def __init__(self, params: Params):
+        super().__init__()
+        self.params = params
+        self.max_length = params["max_length"]
+        self.hidden_size = params["hidden_size"]
+        self.add_pos_encoding = params.get("add_pos_encoding", False)
+        self.layers = nn.ModuleList(
+            EncoderLayer(params, self.max_length)
+            for _ in range(params["num_hidden_layers"])
+        )
+        self.output_norm = nn.LayerNorm(self.hidden_size, eps=1e-6)
+        self.fc1 = nn.Linear(self.hidden_size, constants.SEQ_VOCAB_SIZE)
+        nn.init.xavier_uniform_(self.fc1.weight)
+        nn.init.zeros_(self.fc1.bias)
+        if self.add_pos_encoding:
+            pe = sinusoidal_position_encoding(
+                self.max_length, self.hidden_size
+            )
+            self.register_buffer("pos_encoding", pe, persistent=False)
+        self.post_dropout = params.get("layer_postprocess_dropout", 0.0)
+
+    def _prepare_inputs(self, rows: torch.Tensor) -> torch.Tensor:
+        """[B, R, L, 1] or [B, R, L] -> [B, L, R] feature matrix."""
+        if rows.dim() == 4:
+            rows = rows.squeeze(-1)
+        return rows.transpose(1, 2)
+
+    def embed(self, inputs: torch.Tensor) -> torch.Tensor:
+        """Identity in the base class; learned-values subclass overrides."""
+        return inputs
+
+    def encode(
+        self, rows: torch.Tensor, training: bool = False,
+        need_weights: bool = False,
+    ) -> Dict[str, torch.Tensor]:
+        inputs = self._prepare_inputs(rows)
+        x = self.embed(inputs)
+        t = x.shape[1]
+        if x.shape[2] % 2 != 0:
+            # Odd hidden: pad one zero feature (networks.py:300-306).
+            x = F.pad(x, (0, 1))
+        if self.add_pos_encoding:
+            x = x + self.pos_encoding[:t].to(x.dtype)
+        if training and self.post_dropout > 0:
+            x = F.dropout(x, p=self.post_dropout, training=True)
+        outputs: Dict[str, torch.Tensor] = {}
+        for n, layer in enumerate(self.layers):
+            x, w = layer(x, training, need_weights)
+            if need_weights:
+                outputs[f"attention_scores_{n}"] = w
+        final = self.output_norm(x.float())
+        outputs["final_output"] = final
+        outputs["logits"] = self.fc1(final)
+        return outputs
+
+    def forward(
+        self, rows: torch.Tensor, training: bool = False
+    ) -> torch.Tensor:
+        out = self.encode(rows, training=training)
+        return torch.softmax(out["logits"].float(), dim=-1)
+
+    def predict(self, rows: torch.Tensor) -> torch.Tensor:
+        return self.forward(rows, training=False)
+
+
+class EncoderOnlyLearnedValuesTransformer(EncoderOnlyTransformer):
+    """The production model: learned per-row embeddings + condenser.
+
+    Parity: networks.py:368-520. Embedding concat order is row-major within
+    each feature block: bases rows, PW rows, IP rows, strand rows, CCS row
+    (shares the bases table, networks.py:484-489), optional ccs_bq row
+    (input shifted +1, networks.py:492-497), SN rows.
+    """
+
+    def __init__(self, params: Params):
+        super().__init__(params)
+        p = params
+        self.max_passes = p["max_passes"]
+        self.use_ccs_bq = bool(p.get("use_ccs_bq", False))
+        self.indices = get_indices(self.max_passes, self.use_ccs_bq)
+        self.bases_embedding = ScaledEmbedding(
+            constants.SEQ_VOCAB_SIZE, p["per_base_hidden_size"]
+        )
+        self.pw_embedding = ScaledEmbedding(p["PW_MAX"] + 1, p["pw_hidden_size"])
+        self.ip_embedding = ScaledEmbedding(p["IP_MAX"] + 1, p["ip_hidden_size"])
+        self.strand_embedding = ScaledEmbedding(
+            p["STRAND_MAX"] + 1, p["strand_hidden_size"]
+        )
+        self.sn_embedding = ScaledEmbedding(p["SN_MAX"] + 1, p["sn_hidden_size"])
+        if self.use_ccs_bq:
+            self.ccs_bq_embedding = ScaledEmbedding(
+                p["CCS_BQ_MAX"], p["ccs_bq_hidden_size"]
+            )
+        self.condense = bool(p.get("condense_transformer_input", False))
+        if self.condense:
+            dim = (
+                p["per_base_hidden_size"] * (self.max_passes + 1)
+                + p["pw_hidden_size"] * self.max_passes
+                + p["ip_hidden_size"] * self.max_passes
+                + p["strand_hidden_size"] * self.max_passes
+                + (p["ccs_bq_hidden_size"] if self.use_ccs_bq else 0)
+                + p["sn_hidden_size"] * 4
+            )
+            self.condenser = nn.Linear(
+                dim, p["transformer_input_size"], bias=False
+            )
+            nn.init.xavier_uniform_(self.condenser.weight)
+
+    @staticmethod
+    def _block(
+        emb: ScaledEmbedding, ids: torch.Tensor
+    ) -> torch.Tensor:
+        """Embeds a [B, rows, L] id block -> [B, L, rows*width]."""
+        b, r, l = ids.shape
+        e = emb(ids.reshape(b, r * l)).reshape(b, r, l, emb.width)
+        return e.permute(0, 2, 1, 3).reshape(b, l, r * emb.width)
+
+    def embed(self, inputs: torch.Tensor) -> torch.Tensor:
+        # inputs: [B, L, R] float features; slice per-feature row ranges.
+        (bi, pwi, ipi, sti, ci, bqi, sni) = self.indices
+        x = inputs.transpose(1, 2)  # [B, R, L]
+        ids = x.long()
+        parts = [
+            self._block(self.bases_embedding, ids[:, bi[0]:bi[1]]),
+            self._block(self.pw_embedding, ids[:, pwi[0]:pwi[1]]),
+            self._block(self.ip_embedding, ids[:, ipi[0]:ipi[1]]),
+            self._block(self.strand_embedding, ids[:, sti[0]:sti[1]]),
+            self._block(self.bases_embedding, ids[:, ci[0]:ci[1]]),
+        ]
+        if self.use_ccs_bq:
+            parts.append(
+                self._block(self.ccs_bq_embedding, ids[:, bqi[0]:bqi[1]] + 1)
+            )
+        parts.append(self._block(self.sn_embedding, ids[:, sni[0]:sni[1]]))
+        emb = torch.cat(parts, dim=-1)
+        if self.condense:
+            emb = self.condenser(emb)
+        return emb
+
+
+class FullyConnectedNet(nn.Module):
+    """Flatten -> Dense-ReLU stack -> Dense(L*5) -> softmax (networks.py:67-92)."""
+
+    def __init__(self, params: Params):
+        super().__init__()
+        self.params = params
+        in_dim = (
+            params["hidden_size"] * params["max_length"] * params["num_channels"]
+        )
+        dims = [in_dim] + list(params["fc_size"])
+        self.hidden = nn.ModuleList(
+            nn.Linear(dims[i], dims[i + 1]) for i in range(len(params["fc_size"]))
+        )
+        self.out = nn.Linear(
+            dims[-1], params["max_length"] * constants.SEQ_VOCAB_SIZE
+        )
+        self.dropout = params.get("fc_dropout", 0.0)
+        self.max_length = params["max_length"]
+
+    def forward(
+        self, rows: torch.Tensor, training: bool = False
+    ) -> torch.Tensor:
+        if rows.dim() == 3:
+            rows = rows.unsqueeze(-1)
+        x = rows.flatten(1)
+        for layer in self.hidden:
+            x = F.relu(layer(x))
+            if training and self.dropout > 0:
+                x = F.dropout(x, p=self.dropout, training=True)
+        x = self.out(x).view(-1, self.max_length, constants.SEQ_VOCAB_SIZE)
+        return torch.softmax(x.float(), dim=-1)
+
+    def predict(self, rows: torch.Tensor) -> torch.Tensor:
+        return self.forward(rows, training=False)
+
+
+class ConvNet(nn.Module):
+    """Compact conv baseline over the [B, 1, R, L] matrix.
+
+    Stands in for the reference ConvNet's keras ResNet50V2 backbone
+    (networks.py:121-170) with an MI355X-friendly residual CNN; same I/O:
+    probs [B, L, 5].
+    """
+
+    def __init__(self, params: Params, channels: int = 64, blocks: int = 4):
+        super().__init__()
+        self.params = params
+        self.max_length = params["max_length"]
+        self.stem = nn.Conv2d(1, channels, 3, padding=1)
+        self.blocks = nn.ModuleList()
+        for _ in range(blocks):
+            self.blocks.append(
+                nn.Sequential(
+                    nn.BatchNorm2d(channels),
+                    nn.ReLU(),
+                    nn.Conv2d(channels, channels, 3, padding=1),
+                    nn.BatchNorm2d(channels),
+                    nn.ReLU(),
+                    nn.Conv2d(channels, channels, 3, padding=1),
+                )
+            )
+        self.head = nn.Linear(channels, constants.SEQ_VOCAB_SIZE)
+
+    def forward(
+        self, rows: torch.Tensor, training: bool = False
+    ) -> torch.Tensor:
+        if rows.dim() == 4:
+            rows = rows.squeeze(-1)
+        x = self.stem(rows.unsqueeze(1))  # [B, C, R, L]
+        for block in self.blocks:
+            x = x + block(x)
+        x = x.mean(dim=2).transpose(1, 2)  # [B, L, C]
+        return torch.softmax(self.head(x).float(), dim=-1)
+
+    def predict(self, rows: torch.Tensor) -> torch.Tensor:
+        return self.forward(rows, training=False)
+
+
+def get_model(params: Params) -> nn.Module:
+    """Model factory (reference model_utils.py:142-152)."""
+    name = params["model_name"]
+    if name == "fc":
+        return FullyConnectedNet(params)
+    if name == "conv_net":
+        return ConvNet(params)
+    if name in ("transformer",):
+        return EncoderOnlyTransformer(params)
+    if "transformer_learn_values" in name:
+        return EncoderOnlyLearnedValuesTransformer(params)
+    raise ValueError(f"Unknown model_name: {name}")
