@@ -1,0 +1,174 @@
+#!/usr/bin/env python3
+"""DeepConsensus-AMD flagship benchmark: windowed consensus inference.
+
+Measures the BASELINE.json headline metric — ZMWs/sec of v1.2-architecture
+model inference on chem2.2-shaped data (15 kb insert => 150 windows of 100 bp
+per ZMW, max_passes=20, 85 feature rows) — on synthetic subread windows with
+random-init weights, bf16 compute on MI355X.
+
+One timed step = one batch of windows through the full serving path:
+pinned-host H2D copy -> fused embed+condense (HIP) -> bf16 encoder ->
+fused LN+head+QV (HIP) -> D2H of uint8 base/qual calls.
+
+Reference CPU baseline: 0.76 ZMW/s on one 16-vCPU shard
+(docs/quick_start.md:315-320).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--batch-size B]
+Under torchrun, reads RANK/LOCAL_RANK/WORLD_SIZE from the environment.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from deepconsensus_amd.models import config as cfg
+from deepconsensus_amd.models.model import get_model
+from deepconsensus_amd.models.runner import InferenceRunner
+
+WINDOWS_PER_ZMW = 150  # chem2.2 15 kb insert / 100 bp windows
+BASELINE_ZMW_PER_SEC = 0.76
+
+
+def make_synthetic_windows(params, batch: int, seed: int) -> np.ndarray:
+    """Random chem2.2-shaped window feature tensors [B, 85, 100] fp32."""
+    rng = np.random.default_rng(seed)
+    R, L, mp = params.total_rows, params.max_length, params.max_passes
+    rows = np.zeros((batch, R, L), dtype=np.float32)
+    # ~20% gaps in subread bases, like spaced-out alignments.
+    bases = rng.integers(0, 5, size=(batch, mp, L))
+    rows[:, 0:mp] = bases
+    rows[:, mp : 2 * mp] = rng.integers(0, 256, size=(batch, mp, L))
+    rows[:, 2 * mp : 3 * mp] = rng.integers(0, 256, size=(batch, mp, L))
+    rows[:, 3 * mp : 4 * mp] = rng.integers(1, 3, size=(batch, mp, L))
+    rows[:, 4 * mp] = rng.integers(0, 5, size=(batch, L))
+    rows[:, -4:] = rng.integers(5, 30, size=(batch, 4, 1))
+    return rows
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch-size", type=int, default=2048)
+    ap.add_argument("--pool-batches", type=int, default=8)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+
+    have_cuda = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if have_cuda else "cpu"
+    if have_cuda:
+        torch.cuda.set_device(local_rank)
+
+    if distributed:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group(
+            backend="nccl" if have_cuda else "gloo",
+            rank=rank,
+            world_size=world,
+        )
+
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=False)
+    batch = args.batch_size if have_cuda else 32
+
+    torch.manual_seed(1234)
+    model = get_model(params)
+    runner = InferenceRunner(
+        params, model, device=device, calibration="0,1.197654,-0.99781"
+    )
+
+    # Pinned host pool of pre-featurized batches (the host preprocess stage is
+    # pipelined/off-path in production; windows arrive pre-packed).
+    pool = []
+    for i in range(args.pool_batches):
+        t = torch.from_numpy(make_synthetic_windows(params, batch, 97 + i))
+        if have_cuda:
+            t = t.pin_memory()
+        pool.append(t)
+
+    def one_step(i: int):
+        rows = pool[i % len(pool)]
+        bases, quals = runner.forward_windows(rows)
+        # D2H of the uint8 calls (the output contract of the serving step).
+        return bases.to("cpu", non_blocking=True), quals.to(
+            "cpu", non_blocking=True
+        )
+
+    # Warmup.
+    for i in range(args.warmup):
+        one_step(i)
+    if have_cuda:
+        torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    if have_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if have_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    windows_per_sec = args.steps * batch / elapsed * world
+    zmw_per_sec = windows_per_sec / WINDOWS_PER_ZMW
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "zmw_per_sec",
+            "value": round(zmw_per_sec, 3),
+            "unit": "ZMWs/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(zmw_per_sec / BASELINE_ZMW_PER_SEC, 2),
+            "dtype": "bf16" if have_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "transformer_learn_values v1.2 (hidden 280, 6 layers, "
+                "heads 2, band +-12, max_passes 20)",
+                "global_batch": batch * world,
+                "seq_len": 100,
+                "windows_per_zmw": WINDOWS_PER_ZMW,
+                "parallelism": f"dp{world}",
+                "native_kernels": bool(runner.native),
+            },
+        }
+        print(json.dumps(result))
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,41 @@
+"""In-tree build of the HIP/CDNA4 extensions (gfx950 only).
+
+Builds with torch.utils.cpp_extension driving hipcc under
+PYTORCH_ROCM_ARCH=gfx950, into deepconsensus_amd/ops/_build so the .so
+travels with the repo snapshot to GPU boxes (no JIT cache dependence).
+"""
+from __future__ import annotations
+
+import os
+import shutil
+
+_OPS_DIR = os.path.dirname(os.path.abspath(__file__))
+_BUILD_DIR = os.path.join(_OPS_DIR, "_build")
+_SRC = os.path.join(_OPS_DIR, "hip", "dc_kernels.hip")
+EXT_NAME = "dc_hip_kernels"
+
+
+def build(verbose: bool = False):
+    """Compiles (if needed) and loads the dc_hip_kernels extension."""
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    os.makedirs(_BUILD_DIR, exist_ok=True)
+    from torch.utils.cpp_extension import load
+
+    module = load(
+        name=EXT_NAME,
+        sources=[_SRC],
+        build_directory=_BUILD_DIR,
+        extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
+        verbose=verbose,
+    )
+    return module
+
+
+def clean():
+    if os.path.isdir(_BUILD_DIR):
+        shutil.rmtree(_BUILD_DIR)
+
+
+if __name__ == "__main__":
+    build(verbose=True)
+    print("built", EXT_NAME, "in", _BUILD_DIR)
