@@ -2,8 +2,9 @@
 
 MI355X-native execution of the reference's run_model_on_examples
 (quick_inference.py:341-415): on GPU the path is
-  fused_embed_condense (HIP, K2+K3)
-  -> bf16 encoder stack (rocBLAS GEMMs + banded attention)
+  embed_gather (HIP, K2) -> condenser GEMM (hipBLASLt, K3)
+  -> per layer: fused-QKV GEMM -> banded_attn (HIP, K5-K7) -> out-proj GEMM
+     -> ReZero add -> FFN GEMMs (bias fused) -> ReZero add, all bf16
   -> fused_ln_head_qv (HIP, K10+K11+K12) emitting uint8 base ids + QVs.
 On CPU it falls back to the fp32 torch reference (same numerics contract).
 
@@ -11,18 +12,18 @@ On a GPU machine the HIP extension is required (no silent eager fallback).
 """
 from __future__ import annotations
 
+import copy
 import math
-from typing import Optional, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch
 
 from deepconsensus_amd import ops as dc_ops
 from deepconsensus_amd.calibration.calibration import (
-    QualityCalibrationValues,
     parse_calibration_string,
 )
-from deepconsensus_amd.models.config import Params, get_indices
+from deepconsensus_amd.models.config import Params
 from deepconsensus_amd.models.model import (
     EncoderOnlyLearnedValuesTransformer,
     get_model,
@@ -30,26 +31,20 @@ from deepconsensus_amd.models.model import (
 from deepconsensus_amd.utils import constants
 
 
-def build_fused_tables(
-    model: EncoderOnlyLearnedValuesTransformer,
-) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Precomputes per-row fused embed+condense tables.
+def build_gather_tables(model: EncoderOnlyLearnedValuesTransformer):
+    """Builds the embed_gather kernel's scaled tables + 16-byte chunk map.
 
-    T'_row[id] = (table_f[id] * sqrt(w) * (id != 0)) @ Wc[row_cols] so the
-    device kernel reduces K2+K3 to R gather-adds of H-wide bf16 rows.
-
-    Returns (fused_table [V_total, H] bf16, row_offset, row_shift, row_vocab).
+    Folds the sqrt(width) scale, the id-0 zero mask, and the ccs_bq +1 shift
+    (networks.py:42-63,492-497) into flat bf16 tables; maps every 8-column
+    output chunk to its source (input row, table slice) entries.
     """
-    p = model.params
-    H = p["transformer_input_size"]
-    W = model.condenser.weight.detach().float()  # [H, concat_dim]
     (bi, pwi, ipi, sti, ci, bqi, sni) = model.indices
 
     def scaled(emb):
         t = emb.table.detach().float() * math.sqrt(emb.width)
         t = t.clone()
         t[0].zero_()
-        return t
+        return t.to(torch.bfloat16)
 
     tables = {
         "bases": scaled(model.bases_embedding),
@@ -61,46 +56,58 @@ def build_fused_tables(
     if model.use_ccs_bq:
         tables["ccs_bq"] = scaled(model.ccs_bq_embedding)
 
-    # Row plan in input-row order with concat-column tracking.
-    plan = []  # (table_name, shift)
-    for _ in range(bi[0], bi[1]):
-        plan.append(("bases", 0))
-    for _ in range(pwi[0], pwi[1]):
-        plan.append(("pw", 0))
-    for _ in range(ipi[0], ipi[1]):
-        plan.append(("ip", 0))
-    for _ in range(sti[0], sti[1]):
-        plan.append(("strand", 0))
-    for _ in range(ci[0], ci[1]):
-        plan.append(("bases", 0))
+    order = ["bases", "pw", "ip", "strand", "sn"]
     if model.use_ccs_bq:
-        plan.append(("ccs_bq", 1))
-    for _ in range(sni[0], sni[1]):
-        plan.append(("sn", 0))
-
-    fused_rows = []
-    row_offset, row_shift, row_vocab = [], [], []
-    col = 0
-    offset = 0
-    for name, shift in plan:
+        order.append("ccs_bq")
+    flat_parts, table_base = [], {}
+    off = 0
+    for name in order:
         t = tables[name]
-        w = t.shape[1]
-        cols = W[:, col : col + w]  # [H, w]
-        fused = t @ cols.T  # [vocab, H]
-        fused_rows.append(fused)
-        row_offset.append(offset)
-        row_shift.append(shift)
-        row_vocab.append(t.shape[0])
-        offset += t.shape[0]
-        col += w
-    assert col == W.shape[1], (col, W.shape)
-    fused_table = torch.cat(fused_rows, 0).to(torch.bfloat16).contiguous()
-    return (
-        fused_table,
-        torch.tensor(row_offset, dtype=torch.int32),
-        torch.tensor(row_shift, dtype=torch.int32),
-        torch.tensor(row_vocab, dtype=torch.int32),
+        table_base[name] = off
+        flat_parts.append(t.reshape(-1))
+        off += t.numel()
+    table_flat = torch.cat(flat_parts).contiguous()
+
+    # Per input row: (table name, shift).
+    plan: List[Tuple[str, int]] = []
+    plan += [("bases", 0)] * (bi[1] - bi[0])
+    plan += [("pw", 0)] * (pwi[1] - pwi[0])
+    plan += [("ip", 0)] * (ipi[1] - ipi[0])
+    plan += [("strand", 0)] * (sti[1] - sti[0])
+    plan += [("bases", 0)] * (ci[1] - ci[0])
+    if model.use_ccs_bq:
+        plan += [("ccs_bq", 1)]
+    plan += [("sn", 0)] * (sni[1] - sni[0])
+
+    row_shift = torch.tensor([s for _, s in plan], dtype=torch.int32)
+    row_vocab = torch.tensor(
+        [tables[n].shape[0] for n, _ in plan], dtype=torch.int32
     )
+
+    # Column layout -> 8-col chunks -> entries (row, elem_base, width).
+    cols: List[Tuple[int, int, int]] = []  # per col: (row, elem_base, width)
+    for r, (name, _) in enumerate(plan):
+        w = tables[name].shape[1]
+        base = table_base[name]
+        for j in range(w):
+            cols.append((r, base, w))
+    concat = len(cols)
+    assert concat % 8 == 0, f"concat width {concat} not 8-aligned"
+    nchunk = concat // 8
+    chunk_cnt = torch.zeros(nchunk, dtype=torch.int32)
+    chunk_entries = torch.zeros(nchunk * 4, 4, dtype=torch.int32)
+    for c in range(nchunk):
+        entries = []
+        j = c * 8
+        while j < (c + 1) * 8:
+            r, base, w = cols[j]
+            entries.append((r, base, w))
+            j += w
+        assert len(entries) <= 4, "chunk spans >4 table rows"
+        chunk_cnt[c] = len(entries)
+        for k, (r, base, w) in enumerate(entries):
+            chunk_entries[c * 4 + k] = torch.tensor([r, base, w, 0])
+    return table_flat, row_shift, row_vocab, chunk_cnt, chunk_entries.reshape(-1)
 
 
 class InferenceRunner:
@@ -136,49 +143,94 @@ class InferenceRunner:
         else:
             self.ext = None
 
+        self.model = model.to(self.device)
         if self.native:
-            ft, ro, rs, rv = build_fused_tables(model)
-            self.fused_table = ft.to(self.device)
-            self.row_offset = ro.to(self.device)
-            self.row_shift = rs.to(self.device)
-            self.row_vocab = rv.to(self.device)
-            # Encoder stack in bf16 on device; LN/head params fp32. Deep-copied
-            # so the fp32 reference model stays intact for validation.
-            import copy
+            self._prepare_native(model)
 
-            self.model = model.to(self.device)
-            self.layers_bf16 = copy.deepcopy(model.layers).to(
-                torch.bfloat16
-            ).to(self.device)
-            self.pos = model.pos_encoding.to(self.device).to(torch.bfloat16) \
-                if model.add_pos_encoding else None
-            self.ln_gamma = model.output_norm.weight.detach().float().to(
-                self.device
-            )
-            self.ln_beta = model.output_norm.bias.detach().float().to(
-                self.device
-            )
-            self.w_head = model.fc1.weight.detach().float().contiguous().to(
-                self.device
-            )
-            self.b_head = model.fc1.bias.detach().float().to(self.device)
+    def _prepare_native(self, model) -> None:
+        dev, bf16 = self.device, torch.bfloat16
+        tf, rs, rv, cc, ce = build_gather_tables(model)
+        self.table_flat = tf.to(dev)
+        self.row_shift = rs.to(dev)
+        self.row_vocab = rv.to(dev)
+        self.chunk_cnt = cc.to(dev)
+        self.chunk_entries = ce.to(dev)
+        self.win = model.params.get("attn_win_size")
+        self.num_heads = model.params["num_heads"]
+        # Condenser (no bias): store W^T contiguous for x @ Wt.
+        self.cond_wt = (
+            model.condenser.weight.detach().t().contiguous().to(bf16).to(dev)
+        )
+        self.pos = (
+            model.pos_encoding.to(bf16).to(dev)
+            if model.add_pos_encoding
+            else None
+        )
+        # Packed per-layer weights (fast rezero path); generic fallback keeps
+        # a bf16 deepcopy of the layer stack for non-rezero configs.
+        self.rezero_fast = all(
+            l.attn_wrap.rezero and l.ffn_wrap.rezero for l in model.layers
+        ) and self.win is not None
+        if self.rezero_fast:
+            self.layer_w = []
+            for l in model.layers:
+                wq = l.attn.q_proj.weight.detach()
+                wk = l.attn.k_proj.weight.detach()
+                wv = l.attn.v_proj.weight.detach()
+                wqkv_t = (
+                    torch.cat([wq, wk, wv], 0).t().contiguous().to(bf16).to(dev)
+                )
+                self.layer_w.append(
+                    dict(
+                        wqkv_t=wqkv_t,
+                        wout_t=l.attn.out_proj.weight.detach()
+                        .t().contiguous().to(bf16).to(dev),
+                        w1_t=l.ffn.filter_layer.weight.detach()
+                        .t().contiguous().to(bf16).to(dev),
+                        b1=l.ffn.filter_layer.bias.detach().to(bf16).to(dev),
+                        w2_t=l.ffn.output_layer.weight.detach()
+                        .t().contiguous().to(bf16).to(dev),
+                        b2=l.ffn.output_layer.bias.detach().to(bf16).to(dev),
+                        alpha_attn=float(l.attn_wrap.alpha.detach()),
+                        alpha_ffn=float(l.ffn_wrap.alpha.detach()),
+                    )
+                )
         else:
-            self.model = model.to(self.device)
+            self.layers_bf16 = (
+                copy.deepcopy(model.layers).to(bf16).to(dev)
+            )
+        self.ln_gamma = model.output_norm.weight.detach().float().to(dev)
+        self.ln_beta = model.output_norm.bias.detach().float().to(dev)
+        self.w_head = model.fc1.weight.detach().float().contiguous().to(dev)
+        self.b_head = model.fc1.bias.detach().float().to(dev)
 
     @torch.no_grad()
     def encode_native(self, rows: torch.Tensor) -> torch.Tensor:
         """Native path up to (but excluding) the final LayerNorm: [B,L,H] bf16."""
-        x = self.ext.fused_embed_condense(
-            rows.contiguous(),
-            self.fused_table,
-            self.row_offset,
-            self.row_shift,
-            self.row_vocab,
-        )  # [B, L, H] bf16
+        emb = self.ext.embed_gather(
+            rows.contiguous(), self.table_flat, self.row_shift,
+            self.row_vocab, self.chunk_cnt, self.chunk_entries,
+        )  # [B, L, concat] bf16
+        b, l, _ = emb.shape
+        x = emb.reshape(b * l, -1) @ self.cond_wt  # [B*L, H]
+        x = x.view(b, l, -1)
         if self.pos is not None:
-            x = x + self.pos[: x.shape[1]]
-        for layer in self.layers_bf16:
-            x, _ = layer(x, training=False)
+            x = x + self.pos[:l]
+        if self.rezero_fast:
+            h = x.shape[-1]
+            for lw in self.layer_w:
+                flat = x.view(b * l, h)
+                qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
+                a = self.ext.banded_attn(qkv, self.num_heads, self.win)
+                y = a.view(b * l, h) @ lw["wout_t"]
+                x = torch.add(flat, y, alpha=lw["alpha_attn"]).view(b, l, h)
+                flat = x.view(b * l, h)
+                ff = torch.addmm(lw["b1"], flat, lw["w1_t"]).relu_()
+                y = torch.addmm(lw["b2"], ff, lw["w2_t"])
+                x = torch.add(flat, y, alpha=lw["alpha_ffn"]).view(b, l, h)
+        else:
+            for layer in self.layers_bf16:
+                x, _ = layer(x, training=False)
         return x
 
     @torch.no_grad()

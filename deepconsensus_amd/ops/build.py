@@ -11,7 +11,11 @@ import shutil
 
 _OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 _BUILD_DIR = os.path.join(_OPS_DIR, "_build")
-_SRC = os.path.join(_OPS_DIR, "hip", "dc_kernels.hip")
+_SRCS = [
+    os.path.join(_OPS_DIR, "hip", "dc_kernels.hip"),
+    os.path.join(_OPS_DIR, "hip", "banded_attn.hip"),
+    os.path.join(_OPS_DIR, "hip", "embed_gather.hip"),
+]
 EXT_NAME = "dc_hip_kernels"
 
 
@@ -23,7 +27,7 @@ def build(verbose: bool = False):
 
     module = load(
         name=EXT_NAME,
-        sources=[_SRC],
+        sources=_SRCS,
         build_directory=_BUILD_DIR,
         extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
         verbose=verbose,

@@ -1,12 +1,8 @@
 // DeepConsensus-AMD HIP/CDNA4 kernels (gfx950 / MI355X).
 //
 // Kernel inventory:
-//  * fused_embed_condense — K2+K3: the subread-stack embedding gathers
-//    (reference networks.py:436-516) fused with the no-bias condenser GEMM
-//    (networks.py:426-434) via precomputed per-row fused tables
-//    T'_row[id] = (table_f[id] * sqrt(w) * (id != 0)) @ Wc[row_slice].
-//    One pass over the [B, R, L] feature tensor emits [B, L, H] bf16 directly,
-//    never materializing the 560-wide concat.
+//  * embed_gather (embed_gather.hip) — K2; banded_attn (banded_attn.hip)
+//    — K5-K7 core.
 //  * fused_ln_head_qv — K10+K11+K12: final LayerNorm (eps 1e-6, fp32,
 //    encoder_stack.py:131-133) + Dense(5) head (networks.py:342-345) +
 //    softmax + argmax + Phred QV with linear calibration and cap
@@ -24,118 +20,6 @@
 #define DC_CHECK(x, msg) TORCH_CHECK(x, msg)
 
 using bf16 = __hip_bfloat16;
-
-// ---------------------------------------------------------------------------
-// fused_embed_condense
-// ---------------------------------------------------------------------------
-// rows:        [B, R, L] float32 (feature values; clipped upstream)
-// fused_table: [V_total, H] bf16
-// row_offset:  [R] int32  start row of this input row's fused table
-// row_shift:   [R] int32  added to the raw value before lookup (ccs_bq: +1)
-// row_vocab:   [R] int32  vocab size of this row's table (clamp bound)
-// out:         [B, L, H] bf16
-//
-// Geometry: one workgroup = TPP*TILE_L threads = one (b, l-tile); ids staged
-// in LDS; each thread owns NDIM output dims of one position, accumulating in
-// fp32 VGPRs with 16-byte bf16x8 table loads.
-namespace {
-
-constexpr int TILE_L = 64;   // positions per workgroup
-constexpr int TPP = 5;       // threads per position
-constexpr int NDIM = 56;     // dims per thread (TPP * NDIM = 280)
-
-__global__ __launch_bounds__(TILE_L * TPP) void fused_embed_condense_kernel(
-    const float* __restrict__ rows,
-    const bf16* __restrict__ fused_table,
-    const int* __restrict__ row_offset,
-    const int* __restrict__ row_shift,
-    const int* __restrict__ row_vocab,
-    bf16* __restrict__ out,
-    int B, int R, int L, int H) {
-  __shared__ int ids[128][TILE_L];  // R <= 128 supported (max_passes <= 30)
-
-  const int tiles_per_b = (L + TILE_L - 1) / TILE_L;
-  const int b = blockIdx.x / tiles_per_b;
-  const int l0 = (blockIdx.x % tiles_per_b) * TILE_L;
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;  // 5 waves
-
-  // Stage the id tile: wave w loads rows r = w, w+5, ... (64 lanes = 64 l's).
-  const float* rows_b = rows + (size_t)b * R * L;
-  for (int r = wave; r < R; r += TILE_L * TPP / 64) {
-    int l = l0 + lane;
-    float v = (l < L) ? rows_b[(size_t)r * L + l] : 0.f;
-    int id = (int)v + row_shift[r];
-    int vmax = row_vocab[r] - 1;
-    id = id < 0 ? 0 : (id > vmax ? vmax : id);
-    ids[r][lane] = row_offset[r] + id;
-  }
-  __syncthreads();
-
-  // Accumulate: thread handles position p = tid / TPP, dim chunk c = tid % TPP.
-  const int p = tid / TPP;
-  const int c = tid % TPP;
-  const int l = l0 + p;
-  if (l >= L) return;
-
-  float acc[NDIM];
-#pragma unroll
-  for (int d = 0; d < NDIM; ++d) acc[d] = 0.f;
-
-  const int dim0 = c * NDIM;
-  for (int r = 0; r < R; ++r) {
-    const bf16* trow = fused_table + (size_t)ids[r][p] * H + dim0;
-    // NDIM = 56 bf16 = 7 x 16-byte loads.
-#pragma unroll
-    for (int v = 0; v < NDIM / 8; ++v) {
-      // 16B vector load of 8 bf16.
-      const uint4 raw = *reinterpret_cast<const uint4*>(trow + v * 8);
-      const bf16* e = reinterpret_cast<const bf16*>(&raw);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[v * 8 + j] += __bfloat162float(e[j]);
-    }
-  }
-
-  bf16* orow = out + ((size_t)b * L + l) * H + dim0;
-#pragma unroll
-  for (int v = 0; v < NDIM / 8; ++v) {
-    uint4 raw;
-    bf16* e = reinterpret_cast<bf16*>(&raw);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) e[j] = __float2bfloat16(acc[v * 8 + j]);
-    *reinterpret_cast<uint4*>(orow + v * 8) = raw;
-  }
-}
-
-}  // namespace
-
-at::Tensor fused_embed_condense(
-    at::Tensor rows, at::Tensor fused_table, at::Tensor row_offset,
-    at::Tensor row_shift, at::Tensor row_vocab) {
-  DC_CHECK(rows.is_cuda() && rows.dtype() == at::kFloat,
-           "rows must be float32 on device");
-  DC_CHECK(fused_table.dtype() == at::kBFloat16, "fused_table must be bf16");
-  DC_CHECK(rows.is_contiguous() && fused_table.is_contiguous(),
-           "contiguous inputs required");
-  const int B = rows.size(0), R = rows.size(1), L = rows.size(2);
-  const int H = fused_table.size(1);
-  DC_CHECK(R <= 128, "R must be <= 128");
-  DC_CHECK(H % (TPP * 8) == 0 && H / TPP == NDIM,
-           "H must equal 280 for this kernel build");
-  auto out = at::empty({B, L, H}, rows.options().dtype(at::kBFloat16));
-  const int tiles_per_b = (L + TILE_L - 1) / TILE_L;
-  dim3 grid(B * tiles_per_b);
-  dim3 block(TILE_L * TPP);
-  hipStream_t stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(fused_embed_condense_kernel, grid, block, 0, stream,
-                     rows.data_ptr<float>(),
-                     reinterpret_cast<bf16*>(fused_table.data_ptr()),
-                     row_offset.data_ptr<int>(), row_shift.data_ptr<int>(),
-                     row_vocab.data_ptr<int>(),
-                     reinterpret_cast<bf16*>(out.data_ptr()), B, R, L, H);
-  return out;
-}
 
 // ---------------------------------------------------------------------------
 // fused_ln_head_qv
@@ -285,9 +169,17 @@ std::vector<at::Tensor> fused_ln_head_qv(
   return {bases, quals};
 }
 
+// Defined in banded_attn.hip / embed_gather.hip.
+at::Tensor banded_attn(at::Tensor qkv, int64_t H, int64_t win);
+at::Tensor embed_gather(at::Tensor rows, at::Tensor table_flat,
+                        at::Tensor row_shift, at::Tensor row_vocab,
+                        at::Tensor chunk_cnt, at::Tensor chunk_entries);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("fused_embed_condense", &fused_embed_condense,
-        "Fused subread-stack embedding + condenser (K2+K3)");
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
         "Fused final LayerNorm + head + softmax + QV (K10+K11+K12)");
+  m.def("banded_attn", &banded_attn,
+        "Banded multi-head self-attention forward (K5-K7 core)");
+  m.def("embed_gather", &embed_gather,
+        "Subread-stack embedding gather (K2)");
 }

@@ -1,0 +1,103 @@
+// Subread-stack embedding gather (K2) for gfx950.
+//
+// Reference semantics: 85 per-row embedding lookups scaled by sqrt(width)
+// with id-0 zero-masking, concatenated to [B, L, concat_width]
+// (networks.py:436-506 + ModifiedOnDeviceEmbedding:42-63). The scale and
+// mask are folded into pre-scaled bf16 tables host-side; the condenser GEMM
+// (K3) runs separately on hipBLASLt.
+//
+// Geometry: the concat width is processed in 16-byte chunks (8 bf16 cols);
+// thread t of a 64-position tile owns chunk t%NCHUNK of position t/NCHUNK,
+// so stores are perfectly coalesced 16-B writes. Each chunk resolves via a
+// host-built map to 1..4 (row, table, width) entries; the id tile is staged
+// in LDS.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+namespace {
+
+constexpr int TILE_P = 64;   // positions per workgroup
+constexpr int MAX_R = 160;   // input rows supported (max_passes <= 38)
+
+__global__ __launch_bounds__(256) void embed_gather_kernel(
+    const float* __restrict__ rows,
+    const bf16* __restrict__ table_flat,   // concatenated scaled tables
+    const int* __restrict__ row_shift,     // [R]
+    const int* __restrict__ row_vocab,     // [R]
+    const int* __restrict__ chunk_cnt,     // [NCHUNK]
+    const int4* __restrict__ chunk_entries,  // [NCHUNK*4] (row, elem_base, width, pad)
+    bf16* __restrict__ out,
+    int B, int R, int L, int nchunk) {
+  __shared__ short ids[MAX_R][TILE_P];
+
+  const int tiles_per_b = (L + TILE_P - 1) / TILE_P;
+  const int b = blockIdx.x / tiles_per_b;
+  const int l0 = (blockIdx.x % tiles_per_b) * TILE_P;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  const float* rows_b = rows + (size_t)b * R * L;
+  for (int r = wave; r < R; r += 4) {
+    const int l = l0 + lane;
+    const float v = (l < L) ? rows_b[(size_t)r * L + l] : 0.f;
+    int id = (int)v + row_shift[r];
+    const int vmax = row_vocab[r] - 1;
+    ids[r][lane] = (short)(id < 0 ? 0 : (id > vmax ? vmax : id));
+  }
+  __syncthreads();
+
+  const int H = nchunk * 8;
+  const int npos = min(TILE_P, L - l0);
+  for (int t = tid; t < npos * nchunk; t += 256) {
+    const int p = t / nchunk, c = t % nchunk;
+    ushort val[8];
+    int pos = 0;
+    const int cnt = chunk_cnt[c];
+    for (int k = 0; k < cnt; ++k) {
+      const int4 e = chunk_entries[c * 4 + k];  // x=row, y=elem_base, z=width
+      const int id = ids[e.x][p];
+      const ushort* src = reinterpret_cast<const ushort*>(
+          table_flat + e.y + (size_t)id * e.z);
+      for (int j = 0; j < e.z; ++j) val[pos++] = src[j];
+    }
+    uint4 raw;
+    memcpy(&raw, val, 16);
+    *reinterpret_cast<uint4*>(
+        out + ((size_t)b * L + l0 + p) * H + c * 8) = raw;
+  }
+}
+
+}  // namespace
+
+at::Tensor embed_gather(
+    at::Tensor rows, at::Tensor table_flat, at::Tensor row_shift,
+    at::Tensor row_vocab, at::Tensor chunk_cnt, at::Tensor chunk_entries) {
+  TORCH_CHECK(rows.is_cuda() && rows.dtype() == at::kFloat,
+              "rows must be float32 on device");
+  TORCH_CHECK(table_flat.dtype() == at::kBFloat16, "tables must be bf16");
+  auto rc = rows.contiguous();
+  const int B = rc.size(0), R = rc.size(1), L = rc.size(2);
+  TORCH_CHECK(R <= MAX_R, "too many rows");
+  const int nchunk = chunk_cnt.size(0);
+  auto out = at::empty({B, L, nchunk * 8},
+                       rc.options().dtype(at::kBFloat16));
+  const int tiles_per_b = (L + TILE_P - 1) / TILE_P;
+  dim3 grid(B * tiles_per_b);
+  dim3 block(256);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(embed_gather_kernel, grid, block, 0, stream,
+                     rc.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(table_flat.data_ptr()),
+                     row_shift.data_ptr<int>(), row_vocab.data_ptr<int>(),
+                     chunk_cnt.data_ptr<int>(),
+                     reinterpret_cast<int4*>(chunk_entries.data_ptr<int>()),
+                     reinterpret_cast<bf16*>(out.data_ptr()),
+                     B, R, L, nchunk);
+  return out;
+}
