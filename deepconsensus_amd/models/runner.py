@@ -204,6 +204,15 @@ class InferenceRunner:
         self.w_head = model.fc1.weight.detach().float().contiguous().to(dev)
         self.b_head = model.fc1.bias.detach().float().to(dev)
 
+    def _attn(self, qkv: torch.Tensor) -> torch.Tensor:
+        b, l, w = qkv.shape
+        d = w // (3 * self.num_heads)
+        if d == 140 and 32 <= l <= 104 and 1 <= self.win <= 12:
+            return self.ext.banded_attn_mfma(
+                qkv, self.num_heads, self.win, d ** -0.5
+            )
+        return self.ext.banded_attn(qkv, self.num_heads, self.win)
+
     @torch.no_grad()
     def encode_native(self, rows: torch.Tensor) -> torch.Tensor:
         """Native path up to (but excluding) the final LayerNorm: [B,L,H] bf16."""
@@ -221,7 +230,7 @@ class InferenceRunner:
             for lw in self.layer_w:
                 flat = x.view(b * l, h)
                 qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
-                a = self.ext.banded_attn(qkv, self.num_heads, self.win)
+                a = self._attn(qkv)
                 y = a.view(b * l, h) @ lw["wout_t"]
                 x = torch.add(flat, y, alpha=lw["alpha_attn"]).view(b, l, h)
                 flat = x.view(b * l, h)

@@ -14,6 +14,7 @@ _BUILD_DIR = os.path.join(_OPS_DIR, "_build")
 _SRCS = [
     os.path.join(_OPS_DIR, "hip", "dc_kernels.hip"),
     os.path.join(_OPS_DIR, "hip", "banded_attn.hip"),
+    os.path.join(_OPS_DIR, "hip", "banded_attn_mfma.hip"),
     os.path.join(_OPS_DIR, "hip", "embed_gather.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"

@@ -1,0 +1,230 @@
+// MFMA banded attention forward for gfx950 (production shape D=140, L<=104).
+//
+// Same math as banded_attn.hip (attention_layer.py:196-218: banded QK^T,
+// fp32 softmax over |i-j| <= win, P V), restructured for the CDNA4 matrix
+// cores:
+//  * one workgroup = one (b, h); 4 waves, each owning 32 query rows;
+//  * swapped QK^T — mfma(K, Q) — so each lane holds a full query row's band
+//    scores and softmax is in-lane (one shfl_xor(32) pair to merge halves);
+//  * P repacked to MFMA A-fragments with v_cvt_pk_bf16_f32 +
+//    permlane32_swap (the cdna_hip_programming.md T12 pattern);
+//  * K staged row-major in LDS (stride 152 elems: conflict-free b128 lane
+//    groups), V staged TRANSPOSED (Vt[dim][key], stride 168, keys shifted by
+//    V_OFF=12 so every PV fragment read is 16-byte aligned even at the
+//    negative key-window of the first row block);
+//  * 38 v_mfma_f32_32x32x16_bf16 per wave per (b,h), fp32 accumulation;
+//    ~79 KB LDS => 2 workgroups per CU.
+//
+// Out-of-band and out-of-range keys are masked to -inf before softmax; the
+// corresponding P entries are exactly 0, and the V slots they multiply are
+// zero-filled, so clamped/padded reads never contribute.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int AM_L = 104;        // max rows (production window: 100)
+constexpr int AM_D = 140;        // head dim (production: 280/2)
+constexpr int K_STRIDE = 152;    // K LDS row stride (bf16), 16B-aligned rows
+// Key slot shift is `win` at runtime: slot = win + key, so PV fragment
+// reads land at 32w+16s+8hi (always 16-B aligned) for any win <= 12.
+constexpr int V_STRIDE = 168;    // Vt row stride (key slots)
+constexpr int V_ROWS = 144;      // Vt rows (dims 0..143, 140..143 zero)
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+__global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
+    const bf16* __restrict__ qkv, bf16* __restrict__ out,
+    int B, int L, int H, int win, float scale) {
+  constexpr int D = AM_D;
+  const int b = blockIdx.x / H;
+  const int h = blockIdx.x % H;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int c = lane & 31;
+  const int hi = lane >> 5;
+
+  __shared__ __attribute__((aligned(16))) bf16 k_lds[AM_L][K_STRIDE];
+  __shared__ __attribute__((aligned(16))) bf16 vt_lds[V_ROWS][V_STRIDE];
+
+  const size_t RS = (size_t)3 * H * D;  // qkv row stride (elems)
+  const bf16* base = qkv + (size_t)b * L * RS + (size_t)h * D;
+
+  // ---- Stage K row-major (dims 140..151 zeroed). ----
+  for (int idx = tid; idx < L * (K_STRIDE / 2); idx += 256) {
+    const int r = idx / (K_STRIDE / 2), d2 = idx % (K_STRIDE / 2);
+    unsigned v = 0;
+    if (2 * d2 + 1 < D) {
+      v = *reinterpret_cast<const unsigned*>(base + r * RS + H * D + 2 * d2);
+    }
+    *reinterpret_cast<unsigned*>(&k_lds[r][2 * d2]) = v;
+  }
+  // ---- Stage V transposed: V[key][dim] -> vt_lds[dim][V_OFF + key]. ----
+  for (int idx = tid; idx < L * 70; idx += 256) {
+    const int r = idx / 70, d2 = idx % 70;
+    const unsigned v = *reinterpret_cast<const unsigned*>(
+        base + r * RS + 2 * H * D + 2 * d2);
+    vt_lds[2 * d2][win + r] =
+        __ushort_as_bfloat16((unsigned short)(v & 0xffffu));
+    if (2 * d2 + 1 < D)
+      vt_lds[2 * d2 + 1][win + r] =
+          __ushort_as_bfloat16((unsigned short)(v >> 16));
+  }
+  // Zero-fill: key slots [0, V_OFF) and [V_OFF+L, V_STRIDE) on dims < D,
+  // and all slots on dims >= D (odd dim 141 is also written above w/ zero).
+  for (int d = 0; d < V_ROWS; ++d) {
+    for (int s = tid; s < V_STRIDE; s += 256) {
+      const bool pad_row = d >= D;
+      const bool pad_col = (s < win) || (s >= win + L);
+      if (pad_row || pad_col)
+        vt_lds[d][s] = __ushort_as_bfloat16((unsigned short)0);
+    }
+  }
+
+  // ---- Load Q B-fragments (this lane's query row, 9 k-steps of 16). ----
+  const int l0w = 32 * wave;
+  const int qrow = l0w + c;
+  bf16x8 qf[9];
+  {
+    const bool qv = qrow < L;
+    const unsigned short* qp =
+        reinterpret_cast<const unsigned short*>(base + (size_t)qrow * RS);
+#pragma unroll
+    for (int s = 0; s < 9; ++s) {
+      const int d0 = 16 * s + 8 * hi;
+      bf16x8 t = {};
+      if (qv) {
+        if (d0 + 8 <= D) {
+          t = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const bf16*>(qp) + d0);
+        } else {
+          unsigned short* tw = reinterpret_cast<unsigned short*>(&t);
+          for (int j = 0; j < 8; ++j) tw[j] = (d0 + j < D) ? qp[d0 + j] : 0;
+        }
+      }
+      qf[s] = t;
+    }
+  }
+  __syncthreads();
+
+  // ---- QK^T (swapped): S^T tiles; lane holds qrow=l0w+c, keys in regs. ----
+  const int kw0 = l0w - win;  // key window start (may be negative)
+  float st[32];
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    f32x16 acc = {};
+    const int krow = min(max(kw0 + 32 * t + c, 0), L - 1);
+#pragma unroll
+    for (int s = 0; s < 9; ++s) {
+      const bf16x8 a =
+          *reinterpret_cast<const bf16x8*>(&k_lds[krow][16 * s + 8 * hi]);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[s], acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) st[16 * t + r] = acc[r];
+  }
+
+  // ---- Masked softmax, in-lane + one half-merge. ----
+  // Key-local index of st[i]: kl(i) = 32*(i/16) + (i&3) + 8*((i&15)>>2) + 4*hi.
+  float m = -1e30f;
+#pragma unroll
+  for (int i = 0; i < 32; ++i) {
+    const int kl = 32 * (i >> 4) + (i & 3) + 8 * ((i & 15) >> 2) + 4 * hi;
+    const int kg = kw0 + kl;
+    const bool valid =
+        (kl >= c) && (kl <= c + 2 * win) && (kg >= 0) && (kg < L);
+    st[i] = valid ? st[i] * scale : -1e30f;
+    m = fmaxf(m, st[i]);
+  }
+  m = fmaxf(m, __shfl_xor(m, 32, 64));
+  float denom = 0.f;
+#pragma unroll
+  for (int i = 0; i < 32; ++i) {
+    st[i] = (st[i] <= -1e29f) ? 0.f : __expf(st[i] - m);
+    denom += st[i];
+  }
+  denom += __shfl_xor(denom, 32, 64);
+  const float inv = (denom > 0.f) ? 1.f / denom : 0.f;
+#pragma unroll
+  for (int i = 0; i < 32; ++i) st[i] *= inv;
+
+  // ---- Repack P to A-fragments: 4 k-steps x 4 dwords (T12 pattern).
+  // Lane pair (l, l+32) holds interleaved keys; after cvt_pk + two
+  // permlane32_swaps per k-step, lane l's fragment covers its 8 contiguous
+  // keys 16s+8*hi .. +7.
+  bf16x8 pa[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    const unsigned x = cvt_pk_bf16(st[8 * s + 0], st[8 * s + 1]);
+    const unsigned y = cvt_pk_bf16(st[8 * s + 2], st[8 * s + 3]);
+    const unsigned x2 = cvt_pk_bf16(st[8 * s + 4], st[8 * s + 5]);
+    const unsigned y2 = cvt_pk_bf16(st[8 * s + 6], st[8 * s + 7]);
+    const auto rx = __builtin_amdgcn_permlane32_swap(x, x2, false, false);
+    const auto ry = __builtin_amdgcn_permlane32_swap(y, y2, false, false);
+    unsigned u[4] = {(unsigned)rx[0], (unsigned)ry[0], (unsigned)rx[1],
+                     (unsigned)ry[1]};
+    pa[s] = *reinterpret_cast<const bf16x8*>(u);
+  }
+
+  // ---- PV: out[32 qrows x 140] in 5 col tiles. ----
+  // B-frag: Vt[vdim][win + kw0 + 16s + 8*hi ... +7] = Vt[vdim][32w + 16s
+  // + 8*hi ...], which is >= 0 and == 0 (mod 8): reads stay 16-B aligned.
+  bf16* ob = out + ((size_t)b * L) * (H * D) + (size_t)h * D;
+#pragma unroll
+  for (int ct = 0; ct < 5; ++ct) {
+    f32x16 acc = {};
+    const int vdim = min(32 * ct + c, V_ROWS - 1);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int kk = win + kw0 + 16 * s + 8 * hi;
+      const bf16x8 bfrag =
+          *reinterpret_cast<const bf16x8*>(&vt_lds[vdim][kk]);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s], bfrag, acc, 0, 0, 0);
+    }
+    const int col = 32 * ct + c;
+    if (col < D) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qr = l0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        if (qr < L)
+          ob[(size_t)qr * H * D + col] = __float2bfloat16(acc[r]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
+                            double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == at::kBFloat16,
+              "qkv must be bf16 on device");
+  auto q = qkv.contiguous();
+  const int B = q.size(0), L = q.size(1);
+  const int D = q.size(2) / (3 * H);
+  TORCH_CHECK(D == AM_D && L <= AM_L && L >= 32,
+              "banded_attn_mfma requires D=140, 32<=L<=104");
+  TORCH_CHECK(win <= 12 && win >= 1, "win must be in [1, 12]");
+  auto out = at::empty({B, L, H * D}, q.options());
+  dim3 grid(B * H);
+  dim3 block(256);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(banded_attn_mfma_kernel, grid, block, 0, stream,
+                     reinterpret_cast<bf16*>(q.data_ptr()),
+                     reinterpret_cast<bf16*>(out.data_ptr()),
+                     B, L, (int)H, (int)win, (float)scale);
+  return out;
+}

@@ -171,6 +171,8 @@ std::vector<at::Tensor> fused_ln_head_qv(
 
 // Defined in banded_attn.hip / embed_gather.hip.
 at::Tensor banded_attn(at::Tensor qkv, int64_t H, int64_t win);
+at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
+                            double scale);
 at::Tensor embed_gather(at::Tensor rows, at::Tensor table_flat,
                         at::Tensor row_shift, at::Tensor row_vocab,
                         at::Tensor chunk_cnt, at::Tensor chunk_entries);
@@ -179,7 +181,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
         "Fused final LayerNorm + head + softmax + QV (K10+K11+K12)");
   m.def("banded_attn", &banded_attn,
-        "Banded multi-head self-attention forward (K5-K7 core)");
+        "Banded multi-head self-attention forward (K5-K7 core, generic)");
+  m.def("banded_attn_mfma", &banded_attn_mfma,
+        "Banded MHA forward on MFMA (D=140, L<=104)");
   m.def("embed_gather", &embed_gather,
         "Subread-stack embedding gather (K2)");
 }

@@ -193,3 +193,25 @@ def test_end_to_end_native_calls(setup):
     assert bases.shape == (16, 100)
     assert quals.shape == (16, 100)
     assert int(quals.max()) <= 93
+
+
+def test_banded_attn_mfma_matches_torch(setup):
+    """MFMA attention kernel vs fp32 torch reference, multiple L and win."""
+    params, model, runner, rows = setup
+    for L, win, seed in [(100, 12, 0), (100, 6, 1), (64, 12, 2), (104, 12, 4)]:
+        torch.manual_seed(seed)
+        B, H, D = 4, 2, 140
+        qkv = torch.randn(B, L, 3 * H * D, device="cuda").to(torch.bfloat16)
+        out = runner.ext.banded_attn_mfma(qkv, H, win, D ** -0.5).float()
+        qb = qkv[..., : H * D].float().view(B, L, H, D).permute(0, 2, 1, 3)
+        kb = qkv[..., H * D: 2 * H * D].float().view(B, L, H, D).permute(0, 2, 1, 3)
+        vb = qkv[..., 2 * H * D:].float().view(B, L, H, D).permute(0, 2, 1, 3)
+        logits = torch.matmul(qb * D ** -0.5, kb.transpose(-1, -2))
+        i = torch.arange(L, device="cuda")
+        mask = (i[:, None] - i[None, :]).abs() <= win
+        logits = torch.where(mask, logits, torch.tensor(-1e9, device="cuda"))
+        w = torch.softmax(logits, -1)
+        ref = torch.matmul(w, vb).permute(0, 2, 1, 3).reshape(B, L, H * D)
+        err = (out - ref).abs()
+        assert err.max().item() < 0.06, (L, win, err.max().item())
+        assert err.mean().item() < 0.006, (L, win, err.mean().item())
