@@ -94,21 +94,50 @@ def main():
     )
 
     # Pinned host pool of pre-featurized batches (the host preprocess stage is
-    # pipelined/off-path in production; windows arrive pre-packed).
+    # pipelined/off-path in production; windows arrive pre-packed). Features
+    # fit int16 (max value 500), halving H2D traffic vs float32.
+    use_i16 = have_cuda and runner.native
     pool = []
     for i in range(args.pool_batches):
-        t = torch.from_numpy(make_synthetic_windows(params, batch, 97 + i))
+        arr = make_synthetic_windows(params, batch, 97 + i)
+        t = torch.from_numpy(arr.astype(np.int16) if use_i16 else arr)
         if have_cuda:
             t = t.pin_memory()
         pool.append(t)
 
-    def one_step(i: int):
-        rows = pool[i % len(pool)]
-        bases, quals = runner.forward_windows(rows)
-        # D2H of the uint8 calls (the output contract of the serving step).
-        return bases.to("cpu", non_blocking=True), quals.to(
-            "cpu", non_blocking=True
-        )
+    if have_cuda:
+        # Double-buffered H2D on a copy stream, overlapped with compute.
+        copy_stream = torch.cuda.Stream()
+        dev_bufs = [torch.empty_like(pool[0], device=device) for _ in range(2)]
+        ready = [torch.cuda.Event(), torch.cuda.Event()]
+        consumed = [torch.cuda.Event(), torch.cuda.Event()]
+
+        def prefetch(i: int):
+            buf = i % 2
+            with torch.cuda.stream(copy_stream):
+                # Don't overwrite a buffer a previous step is still reading.
+                copy_stream.wait_event(consumed[buf])
+                dev_bufs[buf].copy_(pool[i % len(pool)], non_blocking=True)
+                ready[buf].record(copy_stream)
+
+        prefetch(0)
+
+        def one_step(i: int):
+            buf = i % 2
+            cur = torch.cuda.current_stream()
+            cur.wait_event(ready[buf])
+            bases, quals = runner.forward_windows(dev_bufs[buf])
+            consumed[buf].record(cur)
+            prefetch(i + 1)
+            # D2H of the uint8 calls (the serving step's output contract).
+            return bases.to("cpu", non_blocking=True), quals.to(
+                "cpu", non_blocking=True
+            )
+    else:
+        def one_step(i: int):
+            rows = pool[i % len(pool)]
+            bases, quals = runner.forward_windows(rows)
+            return bases.cpu(), quals.cpu()
 
     # Warmup.
     for i in range(args.warmup):

@@ -195,6 +195,23 @@ class InferenceRunner:
                         alpha_ffn=float(l.ffn_wrap.alpha.detach()),
                     )
                 )
+            # Fold the ReZero alphas into the out-proj / FFN2 weights so the
+            # residual adds fuse into the GEMM epilogues (addmm beta=1).
+            for lw in self.layer_w:
+                lw["wout_t_a"] = lw["wout_t"] * lw["alpha_attn"]
+                lw["w2_t_a"] = lw["w2_t"] * lw["alpha_ffn"]
+                lw["b2_a"] = lw["b2"] * lw["alpha_ffn"]
+            # hipBLASLt fused bias+ReLU epilogue, when this torch exposes it.
+            self._addmm_act = hasattr(torch, "_addmm_activation")
+            if self._addmm_act:
+                try:
+                    torch._addmm_activation(
+                        torch.zeros(4, device=dev, dtype=bf16),
+                        torch.zeros(2, 4, device=dev, dtype=bf16),
+                        torch.zeros(4, 4, device=dev, dtype=bf16),
+                    )
+                except Exception:
+                    self._addmm_act = False
         else:
             self.layers_bf16 = (
                 copy.deepcopy(model.layers).to(bf16).to(dev)
@@ -227,16 +244,18 @@ class InferenceRunner:
             x = x + self.pos[:l]
         if self.rezero_fast:
             h = x.shape[-1]
+            flat = x.reshape(b * l, h)
             for lw in self.layer_w:
-                flat = x.view(b * l, h)
                 qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
                 a = self._attn(qkv)
-                y = a.view(b * l, h) @ lw["wout_t"]
-                x = torch.add(flat, y, alpha=lw["alpha_attn"]).view(b, l, h)
-                flat = x.view(b * l, h)
-                ff = torch.addmm(lw["b1"], flat, lw["w1_t"]).relu_()
-                y = torch.addmm(lw["b2"], ff, lw["w2_t"])
-                x = torch.add(flat, y, alpha=lw["alpha_ffn"]).view(b, l, h)
+                # Residual fused into the GEMM epilogue (alpha pre-folded).
+                flat = torch.addmm(flat, a.view(b * l, h), lw["wout_t_a"])
+                if self._addmm_act:
+                    ff = torch._addmm_activation(lw["b1"], flat, lw["w1_t"])
+                else:
+                    ff = torch.addmm(lw["b1"], flat, lw["w1_t"]).relu_()
+                flat = torch.addmm(flat, ff, lw["w2_t_a"]).add_(lw["b2_a"])
+            x = flat.view(b, l, h)
         else:
             for layer in self.layers_bf16:
                 x, _ = layer(x, training=False)

@@ -24,8 +24,9 @@ namespace {
 constexpr int TILE_P = 64;   // positions per workgroup
 constexpr int MAX_R = 160;   // input rows supported (max_passes <= 38)
 
+template <typename TIn>
 __global__ __launch_bounds__(256) void embed_gather_kernel(
-    const float* __restrict__ rows,
+    const TIn* __restrict__ rows,
     const bf16* __restrict__ table_flat,   // concatenated scaled tables
     const int* __restrict__ row_shift,     // [R]
     const int* __restrict__ row_vocab,     // [R]
@@ -42,10 +43,10 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  const float* rows_b = rows + (size_t)b * R * L;
+  const TIn* rows_b = rows + (size_t)b * R * L;
   for (int r = wave; r < R; r += 4) {
     const int l = l0 + lane;
-    const float v = (l < L) ? rows_b[(size_t)r * L + l] : 0.f;
+    const TIn v = (l < L) ? rows_b[(size_t)r * L + l] : (TIn)0;
     int id = (int)v + row_shift[r];
     const int vmax = row_vocab[r] - 1;
     ids[r][lane] = (short)(id < 0 ? 0 : (id > vmax ? vmax : id));
@@ -78,8 +79,9 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
 at::Tensor embed_gather(
     at::Tensor rows, at::Tensor table_flat, at::Tensor row_shift,
     at::Tensor row_vocab, at::Tensor chunk_cnt, at::Tensor chunk_entries) {
-  TORCH_CHECK(rows.is_cuda() && rows.dtype() == at::kFloat,
-              "rows must be float32 on device");
+  TORCH_CHECK(rows.is_cuda() &&
+                  (rows.dtype() == at::kFloat || rows.dtype() == at::kShort),
+              "rows must be float32 or int16 on device");
   TORCH_CHECK(table_flat.dtype() == at::kBFloat16, "tables must be bf16");
   auto rc = rows.contiguous();
   const int B = rc.size(0), R = rc.size(1), L = rc.size(2);
@@ -91,13 +93,24 @@ at::Tensor embed_gather(
   dim3 grid(B * tiles_per_b);
   dim3 block(256);
   hipStream_t stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(embed_gather_kernel, grid, block, 0, stream,
-                     rc.data_ptr<float>(),
-                     reinterpret_cast<bf16*>(table_flat.data_ptr()),
-                     row_shift.data_ptr<int>(), row_vocab.data_ptr<int>(),
-                     chunk_cnt.data_ptr<int>(),
-                     reinterpret_cast<int4*>(chunk_entries.data_ptr<int>()),
-                     reinterpret_cast<bf16*>(out.data_ptr()),
-                     B, R, L, nchunk);
+  if (rc.dtype() == at::kFloat) {
+    hipLaunchKernelGGL(embed_gather_kernel<float>, grid, block, 0, stream,
+                       rc.data_ptr<float>(),
+                       reinterpret_cast<bf16*>(table_flat.data_ptr()),
+                       row_shift.data_ptr<int>(), row_vocab.data_ptr<int>(),
+                       chunk_cnt.data_ptr<int>(),
+                       reinterpret_cast<int4*>(chunk_entries.data_ptr<int>()),
+                       reinterpret_cast<bf16*>(out.data_ptr()),
+                       B, R, L, nchunk);
+  } else {
+    hipLaunchKernelGGL(embed_gather_kernel<short>, grid, block, 0, stream,
+                       rc.data_ptr<short>(),
+                       reinterpret_cast<bf16*>(table_flat.data_ptr()),
+                       row_shift.data_ptr<int>(), row_vocab.data_ptr<int>(),
+                       chunk_cnt.data_ptr<int>(),
+                       reinterpret_cast<int4*>(chunk_entries.data_ptr<int>()),
+                       reinterpret_cast<bf16*>(out.data_ptr()),
+                       B, R, L, nchunk);
+  }
   return out;
 }
