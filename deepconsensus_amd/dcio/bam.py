@@ -1,0 +1,538 @@
+"""Minimal BAM/BGZF reader + writer (pure Python, zlib-based).
+
+The reference uses pysam for all BAM I/O (pre_lib.py:50-91 SubreadGrouper,
+quick_inference.py:447-482 stream_bam, :738-760 BAM output). pysam is not
+available in this environment, so this module implements the subset of the
+BAM format (SAM spec v1.6 section 4) the pipeline needs:
+
+* BGZF block framing (gzip members with the BC extra field), read and write;
+* header parsing (text + reference list);
+* alignment record decode/encode: flag, pos, mapq, cigar, seq, qual, and the
+  aux tags used by DeepConsensus (zm, pw, ip, sn, ec, np, rq, RG, wl);
+* a pysam-like BamRead API surface: qname, is_unmapped/is_reverse/
+  is_supplementary, pos, reference_name, cigartuples, seq, query_qualities,
+  get_tag/has_tag, query_alignment_start/end, get_aligned_pairs.
+
+Random access (.bai) is intentionally not implemented; "fetch by reference
+name" for truth-to-CCS lookups is provided by a sequential scan into a dict
+(fetch_index), which matches how DeepConsensus actually uses the index
+(pre_lib.py:1001-1014: one alignment per CCS read name).
+"""
+from __future__ import annotations
+
+import io
+import struct
+import zlib
+from typing import Any, Dict, Iterator, List, Optional, Sequence, Tuple
+
+import numpy as np
+
+from deepconsensus_amd.utils import constants
+
+BAM_MAGIC = b"BAM\x01"
+_CIGAR_CHARS = "MIDNSHP=XB"
+_SEQ_NT16 = "=ACMGRSVTWYHKDBN"
+_NT16_OF = {c: i for i, c in enumerate(_SEQ_NT16)}
+
+# Flag bits.
+FUNMAP = 0x4
+FREVERSE = 0x10
+FSECONDARY = 0x100
+FSUPPLEMENTARY = 0x800
+
+
+# ---------------------------------------------------------------------------
+# BGZF framing
+# ---------------------------------------------------------------------------
+
+BGZF_EOF = bytes.fromhex(
+    "1f8b08040000000000ff0600424302001b0003000000000000000000"
+)
+
+
+def _read_bgzf_blocks(fh) -> Iterator[bytes]:
+    """Yields decompressed BGZF block payloads."""
+    while True:
+        header = fh.read(12)
+        if len(header) == 0:
+            return
+        if len(header) < 12:
+            raise ValueError("truncated BGZF header")
+        magic, _mtime, _xfl, _os, xlen = struct.unpack("<4sLBBH", header)
+        if magic[:2] != b"\x1f\x8b":
+            raise ValueError("not a gzip stream")
+        extra = fh.read(xlen)
+        bsize = None
+        off = 0
+        while off + 4 <= len(extra):
+            si1, si2, slen = struct.unpack_from("<BBH", extra, off)
+            if si1 == 66 and si2 == 67 and slen == 2:
+                bsize = struct.unpack_from("<H", extra, off + 4)[0]
+            off += 4 + slen
+        if bsize is None:
+            raise ValueError("missing BGZF BC extra field")
+        cdata_len = bsize - xlen - 19
+        cdata = fh.read(cdata_len)
+        _crc, isize = struct.unpack("<LL", fh.read(8))
+        data = zlib.decompress(cdata, -15)
+        if len(data) != isize:
+            raise ValueError("BGZF ISIZE mismatch")
+        if data:
+            yield data
+
+
+def _bgzf_compress_block(data: bytes, level: int = 6) -> bytes:
+    co = zlib.compressobj(level, zlib.DEFLATED, -15)
+    cdata = co.compress(data) + co.flush()
+    bsize = len(cdata) + 25 + 1
+    header = struct.pack(
+        "<4sLBBHBBHH",
+        b"\x1f\x8b\x08\x04",
+        0,
+        0,
+        255,
+        6,
+        66,
+        67,
+        2,
+        bsize,
+    )
+    footer = struct.pack("<LL", zlib.crc32(data) & 0xFFFFFFFF, len(data))
+    return header + cdata + footer
+
+
+class BgzfWriter:
+    """Streams data into BGZF blocks (<=64 KiB payload each)."""
+
+    def __init__(self, fh, level: int = 6):
+        self.fh = fh
+        self.level = level
+        self.buf = bytearray()
+
+    def write(self, data: bytes) -> None:
+        self.buf += data
+        while len(self.buf) >= 65000:
+            chunk = bytes(self.buf[:65000])
+            del self.buf[:65000]
+            self.fh.write(_bgzf_compress_block(chunk, self.level))
+
+    def close(self) -> None:
+        if self.buf:
+            self.fh.write(_bgzf_compress_block(bytes(self.buf), self.level))
+            self.buf = bytearray()
+        self.fh.write(BGZF_EOF)
+        self.fh.flush()
+
+
+# ---------------------------------------------------------------------------
+# Alignment record
+# ---------------------------------------------------------------------------
+
+
+class BamRead:
+    """One BAM alignment record with a pysam-like API subset."""
+
+    __slots__ = (
+        "qname", "flag", "ref_id", "pos", "mapq", "cigartuples", "seq",
+        "query_qualities", "tags", "_header", "next_ref_id", "next_pos",
+        "tlen",
+    )
+
+    def __init__(
+        self,
+        qname: str = "",
+        flag: int = 0,
+        ref_id: int = -1,
+        pos: int = -1,
+        mapq: int = 255,
+        cigartuples: Optional[List[Tuple[int, int]]] = None,
+        seq: str = "",
+        query_qualities: Optional[Sequence[int]] = None,
+        tags: Optional[Dict[str, Any]] = None,
+        header: Optional["BamHeader"] = None,
+    ):
+        self.qname = qname
+        self.flag = flag
+        self.ref_id = ref_id
+        self.pos = pos
+        self.mapq = mapq
+        self.cigartuples = cigartuples or []
+        self.seq = seq
+        self.query_qualities = (
+            np.asarray(query_qualities, dtype=np.int16)
+            if query_qualities is not None
+            else None
+        )
+        self.tags = tags or {}
+        self._header = header
+        self.next_ref_id = -1
+        self.next_pos = -1
+        self.tlen = 0
+
+    # pysam-compatible surface --------------------------------------------
+    @property
+    def query_sequence(self) -> str:
+        return self.seq
+
+    @query_sequence.setter
+    def query_sequence(self, v: str) -> None:
+        self.seq = v
+
+    @property
+    def is_unmapped(self) -> bool:
+        return bool(self.flag & FUNMAP)
+
+    @property
+    def is_reverse(self) -> bool:
+        return bool(self.flag & FREVERSE)
+
+    @property
+    def is_supplementary(self) -> bool:
+        return bool(self.flag & FSUPPLEMENTARY)
+
+    @property
+    def reference_name(self) -> Optional[str]:
+        if self._header is None or self.ref_id < 0:
+            return None
+        return self._header.references[self.ref_id][0]
+
+    @property
+    def reference_start(self) -> int:
+        return self.pos
+
+    @property
+    def cigarstring(self) -> str:
+        return "".join(f"{n}{_CIGAR_CHARS[op]}" for op, n in self.cigartuples)
+
+    @property
+    def cigar(self) -> List[Tuple[int, int]]:
+        return self.cigartuples
+
+    def get_tag(self, name: str) -> Any:
+        return self.tags[name]
+
+    def has_tag(self, name: str) -> bool:
+        return name in self.tags
+
+    def set_tag(self, name: str, value: Any) -> None:
+        self.tags[name] = value
+
+    @property
+    def query_alignment_start(self) -> int:
+        qpos = 0
+        for op, n in self.cigartuples:
+            if op == constants.CSOFT_CLIP:
+                qpos += n
+            elif op == constants.CHARD_CLIP:
+                continue
+            else:
+                break
+        return qpos
+
+    @property
+    def query_alignment_end(self) -> int:
+        qlen = len(self.seq)
+        clip = 0
+        for op, n in reversed(self.cigartuples):
+            if op == constants.CHARD_CLIP:
+                continue
+            if op == constants.CSOFT_CLIP:
+                clip += n
+            else:
+                break
+        return qlen - clip
+
+    def get_aligned_pairs(self) -> List[Tuple[Optional[int], Optional[int]]]:
+        """(query_pos, ref_pos) pairs per aligned column (SAM semantics)."""
+        pairs: List[Tuple[Optional[int], Optional[int]]] = []
+        qpos, rpos = 0, self.pos
+        for op, n in self.cigartuples:
+            if op in (constants.CMATCH, constants.CEQUAL, constants.CDIFF):
+                for _ in range(n):
+                    pairs.append((qpos, rpos))
+                    qpos += 1
+                    rpos += 1
+            elif op in (constants.CINS, constants.CSOFT_CLIP):
+                for _ in range(n):
+                    pairs.append((qpos, None))
+                    qpos += 1
+            elif op in (constants.CDEL, constants.CREF_SKIP):
+                for _ in range(n):
+                    pairs.append((None, rpos))
+                    rpos += 1
+            # CHARD_CLIP / CPAD consume nothing.
+        return pairs
+
+    def infer_query_length(self) -> int:
+        return len(self.seq)
+
+    def __repr__(self):
+        return f"BamRead({self.qname} flag={self.flag} pos={self.pos})"
+
+
+class BamHeader:
+    def __init__(self, text: str = "", references=None):
+        self.text = text
+        self.references: List[Tuple[str, int]] = references or []
+
+    @property
+    def ref_index(self) -> Dict[str, int]:
+        return {name: i for i, (name, _l) in enumerate(self.references)}
+
+
+# ---------------------------------------------------------------------------
+# Tag codec
+# ---------------------------------------------------------------------------
+
+_TAG_FMT = {"c": "b", "C": "B", "s": "h", "S": "H", "i": "i", "I": "I",
+            "f": "f"}
+_ARRAY_DTYPE = {"c": np.int8, "C": np.uint8, "s": np.int16, "S": np.uint16,
+                "i": np.int32, "I": np.uint32, "f": np.float32}
+
+
+def _parse_tags(buf: bytes) -> Dict[str, Any]:
+    tags: Dict[str, Any] = {}
+    off = 0
+    n = len(buf)
+    while off + 3 <= n:
+        name = buf[off:off + 2].decode()
+        typ = chr(buf[off + 2])
+        off += 3
+        if typ == "A":
+            tags[name] = chr(buf[off])
+            off += 1
+        elif typ in _TAG_FMT:
+            fmt = _TAG_FMT[typ]
+            size = struct.calcsize(fmt)
+            tags[name] = struct.unpack_from("<" + fmt, buf, off)[0]
+            off += size
+        elif typ == "Z":
+            end = buf.index(0, off)
+            tags[name] = buf[off:end].decode()
+            off = end + 1
+        elif typ == "H":
+            end = buf.index(0, off)
+            tags[name] = buf[off:end].decode()
+            off = end + 1
+        elif typ == "B":
+            sub = chr(buf[off])
+            count = struct.unpack_from("<I", buf, off + 1)[0]
+            dt = _ARRAY_DTYPE[sub]
+            arr = np.frombuffer(
+                buf, dtype=dt, count=count, offset=off + 5
+            ).copy()
+            tags[name] = arr
+            off += 5 + count * arr.dtype.itemsize
+        else:
+            raise ValueError(f"unknown tag type {typ!r}")
+    return tags
+
+
+def _encode_tags(tags: Dict[str, Any]) -> bytes:
+    out = bytearray()
+    for name, val in tags.items():
+        key = name.encode()
+        if isinstance(val, str) and len(val) == 1 and name in ("rs",):
+            out += key + b"A" + val.encode()
+        elif isinstance(val, (bool, int, np.integer)):
+            v = int(val)
+            if -2147483648 <= v <= 2147483647:
+                out += key + b"i" + struct.pack("<i", v)
+            else:
+                out += key + b"I" + struct.pack("<I", v)
+        elif isinstance(val, (float, np.floating)):
+            out += key + b"f" + struct.pack("<f", float(val))
+        elif isinstance(val, str):
+            out += key + b"Z" + val.encode() + b"\x00"
+        elif isinstance(val, (list, tuple, np.ndarray)):
+            arr = np.asarray(val)
+            if arr.dtype.kind == "f":
+                sub, dt = "f", np.float32
+            elif arr.dtype.kind in "iu" and arr.min(initial=0) >= 0 and arr.max(initial=0) <= 255:
+                sub, dt = "C", np.uint8
+            elif arr.dtype.kind in "iu" and abs(arr).max(initial=0) <= 32767:
+                sub, dt = "s", np.int16
+            else:
+                sub, dt = "i", np.int32
+            arr = arr.astype(dt)
+            out += (
+                key + b"B" + sub.encode()
+                + struct.pack("<I", arr.size) + arr.tobytes()
+            )
+        else:
+            raise ValueError(f"cannot encode tag {name}={val!r}")
+    return bytes(out)
+
+
+# ---------------------------------------------------------------------------
+# Reader / writer
+# ---------------------------------------------------------------------------
+
+
+class BamReader:
+    """Sequential BAM reader."""
+
+    def __init__(self, path: str):
+        self.path = path
+        self._fh = open(path, "rb")
+        self._stream = _ConcatStream(_read_bgzf_blocks(self._fh))
+        magic = self._stream.read(4)
+        if magic != BAM_MAGIC:
+            raise ValueError(f"{path} is not a BAM file")
+        (l_text,) = struct.unpack("<l", self._stream.read(4))
+        text = self._stream.read(l_text).decode(errors="replace").rstrip("\x00")
+        (n_ref,) = struct.unpack("<l", self._stream.read(4))
+        refs = []
+        for _ in range(n_ref):
+            (l_name,) = struct.unpack("<l", self._stream.read(4))
+            name = self._stream.read(l_name)[:-1].decode()
+            (l_ref,) = struct.unpack("<l", self._stream.read(4))
+            refs.append((name, l_ref))
+        self.header = BamHeader(text, refs)
+
+    def __iter__(self) -> Iterator[BamRead]:
+        return self
+
+    def __next__(self) -> BamRead:
+        head = self._stream.read(4)
+        if len(head) < 4:
+            raise StopIteration
+        (block_size,) = struct.unpack("<l", head)
+        buf = self._stream.read(block_size)
+        return self._decode(buf)
+
+    def _decode(self, buf: bytes) -> BamRead:
+        (ref_id, pos, l_read_name, mapq, _bin, n_cigar, flag, l_seq,
+         next_ref, next_pos, tlen) = struct.unpack_from("<llBBHHHllll", buf, 0)
+        off = 32
+        qname = buf[off:off + l_read_name - 1].decode()
+        off += l_read_name
+        cigartuples = []
+        for i in range(n_cigar):
+            (v,) = struct.unpack_from("<I", buf, off)
+            cigartuples.append((v & 0xF, v >> 4))
+            off += 4
+        nbytes = (l_seq + 1) // 2
+        seq_chars = []
+        for i in range(l_seq):
+            b = buf[off + i // 2]
+            code = (b >> 4) if i % 2 == 0 else (b & 0xF)
+            seq_chars.append(_SEQ_NT16[code])
+        seq = "".join(seq_chars)
+        off += nbytes
+        quals = np.frombuffer(buf, np.uint8, l_seq, off).astype(np.int16)
+        if l_seq and quals.size and quals[0] == 0xFF:
+            qual_arr = None
+        else:
+            qual_arr = quals
+        off += l_seq
+        tags = _parse_tags(buf[off:])
+        read = BamRead(
+            qname=qname, flag=flag, ref_id=ref_id, pos=pos, mapq=mapq,
+            cigartuples=cigartuples, seq=seq, query_qualities=qual_arr,
+            tags=tags, header=self.header,
+        )
+        read.next_ref_id, read.next_pos, read.tlen = next_ref, next_pos, tlen
+        return read
+
+    def close(self):
+        self._fh.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+
+class _ConcatStream:
+    """Byte stream over an iterator of chunks."""
+
+    def __init__(self, chunks: Iterator[bytes]):
+        self._chunks = chunks
+        self._buf = bytearray()
+
+    def read(self, n: int) -> bytes:
+        while len(self._buf) < n:
+            try:
+                self._buf += next(self._chunks)
+            except StopIteration:
+                break
+        out = bytes(self._buf[:n])
+        del self._buf[:n]
+        return out
+
+
+class BamWriter:
+    """Sequential BAM writer (BGZF)."""
+
+    def __init__(self, path: str, header: BamHeader):
+        self.header = header
+        self._fh = open(path, "wb")
+        self._w = BgzfWriter(self._fh)
+        text = header.text
+        if text and not text.endswith("\n"):
+            text += "\n"
+        payload = BAM_MAGIC + struct.pack("<l", len(text)) + text.encode()
+        payload += struct.pack("<l", len(header.references))
+        for name, length in header.references:
+            nb = name.encode() + b"\x00"
+            payload += struct.pack("<l", len(nb)) + nb
+            payload += struct.pack("<l", length)
+        self._w.write(payload)
+        self._ref_index = header.ref_index
+
+    def write(self, read: BamRead) -> None:
+        qname_b = read.qname.encode() + b"\x00"
+        l_seq = len(read.seq)
+        cig = b"".join(
+            struct.pack("<I", (n << 4) | op) for op, n in read.cigartuples
+        )
+        seq_b = bytearray((l_seq + 1) // 2)
+        for i, ch in enumerate(read.seq):
+            code = _NT16_OF.get(ch, 15)
+            if i % 2 == 0:
+                seq_b[i // 2] = code << 4
+            else:
+                seq_b[i // 2] |= code
+        if read.query_qualities is None:
+            qual_b = b"\xff" * l_seq
+        else:
+            qual_b = np.asarray(
+                read.query_qualities, dtype=np.uint8
+            ).tobytes()
+        tag_b = _encode_tags(read.tags)
+        rec = struct.pack(
+            "<llBBHHHllll",
+            read.ref_id, read.pos, len(qname_b), read.mapq, 0,
+            len(read.cigartuples), read.flag, l_seq,
+            read.next_ref_id, read.next_pos, read.tlen,
+        ) + qname_b + cig + bytes(seq_b) + qual_b + tag_b
+        self._w.write(struct.pack("<l", len(rec)) + rec)
+
+    def close(self):
+        self._w.close()
+        self._fh.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+
+def fetch_index(path: str) -> Dict[str, List[BamRead]]:
+    """Sequentially scans a BAM, grouping reads by reference name.
+
+    Replaces indexed `fetch(ref_name)` for the truth-to-CCS lookup
+    (pre_lib.py:1001-1014): DeepConsensus fetches exactly one primary
+    alignment per CCS read name.
+    """
+    out: Dict[str, List[BamRead]] = {}
+    with BamReader(path) as reader:
+        for read in reader:
+            name = read.reference_name
+            if name is None:
+                continue
+            out.setdefault(name, []).append(read)
+    return out

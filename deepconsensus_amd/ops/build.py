@@ -20,12 +20,32 @@ _SRCS = [
 EXT_NAME = "dc_hip_kernels"
 
 
+_SPACING_SRC = os.path.join(
+    os.path.dirname(_OPS_DIR), "preprocess", "_spacing_cpp.cpp"
+)
+
+
+def build_spacing(verbose: bool = False):
+    """Compiles (if needed) and loads the CPU _spacing extension."""
+    os.makedirs(_BUILD_DIR, exist_ok=True)
+    from torch.utils.cpp_extension import load
+
+    return load(
+        name="_spacing",
+        sources=[_SPACING_SRC],
+        build_directory=_BUILD_DIR,
+        extra_cflags=["-O3"],
+        verbose=verbose,
+    )
+
+
 def build(verbose: bool = False):
     """Compiles (if needed) and loads the dc_hip_kernels extension."""
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.makedirs(_BUILD_DIR, exist_ok=True)
     from torch.utils.cpp_extension import load
 
+    build_spacing(verbose=verbose)
     module = load(
         name=EXT_NAME,
         sources=_SRCS,

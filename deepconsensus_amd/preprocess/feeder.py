@@ -1,0 +1,250 @@
+"""ZMW feeding: subread grouping, CCS construction, truth labels.
+
+Behavioral parity with reference pre_lib.py:50-91 (SubreadGrouper),
+:965-998 (construct_ccs_read), :1001-1058 (fetch_label_alignment,
+read_truth_bedfile, read_truth_split), :1279-1367 (create_proc_feeder) and
+:1370-1384 (subreads_to_dc_example), on the in-repo BAM reader.
+"""
+from __future__ import annotations
+
+import collections
+import functools
+import logging
+from typing import Any, Dict, List, Optional, Union
+
+import numpy as np
+
+from deepconsensus_amd.dcio import bam
+from deepconsensus_amd.preprocess.expand import expand_clip_indent
+from deepconsensus_amd.preprocess.read import Read, space_out_subreads
+from deepconsensus_amd.preprocess.windows import DcConfig, DcExample
+from deepconsensus_amd.utils import constants
+
+Issue = constants.Issue
+
+
+class SubreadGrouper:
+    """Yields all mapped subreads of one ZMW (zm-tag grouped, streaming)."""
+
+    def __init__(self, subreads_to_ccs: str, reader_threads: int = 1):
+        del reader_threads  # the pure-python reader is single-threaded
+        self.bam_reader = iter(bam.BamReader(subreads_to_ccs))
+        self.keep_iter = True
+        self.subread_group: List[bam.BamRead] = []
+        first_read = next(self.bam_reader)
+        self.zmw = first_read.get_tag("zm")
+        if not first_read.is_unmapped:
+            self.subread_group.append(first_read)
+
+    def __iter__(self):
+        return self
+
+    def __next__(self) -> List[bam.BamRead]:
+        if not self.keep_iter:
+            raise StopIteration
+        while self.keep_iter:
+            try:
+                read = next(self.bam_reader)
+                if read.is_unmapped:
+                    continue
+            except StopIteration:
+                self.keep_iter = False
+                break
+            read_zmw = read.get_tag("zm")
+            if read_zmw == self.zmw:
+                self.subread_group.append(read)
+            else:
+                subreads_set = self.subread_group
+                self.subread_group = [read]
+                self.zmw = read_zmw
+                if subreads_set:
+                    return subreads_set
+        if self.subread_group:
+            out = self.subread_group
+            self.subread_group = []
+            return out
+        raise StopIteration
+
+
+def construct_ccs_read(ccs_bam_read: bam.BamRead) -> Read:
+    """CCS read with quality scores + aux tags (pre_lib.py:965-998)."""
+    ccs_seq = np.array(list(ccs_bam_read.seq), dtype="<U1")
+
+    def get_tag(read, tag_name):
+        try:
+            return read.get_tag(tag_name)
+        except KeyError:
+            return None
+
+    return Read(
+        name=ccs_bam_read.qname,
+        bases=ccs_seq,
+        cigar=np.repeat(np.uint8(constants.CMATCH), len(ccs_seq)),
+        pw=np.repeat(np.uint8(0), len(ccs_seq)),
+        ip=np.repeat(np.uint8(0), len(ccs_seq)),
+        sn=np.repeat(0, 4),
+        ec=get_tag(ccs_bam_read, "ec"),
+        np_num_passes=get_tag(ccs_bam_read, "np"),
+        rq=get_tag(ccs_bam_read, "rq"),
+        rg=get_tag(ccs_bam_read, "RG"),
+        strand=constants.Strand.UNKNOWN,
+        base_quality_scores=np.array(ccs_bam_read.query_qualities),
+        ccs_idx=np.arange(len(ccs_seq)),
+    )
+
+
+def fetch_label_alignment(
+    ccs_seqname: str,
+    truth_index: Dict[str, List[bam.BamRead]],
+    truth_range: Dict[str, Any],
+) -> Union[constants.Issue, Read]:
+    """Fetches the label aligned to a ccs sequence (pre_lib.py:1001-1014)."""
+    alns = truth_index.get(ccs_seqname)
+    if not alns:
+        return Issue.TRUTH_ALIGNMENT_NOT_FOUND
+    truth_alignment = alns[0]
+    if truth_alignment.is_supplementary:
+        return Issue.SUPP_TRUTH_ALIGNMENT
+    return expand_clip_indent(truth_alignment, truth_range)
+
+
+def read_truth_bedfile(truth_bed: str) -> Dict[str, Dict[str, Any]]:
+    bed_coords = {}
+    with open(truth_bed) as bedfile:
+        for line in bedfile:
+            contig, begin, end, ccs_seqname = line.strip().split("\t")[:4]
+            bed_coords[ccs_seqname] = {
+                "contig": contig,
+                "begin": int(begin),
+                "end": int(end),
+            }
+    return bed_coords
+
+
+def read_truth_split(split_fname: str) -> Dict[str, str]:
+    contig_split = {}
+    split_regions = {}
+    lower = split_fname.lower()
+    if any(x in lower for x in ["chm13", "hg00", "human"]):
+        genome = "HUMAN"
+    elif "maize" in lower:
+        genome = "MAIZE"
+    else:
+        raise ValueError(
+            f"{split_fname} does not correspond to any genome specified in "
+            "constants.py. Please either change the file name or add new "
+            "train/eval/test regions."
+        )
+    for i in constants.TRAIN_REGIONS[genome]:
+        split_regions[i] = "train"
+    for i in constants.EVAL_REGIONS[genome]:
+        split_regions[i] = "eval"
+    for i in constants.TEST_REGIONS[genome]:
+        split_regions[i] = "test"
+    with open(split_fname) as f:
+        for line in f:
+            contig, chrom = line.split()
+            if chrom in split_regions:
+                contig_split[contig] = split_regions[chrom]
+    return contig_split
+
+
+def create_proc_feeder(
+    subreads_to_ccs: str,
+    ccs_bam: str,
+    dc_config: DcConfig,
+    ins_trim: int = 0,
+    use_ccs_smart_windows: bool = False,
+    truth_bed: Optional[str] = None,
+    truth_to_ccs: Optional[str] = None,
+    truth_split: Optional[str] = None,
+    limit: int = 0,
+    bam_reader_threads: int = 1,
+):
+    """Generator feeding per-ZMW jobs (pre_lib.py:1279-1367)."""
+    main_counter = collections.Counter()
+    subread_grouper = SubreadGrouper(subreads_to_ccs, bam_reader_threads)
+    ccs_bam_h = iter(bam.BamReader(ccs_bam))
+
+    is_training = truth_bed and truth_to_ccs and truth_split
+    if is_training:
+        truth_index = bam.fetch_index(truth_to_ccs)
+        truth_ref_coords = read_truth_bedfile(truth_bed)
+        truth_split_dict = read_truth_split(truth_split)
+
+    def proc_feeder():
+        for read_set in subread_grouper:
+            main_counter["n_zmw_processed"] += 1
+            expand = functools.partial(
+                expand_clip_indent,
+                truth_range=None,
+                ins_trim=ins_trim,
+                counter=main_counter,
+            )
+            subreads = list(map(expand, read_set))
+            ccs_seqname = read_set[0].reference_name
+            while True:
+                ccs_bam_read = next(ccs_bam_h)
+                if ccs_bam_read.qname == ccs_seqname:
+                    break
+            if ccs_bam_read.qname != ccs_seqname:
+                raise ValueError(f"ccs bam does not contain {ccs_seqname}")
+
+            ccs_read = construct_ccs_read(ccs_bam_read)
+            window_widths = None
+            if use_ccs_smart_windows:
+                window_widths = np.array(ccs_bam_read.get_tag("wl"))
+            subreads.append(ccs_read)
+
+            if is_training:
+                truth_range = truth_ref_coords.get(ccs_seqname, None)
+                if not truth_range:
+                    logging.info(
+                        "No truth_range defined for %s.", ccs_seqname
+                    )
+                    main_counter["n_zmw_missing_truth_range"] += 1
+                    continue
+                label = fetch_label_alignment(
+                    ccs_seqname, truth_index, truth_range
+                )
+                if label == Issue.TRUTH_ALIGNMENT_NOT_FOUND:
+                    logging.info(
+                        "Unable to fetch label alignment for %s.",
+                        ccs_seqname,
+                    )
+                    main_counter["n_zmw_no_label_alignment"] += 1
+                    continue
+                elif label == Issue.SUPP_TRUTH_ALIGNMENT:
+                    main_counter["n_zmw_truth_label_supp_alignment"] += 1
+                    continue
+                subreads.append(label)
+                split = truth_split_dict.get(truth_range["contig"], None)
+                if not split:
+                    logging.info("No split defined for %s.", ccs_seqname)
+                    main_counter["n_zmw_missing_contig_split"] += 1
+                    continue
+            else:
+                split = "inference"
+            main_counter[f"n_zmw_{split}"] += 1
+            main_counter["n_zmw_pass"] += 1
+            yield (subreads, ccs_seqname, dc_config, split, window_widths)
+            if limit and main_counter["n_zmw_pass"] >= limit:
+                break
+
+    return proc_feeder, main_counter
+
+
+def subreads_to_dc_example(
+    subreads: List[Read],
+    ccs_seqname: str,
+    dc_config: DcConfig,
+    window_widths: Optional[np.ndarray] = None,
+) -> DcExample:
+    """Spaces reads and wraps them in a DcExample (pre_lib.py:1370-1384)."""
+    aln_reads = space_out_subreads(subreads)
+    return DcExample(
+        name=ccs_seqname,
+        reads=aln_reads,
+        config=dc_config,
+        window_widths=window_widths,
+    )

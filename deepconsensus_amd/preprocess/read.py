@@ -1,0 +1,375 @@
+"""Read abstraction + multi-read gap spacing.
+
+Behavioral parity with the reference's Read dataclass
+(pre_lib.py:111-421) and the spacing state machine
+(pre_lib.py:176-276 setup_spacing/move/add_gap/next_is_insertion/put_spacing,
+:1242-1276 space_out_subreads): all subreads + CCS (+ label) are co-spaced
+left-to-right so every insertion in any read gets a column in all reads;
+insertions at the same junction share (left-aligned) columns; label reads
+write their insertion bases instead of signaling gaps.
+
+The O(total_columns x n_reads) scan is the reference's known CPU hot loop;
+space_out_subreads dispatches to the C++ extension
+(deepconsensus_amd.preprocess._spacing) when built and falls back to the
+pure-Python state machine below (both produce identical _seq_indices).
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import Any, Dict, List, Optional, Union
+
+import numpy as np
+
+from deepconsensus_amd.utils import constants
+from deepconsensus_amd.utils import phred
+
+
+def right_pad(arr: np.ndarray, length: int, value: Any) -> np.ndarray:
+    pad_amt = length - len(arr)
+    return np.pad(arr, (0, pad_amt), "constant", constant_values=value)[
+        :length
+    ]
+
+
+@dataclasses.dataclass
+class Read:
+    """One aligned sequence (subread, CCS, or label) in CCS space."""
+
+    name: str
+    bases: np.ndarray  # '|S1' char array
+    cigar: np.ndarray  # uint8 cigar op per column
+    pw: np.ndarray
+    ip: np.ndarray
+    sn: np.ndarray
+    strand: constants.Strand
+
+    ec: Optional[float] = None
+    np_num_passes: Optional[int] = None
+    rq: Optional[float] = None
+    rg: Optional[str] = None
+
+    ccs_idx: np.ndarray = dataclasses.field(
+        default_factory=lambda: np.empty(0, dtype=int)
+    )
+    base_quality_scores: np.ndarray = dataclasses.field(
+        default_factory=lambda: np.empty(0, dtype=np.uint8)
+    )
+    truth_idx: np.ndarray = dataclasses.field(
+        default_factory=lambda: np.empty(0, int)
+    )
+    truth_range: Union[Dict[str, Any], None] = None
+
+    # Spacing state.
+    _seq_indices: np.ndarray = dataclasses.field(
+        default_factory=lambda: np.empty(0, dtype=int)
+    )
+    _is_insertion: np.ndarray = dataclasses.field(
+        default_factory=lambda: np.empty(0, dtype=bool)
+    )
+    _seq_len: int = 0
+    _idx_seq: int = 0
+    idx_spaced: int = 0
+    spacing_done: bool = False
+
+    # --- spacing state machine (reference pre_lib.py:176-276) -----------
+    def setup_spacing(self):
+        self._seq_indices = np.zeros(len(self.bases), dtype=int)
+        self._is_insertion = self.cigar == constants.CINS
+        self._seq_len = len(self.bases)
+        self._idx_seq = 0
+        self.idx_spaced = 0
+        self.spacing_done = False
+
+    def move(self):
+        self._seq_indices[self._idx_seq] = self.idx_spaced
+        self._idx_seq += 1
+        self.idx_spaced += 1
+
+    def add_gap(self):
+        self.idx_spaced += 1
+
+    def is_out_of_bounds(self) -> bool:
+        return self._idx_seq >= self._seq_len
+
+    def next_is_insertion(self) -> bool:
+        if self.truth_range:
+            while (
+                not self.is_out_of_bounds()
+                and self._is_insertion[self._idx_seq]
+            ):
+                self._seq_indices[self._idx_seq] = self.idx_spaced
+                self._idx_seq += 1
+                self.idx_spaced += 1
+            return False
+        return bool(self._is_insertion[self._idx_seq])
+
+    def put_spacing(self, seq_len: int):
+        """Scatters the original arrays into their spaced positions."""
+        spaced_seq = np.full(seq_len, constants.GAP, dtype="<U1")
+        spaced_pw = np.zeros(seq_len, dtype=np.uint8)
+        spaced_ip = np.zeros(seq_len, dtype=np.uint8)
+        spaced_ccs_idx = np.full(seq_len, -1)
+        spaced_seq[self._seq_indices] = self.bases
+        spaced_pw[self._seq_indices] = self.pw
+        spaced_ip[self._seq_indices] = self.ip
+        spaced_ccs_idx[self._seq_indices] = self.ccs_idx
+        if self.truth_range:
+            spaced_cigar = np.full(
+                seq_len, constants.CHARD_CLIP, dtype=np.uint8
+            )
+            spaced_cigar[self._seq_indices] = self.cigar
+            self.cigar = spaced_cigar
+            truth_pos = np.full(seq_len, -1)
+            truth_idx = np.arange(
+                self.truth_range["begin"], self.truth_range["end"]
+            )
+            truth_aln_base = np.isin(
+                self.cigar, constants.READ_ADVANCING_OPS
+            )
+            assert len(truth_pos[truth_aln_base]) == len(truth_idx)
+            truth_pos[truth_aln_base] = truth_idx
+            self.truth_idx = truth_pos
+
+        self.bases = spaced_seq
+        self.pw = spaced_pw
+        self.ip = spaced_ip
+        self.ccs_idx = spaced_ccs_idx
+
+        if self.base_quality_scores.any():
+            spaced_bq = np.full(seq_len, -1)
+            spaced_bq[self._seq_indices] = self.base_quality_scores
+            self.base_quality_scores = spaced_bq
+
+    # --- derived views (reference pre_lib.py:252-421) --------------------
+    @property
+    def bases_encoded(self) -> np.ndarray:
+        out = np.zeros(self.bases.shape, dtype=constants.NP_DATA_TYPE)
+        for k, base in enumerate(constants.SEQ_VOCAB):
+            out[self.bases == base] = k
+        return out
+
+    @property
+    def avg_base_quality_score(self) -> float:
+        return phred.avg_phred(self.base_quality_scores)
+
+    @property
+    def zmw(self) -> int:
+        return int(self.name.split("/")[1])
+
+    @property
+    def label_coords(self) -> str:
+        if self.is_label:
+            begin = self.label_bounds.start
+            end = self.label_bounds.stop
+            return f'{self.truth_range["contig"]}:{begin}-{end}'
+        return ""
+
+    @property
+    def is_label(self) -> bool:
+        return self.truth_range is not None
+
+    @property
+    def ccs_bounds(self) -> slice:
+        masked = self.ccs_idx[self.ccs_idx != -1]
+        if masked.size == 0:
+            return slice(0, 0)
+        return slice(int(masked.min()), int(masked.max()))
+
+    @property
+    def label_bounds(self) -> slice:
+        masked = self.truth_idx[self.truth_idx != -1]
+        if masked.size == 0:
+            return slice(0, 0)
+        return slice(int(masked.min()), int(masked.max()))
+
+    def ccs_slice(self, start: int, end: int) -> "Read":
+        """Slices on ccs coordinates; bounds inclusive (pre_lib.py:308)."""
+        locs = np.where(
+            np.logical_and(self.ccs_idx >= start, self.ccs_idx <= end)
+        )[0]
+        if locs.any():
+            sl = slice(int(locs.min()), int(locs.max()) + 1)
+        else:
+            sl = slice(0, 0)
+        return Read(
+            name=self.name,
+            bases=self.bases[sl],
+            cigar=self.cigar[sl],
+            pw=self.pw[sl],
+            ip=self.ip[sl],
+            sn=self.sn,
+            strand=self.strand,
+            base_quality_scores=self.base_quality_scores[sl],
+            ec=self.ec,
+            np_num_passes=self.np_num_passes,
+            rq=self.rq,
+            rg=self.rg,
+            ccs_idx=self.ccs_idx[sl],
+            truth_idx=self.truth_idx[sl],
+            truth_range=self.truth_range,
+        )
+
+    def pad(self, pad_width: int) -> "Read":
+        if len(self) >= pad_width:
+            return self
+        return Read(
+            name=self.name,
+            bases=right_pad(self.bases, pad_width, constants.GAP),
+            cigar=right_pad(self.cigar, pad_width, constants.CHARD_CLIP),
+            pw=right_pad(self.pw, pad_width, 0),
+            ip=right_pad(self.ip, pad_width, 0),
+            sn=self.sn,
+            strand=self.strand,
+            base_quality_scores=right_pad(
+                self.base_quality_scores, pad_width, -1
+            ),
+            ec=self.ec,
+            np_num_passes=self.np_num_passes,
+            rq=self.rq,
+            rg=self.rg,
+            ccs_idx=right_pad(self.ccs_idx, pad_width, -1),
+            truth_idx=right_pad(self.truth_idx, pad_width, -1),
+            truth_range=self.truth_range,
+        )
+
+    def remove_gaps(self, pad_width: int) -> Union["Read", None]:
+        keep = self.bases != constants.GAP
+        if self.base_quality_scores.any():
+            bq = self.base_quality_scores[keep]
+        else:
+            bq = np.empty(0, dtype=np.uint8)
+        if int(keep.sum()) > pad_width:
+            return None
+        return Read(
+            name=self.name,
+            bases=self.bases[keep],
+            cigar=self.cigar[keep],
+            pw=self.pw[keep],
+            ip=self.ip[keep],
+            sn=self.sn,
+            strand=self.strand,
+            base_quality_scores=bq,
+            ec=self.ec,
+            np_num_passes=self.np_num_passes,
+            rq=self.rq,
+            rg=self.rg,
+            ccs_idx=self.ccs_idx[keep],
+            truth_idx=self.truth_idx[keep],
+            truth_range=self.truth_range,
+        ).pad(pad_width)
+
+    def __str__(self):
+        return "".join(self.bases)
+
+    def __len__(self):
+        return len(self.bases)
+
+    def __getitem__(self, r_slice: Union[slice, int]) -> "Read":
+        return Read(
+            name=self.name,
+            bases=self.bases[r_slice],
+            cigar=self.cigar[r_slice],
+            pw=self.pw[r_slice],
+            ip=self.ip[r_slice],
+            sn=self.sn,
+            strand=self.strand,
+            base_quality_scores=self.base_quality_scores[r_slice],
+            ec=self.ec,
+            np_num_passes=self.np_num_passes,
+            rq=self.rq,
+            rg=self.rg,
+            ccs_idx=self.ccs_idx[r_slice],
+            truth_idx=self.truth_idx[r_slice],
+        )
+
+    def __repr__(self):
+        if np.any(self.ccs_idx >= 0):
+            start = np.min(self.ccs_idx[self.ccs_idx >= 0])
+            end = np.max(self.ccs_idx, initial=0)
+        else:
+            start, end = 0, 0
+        return (
+            f"Read({self.name}) : CCS({start}-{end}) L={len(self.bases)} "
+            + self.label_coords
+        ).strip()
+
+
+def _space_out_python(subreads: List[Read]) -> int:
+    """Pure-Python spacing state machine (reference pre_lib.py:1242-1270)."""
+    while not all(r.spacing_done for r in subreads):
+        any_insertions = False
+        for r in subreads:
+            if r.spacing_done:
+                continue
+            if r.next_is_insertion():
+                any_insertions = True
+                break
+        for r in subreads:
+            if r.spacing_done:
+                continue
+            if any_insertions and not r.next_is_insertion():
+                r.add_gap()
+            else:
+                if not r.is_out_of_bounds():
+                    r.move()
+                if r.is_out_of_bounds():
+                    r.spacing_done = True
+    return max(r.idx_spaced for r in subreads)
+
+
+_spacing_ext = None
+_spacing_tried = False
+
+
+def _get_spacing_ext():
+    global _spacing_ext, _spacing_tried
+    if not _spacing_tried:
+        _spacing_tried = True
+        try:
+            import importlib.util
+            import os
+
+            so = os.path.join(
+                os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                "ops", "_build", "_spacing.so",
+            )
+            if os.path.exists(so):
+                import torch  # noqa: F401  (extension links libtorch)
+
+                spec = importlib.util.spec_from_file_location("_spacing", so)
+                mod = importlib.util.module_from_spec(spec)
+                spec.loader.exec_module(mod)
+                _spacing_ext = mod
+            else:
+                from deepconsensus_amd.ops import build as _b
+
+                _spacing_ext = _b.build_spacing()
+        except Exception:
+            _spacing_ext = None
+    return _spacing_ext
+
+
+def space_out_subreads(
+    subreads: List[Read], force_python: bool = False
+) -> List[Read]:
+    """Co-spaces all reads; see module docstring."""
+    for r in subreads:
+        r.setup_spacing()
+    ext = None if force_python else _get_spacing_ext()
+    if ext is not None:
+        is_ins = [r._is_insertion.astype(np.uint8) for r in subreads]
+        is_label = np.array(
+            [bool(r.truth_range) for r in subreads], dtype=np.uint8
+        )
+        seq_indices, spaced_lens = ext.space_out(is_ins, is_label)
+        max_len = 0
+        for r, idx, sl in zip(subreads, seq_indices, spaced_lens):
+            r._seq_indices = idx
+            r.idx_spaced = int(sl)
+            r.spacing_done = True
+            max_len = max(max_len, int(sl))
+    else:
+        max_len = _space_out_python(subreads)
+    for r in subreads:
+        r.put_spacing(max_len)
+    return subreads

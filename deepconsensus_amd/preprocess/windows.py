@@ -1,0 +1,384 @@
+"""DcConfig + DcExample: window slicing and feature extraction.
+
+Behavioral parity with the reference (pre_lib.py:450-819): DcConfig computes
+the row layout of the model input matrix (4 per-subread blocks of max_passes
+rows + CCS row + optional ccs_bq row + 4 SN rows); DcExample slices the
+spaced ZMW into fixed or CCS-provided "smart" windows, handles label
+overflow, and emits the (height x width x 1) float32 feature matrix, the
+inference feature dict, or a serialized tf.Example.
+"""
+from __future__ import annotations
+
+import collections
+import dataclasses
+from typing import Any, Dict, List, Optional, Tuple, Union
+
+import numpy as np
+
+from deepconsensus_amd.dcio import example_codec
+from deepconsensus_amd.preprocess.read import Read
+from deepconsensus_amd.utils import constants
+from deepconsensus_amd.utils import phred
+
+
+def dc_config_from_shape(
+    subreads_shape: Tuple[int, int, int], use_ccs_bq: bool = False
+) -> "DcConfig":
+    """Inverse constructor from tensor shape (pre_lib.py:424-447)."""
+    height, width, _ = subreads_shape
+    fixed_height = 6 if use_ccs_bq else 5
+    max_passes, remainder = divmod(
+        height - fixed_height, len(DcConfig.n_subread_features)
+    )
+    if remainder != 0:
+        raise ValueError(f"Invalid subreads shape {subreads_shape!r}.")
+    return DcConfig(max_passes, width, use_ccs_bq)
+
+
+class DcConfig:
+    """Row layout of the model input matrix (pre_lib.py:450-528)."""
+
+    n_subread_features = ["bases", "pw", "ip", "strand"]
+
+    def __init__(self, max_passes: int, max_length: int,
+                 use_ccs_bq: bool = False):
+        self.max_passes = max_passes
+        self.max_length = max_length
+        self.use_ccs_bq = use_ccs_bq
+        self.feature_rows = {
+            "bases": max_passes,
+            "pw": max_passes,
+            "ip": max_passes,
+            "strand": max_passes,
+            "ccs": 1,
+            "ccs_bq": 1 if use_ccs_bq else 0,
+            "sn": 4,
+        }
+        self.feature_indices = {}
+        i_rows = 0
+        for k, v in self.feature_rows.items():
+            self.feature_indices[k] = slice(i_rows, i_rows + v)
+            setattr(self, k, i_rows)
+            i_rows += v
+
+    def indices(self, feature: str, n_subreads: int = 0) -> slice:
+        if n_subreads:
+            assert feature in DcConfig.n_subread_features
+            n_rows = min(n_subreads, self.max_passes)
+            return slice(
+                getattr(self, feature), getattr(self, feature) + n_rows
+            )
+        assert feature not in DcConfig.n_subread_features
+        return slice(
+            getattr(self, feature),
+            getattr(self, feature) + self.feature_rows[feature],
+        )
+
+    @property
+    def tensor_height(self) -> int:
+        return sum(self.feature_rows.values())
+
+    def to_dict(self) -> Dict[str, str]:
+        return {
+            "max_passes": str(self.max_passes),
+            "max_length": str(self.max_length),
+            "tensor_height": str(self.tensor_height),
+            "tensor_width": str(self.max_length),
+        }
+
+
+@dataclasses.dataclass
+class DcExample:
+    """Container generating model inputs from spaced reads."""
+
+    name: str
+    reads: List[Read]
+    config: DcConfig
+    window_widths: Optional[np.ndarray] = None
+    counter: collections.Counter = dataclasses.field(
+        default_factory=collections.Counter
+    )
+
+    _width: Optional[int] = None
+    _ccs_width: Optional[int] = None
+    _overflow: bool = False
+
+    @property
+    def contig(self) -> Optional[str]:
+        if self.label:
+            return self.label.truth_range["contig"]
+        return None
+
+    @property
+    def is_training(self) -> bool:
+        return self.reads[-1].is_label
+
+    @property
+    def ccs(self) -> Read:
+        return self.reads[-2] if self.is_training else self.reads[-1]
+
+    @property
+    def label(self) -> Union[Read, None]:
+        return self.reads[-1] if self.is_training else None
+
+    @property
+    def label_coords(self) -> str:
+        return self.label.label_coords if self.is_training else ""
+
+    @property
+    def subreads(self) -> List[Read]:
+        return self.reads[:-2] if self.is_training else self.reads[:-1]
+
+    @property
+    def n_subreads(self) -> int:
+        return len(self.subreads)
+
+    @property
+    def keep_subreads(self) -> int:
+        return min(self.config.max_passes, self.n_subreads)
+
+    @property
+    def width(self) -> int:
+        if self._width is None:
+            self._width = len(self.ccs.bases)
+        return self._width
+
+    @property
+    def ccs_width(self) -> int:
+        if self._ccs_width is None:
+            self._ccs_width = len(str(self.ccs).rstrip())
+        return self._ccs_width
+
+    @property
+    def is_empty(self) -> bool:
+        return not (self.ccs.ccs_idx >= 0).any()
+
+    @property
+    def ccs_matches_label(self) -> bool:
+        from deepconsensus_amd.preprocess.read import right_pad
+
+        ccs = phred.left_shift_seq(self.ccs.bases_encoded)
+        label = phred.left_shift_seq(self.label.bases_encoded)
+        seq_len = max([len(ccs), len(label)])
+        ccs = right_pad(ccs, seq_len, 0)
+        label = right_pad(label, seq_len, 0)
+        return bool(np.equal(ccs, label).all())
+
+    def calculate_windows(self, example_width: int) -> List[int]:
+        """Window widths, fixed or from CCS smart-window widths
+        (pre_lib.py:625-650)."""
+        window_widths = []
+        last_pos = 0
+        if self.window_widths is not None:
+            ccs_calculated_width = 0
+            for window_width in self.window_widths:
+                original_width = 0
+                window_width_spaced = 0
+                while original_width < window_width:
+                    if (
+                        self.ccs.bases[last_pos + window_width_spaced]
+                        != constants.GAP
+                    ):
+                        original_width += 1
+                    window_width_spaced += 1
+                window_widths.append(window_width_spaced)
+                last_pos += window_width_spaced
+                ccs_calculated_width += window_width_spaced
+            assert ccs_calculated_width == self.ccs_width
+        else:
+            num_full = int(self.ccs_width / example_width)
+            if self.ccs_width % example_width > 0:
+                num_full += 1
+            window_widths = [example_width] * num_full
+        return window_widths
+
+    def iter_examples(self):
+        """Yields per-window DcExamples (pre_lib.py:652-697)."""
+        self.counter = collections.Counter()
+        max_length = self.config.max_length
+        start_pos = 0
+        for window_width in self.calculate_windows(max_length):
+            self.counter[f"example_width_bucket_{window_width}"] += 1
+            window = self[start_pos : start_pos + window_width]
+            if start_pos > self.ccs_width:
+                break
+            start_pos += window_width
+            if window.is_empty:
+                self.counter["n_examples_no_ccs_idx"] += 1
+                continue
+
+            if (
+                self.is_training
+                and len(window.label.bases) > max_length
+            ):
+                adjusted_label = window.label.remove_gaps(max_length)
+                if not adjusted_label:
+                    self.counter["n_examples_label_overflow"] += 1
+                    continue
+                self.counter["n_examples_adjusted_label"] += 1
+                window.reads[-1] = adjusted_label
+
+            self._overflow = False
+            if window_width > max_length:
+                self.counter["n_examples_overflow"] += 1
+                self._overflow = True
+                if self.is_training:
+                    continue
+            else:
+                self.counter["n_examples_skip_large_windows_keep"] += 1
+
+            reads = [x.pad(max_length) for x in window.reads]
+            yield DcExample(
+                self.name, reads, self.config, _overflow=self._overflow
+            )
+
+    def stack_subread_feature(self, name: str) -> np.ndarray:
+        max_passes = self.config.max_passes
+        return np.stack(
+            [getattr(x, name) for x in self.subreads[:max_passes]]
+        )
+
+    def extract_features(self) -> np.ndarray:
+        """The (height x width x 1) float32 model input
+        (pre_lib.py:704-744)."""
+
+        def repeat(feature):
+            return np.repeat(np.expand_dims(feature, -1), self.width, -1)
+
+        n_subreads = self.n_subreads
+        dims = (self.config.tensor_height, self.width)
+        data = np.zeros(shape=dims, dtype=constants.NP_DATA_TYPE)
+
+        bases_idx = self.config.indices("bases", n_subreads)
+        pw_idx = self.config.indices("pw", n_subreads)
+        ip_idx = self.config.indices("ip", n_subreads)
+        strand_idx = self.config.indices("strand", n_subreads)
+        ccs_idx = self.config.indices("ccs")
+        ccs_bq_idx = self.config.indices("ccs_bq")
+        sn_idx = self.config.indices("sn")
+
+        if n_subreads:
+            data[bases_idx] = self.stack_subread_feature("bases_encoded")
+            data[pw_idx] = self.stack_subread_feature("pw")
+            data[ip_idx] = self.stack_subread_feature("ip")
+            strand = self.stack_subread_feature("strand")
+            strand = strand.astype(constants.NP_DATA_TYPE)
+            data[strand_idx] = repeat(strand)
+        data[ccs_idx] = self.ccs.bases_encoded
+        if self.config.use_ccs_bq:
+            data[ccs_bq_idx] = self.ccs.base_quality_scores
+        if n_subreads:
+            data[sn_idx] = repeat(self.subreads[0].sn)
+        return np.expand_dims(data, -1)
+
+    def to_features_dict(self) -> Dict[str, Any]:
+        """Inference feature dict (pre_lib.py:746-762)."""
+        data = self.extract_features()
+        return {
+            "subreads": data,
+            "subreads/num_passes": self.keep_subreads,
+            "name": self.name,
+            "window_pos": self.ccs.ccs_bounds.start,
+            "ccs_base_quality_scores": self.ccs.base_quality_scores,
+            "overflow": self._overflow,
+            "ec": self.ccs.ec,
+            "np_num_passes": self.ccs.np_num_passes,
+            "rq": self.ccs.rq,
+            "rg": self.ccs.rg,
+        }
+
+    def tf_example(self) -> bytes:
+        """Serialized tf.train.Example (pre_lib.py:764-787)."""
+        data = self.extract_features()
+        feats: Dict[str, Tuple[str, Any]] = {
+            "subreads/encoded": (example_codec.BYTES, [data.tobytes()]),
+            "subreads/shape": (example_codec.INT64, list(data.shape)),
+            "subreads/num_passes": (
+                example_codec.INT64,
+                [self.keep_subreads],
+            ),
+            "name": (example_codec.BYTES, [self.name.encode()]),
+            "window_pos": (
+                example_codec.INT64,
+                [int(self.ccs.ccs_bounds.start)],
+            ),
+            "ccs_base_quality_scores": (
+                example_codec.INT64,
+                [int(x) for x in self.ccs.base_quality_scores],
+            ),
+        }
+        if self.is_training:
+            label = self.label.bases_encoded
+            feats["label/encoded"] = (example_codec.BYTES, [label.tobytes()])
+            feats["label/shape"] = (example_codec.INT64, list(label.shape))
+        return example_codec.encode_example(feats)
+
+    def __getitem__(self, r_slice) -> "DcExample":
+        if isinstance(r_slice, int):
+            raise NotImplementedError
+        reads = self.subreads + [self.ccs]
+        reads = [x[r_slice] for x in reads]
+        if self.label:
+            ccs_slice = self.ccs[r_slice].ccs_bounds
+            reads.append(
+                self.label.ccs_slice(ccs_slice.start, ccs_slice.stop)
+            )
+        return DcExample(self.name, reads, self.config)
+
+    def __repr__(self):
+        preview = self[:100]
+        start = preview.ccs.ccs_bounds.start
+        end = preview.ccs.ccs_bounds.stop
+        output = (
+            f"{self.name} CCS({start}-{end}) {self.label_coords}".strip()
+            + f'\n{"-" * (preview.width + 24)}\n'
+        )
+        for subread in self.subreads:
+            subread_range = subread.name.split("/")[2]
+            output += f"{subread_range:<20} {subread.strand} >{str(subread)}\n"
+        output += f'{"CCS":<22} >{str(preview.ccs)}\n'
+        if self.is_training:
+            output += f'{"Label":<22} >{str(self.label)}\n'
+        if self._overflow:
+            output += f'{"overflow":<22} >{self._overflow}\n'
+        return output
+
+
+def decode_bases(bases_encoded: np.ndarray) -> np.ndarray:
+    """Reverses base encoding (pre_lib.py:822-828)."""
+    n_subreads, max_length = bases_encoded.shape
+    bases = np.stack(
+        [np.full(max_length, constants.GAP, dtype="<U1")] * n_subreads
+    )
+    for k, base in enumerate(constants.SEQ_VOCAB):
+        bases[bases_encoded == k] = base
+    return bases
+
+
+def tf_example_to_features_dict(
+    serialized: bytes,
+    inference: bool = False,
+) -> Dict[str, Any]:
+    """Decodes a serialized tf.Example into the features dict
+    (pre_lib.py:902-963, without TF)."""
+    decoded = example_codec.decode_example(serialized)
+    features: Dict[str, Any] = {}
+    features["name"] = decoded["name"][1][0].decode()
+    features["subreads/shape"] = list(decoded["subreads/shape"][1])
+    features["subreads/num_passes"] = int(
+        decoded["subreads/num_passes"][1][0]
+    )
+    features["window_pos"] = int(decoded["window_pos"][1][0])
+    features["ccs_base_quality_scores"] = np.array(
+        decoded["ccs_base_quality_scores"][1], dtype=np.int64
+    )
+    data = np.frombuffer(
+        decoded["subreads/encoded"][1][0], dtype=constants.NP_DATA_TYPE
+    ).reshape(features["subreads/shape"])
+    features["subreads"] = data
+    if not inference and "label/encoded" in decoded:
+        label = np.frombuffer(
+            decoded["label/encoded"][1][0], dtype=constants.NP_DATA_TYPE
+        ).reshape(list(decoded["label/shape"][1]))
+        features["label"] = label
+    return features
