@@ -84,7 +84,8 @@ def _read_bgzf_blocks(fh) -> Iterator[bytes]:
 def _bgzf_compress_block(data: bytes, level: int = 6) -> bytes:
     co = zlib.compressobj(level, zlib.DEFLATED, -15)
     cdata = co.compress(data) + co.flush()
-    bsize = len(cdata) + 25 + 1
+    # BSIZE field = total block length - 1 = 12 + xlen(6) + cdata + 8 - 1.
+    bsize = len(cdata) + 25
     header = struct.pack(
         "<4sLBBHBBHH",
         b"\x1f\x8b\x08\x04",

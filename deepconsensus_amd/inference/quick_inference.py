@@ -1,0 +1,424 @@
+"""`deepconsensus run`: BAM -> polished FASTQ/BAM inference orchestrator.
+
+Behavioral parity with reference quick_inference.py:238-984 — ZMW streaming,
+multiprocess window preprocessing, skip-window fast path (overflow windows
+and windows whose average CCS base quality exceeds --skip_windows_above adopt
+the CCS sequence + calibrated CCS qualities), batched model execution,
+stitching, FASTQ/BAM output with ec/np/rq/RG/zm tags, a per-stage runtime CSV
+and an inference-stats JSON.
+
+MI355X-native differences: the model path is the InferenceRunner HIP pipeline
+(fused embed gather -> bf16 encoder with MFMA banded attention -> fused
+LN+head+QV kernel), windows cross to the device as int16 tensors, and
+preprocessing of the NEXT ZMW batch overlaps model execution of the current
+one (the reference serialized these stages).
+"""
+from __future__ import annotations
+
+import collections
+import concurrent.futures
+import dataclasses
+import itertools
+import json
+import logging
+import os
+import time
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from deepconsensus_amd.calibration import calibration as calibration_lib
+from deepconsensus_amd.models import checkpoint as ckpt_lib
+from deepconsensus_amd.models import config as cfg
+from deepconsensus_amd.models import data as data_lib
+from deepconsensus_amd.models.model import get_model
+from deepconsensus_amd.models.runner import InferenceRunner
+from deepconsensus_amd.postprocess import stitch as stitch_utils
+from deepconsensus_amd.preprocess import feeder as pre_feeder
+from deepconsensus_amd.preprocess.windows import DcConfig
+from deepconsensus_amd.utils import constants, phred
+
+log = logging.getLogger(__name__)
+
+
+@dataclasses.dataclass
+class InferenceOptions:
+    """Options shared across inference stages (quick_inference.py:238-275)."""
+
+    max_length: int = 100
+    example_height: int = 85
+    max_passes: int = 20
+    min_quality: int = 20
+    min_length: int = 0
+    batch_size: int = 1024
+    use_ccs_bq: bool = False
+    cpus: int = 0
+    skip_windows_above: int = 45
+    max_base_quality: int = constants.MAX_QUAL
+    dc_calibration_values: calibration_lib.QualityCalibrationValues = (
+        dataclasses.field(
+            default_factory=lambda: calibration_lib.parse_calibration_string(
+                "skip"
+            )
+        )
+    )
+    ccs_calibration_values: calibration_lib.QualityCalibrationValues = (
+        dataclasses.field(
+            default_factory=lambda: calibration_lib.parse_calibration_string(
+                "skip"
+            )
+        )
+    )
+    batch_zmws: int = 100
+    ins_trim: int = 0
+    use_ccs_smart_windows: bool = False
+
+
+def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
+    """Windows + counters for one ZMW (quick_inference.py:535-564)."""
+    zmw, subreads, dc_config, window_widths = one_zmw
+    dc_whole = pre_feeder.subreads_to_dc_example(
+        subreads=subreads,
+        ccs_seqname=zmw,
+        dc_config=dc_config,
+        window_widths=window_widths,
+    )
+    feature_dicts = [x.to_features_dict() for x in dc_whole.iter_examples()]
+    return feature_dicts, dc_whole.counter
+
+
+def process_skipped_window(
+    feature_dict: Dict[str, Any], options: InferenceOptions
+) -> stitch_utils.DCModelOutput:
+    """CCS passthrough for skipped windows (quick_inference.py:567-594)."""
+    rows = feature_dict["subreads"]
+    (_, _, _, _, ccs_index, _, _) = cfg.get_indices(
+        options.max_passes, options.use_ccs_bq
+    )
+    ccs = rows[ccs_index[0], :, 0]
+    ccs_seq = phred.encoded_sequence_to_string(ccs)
+    ccs_quality_scores = np.asarray(
+        feature_dict["ccs_base_quality_scores"], dtype=np.float64
+    )
+    if options.ccs_calibration_values.enabled:
+        ccs_quality_scores = calibration_lib.calibrate_quality_scores(
+            ccs_quality_scores, options.ccs_calibration_values
+        )
+    ccs_quality_scores = np.minimum(
+        ccs_quality_scores, options.max_base_quality
+    )
+    ccs_quality_scores = np.maximum(ccs_quality_scores, 0)
+    ccs_quality_scores = ccs_quality_scores.astype(np.int32)
+    return stitch_utils.DCModelOutput(
+        window_pos=feature_dict["window_pos"],
+        molecule_name=feature_dict["name"],
+        sequence=ccs_seq,
+        quality_string=phred.quality_scores_to_string(ccs_quality_scores),
+        ec=feature_dict["ec"],
+        np_num_passes=feature_dict["np_num_passes"],
+        rq=feature_dict["rq"],
+        rg=feature_dict["rg"],
+    )
+
+
+def run_model_on_examples(
+    feature_dicts: List[Dict[str, Any]],
+    runner: InferenceRunner,
+    options: InferenceOptions,
+) -> List[stitch_utils.DCModelOutput]:
+    """Batched model execution emitting per-window DCModelOutput."""
+    predictions: List[stitch_utils.DCModelOutput] = []
+    params = runner.params
+    for i in range(0, len(feature_dicts), options.batch_size):
+        chunk = feature_dicts[i : i + options.batch_size]
+        rows = np.stack(
+            [
+                data_lib.format_rows(np.asarray(f["subreads"]), params)[
+                    :, :, 0
+                ]
+                for f in chunk
+            ]
+        ).astype(np.float32)
+        bases_t, quals_t = runner.forward_windows(torch.from_numpy(rows))
+        bases = bases_t.cpu().numpy()
+        quals = quals_t.cpu().numpy()
+        for j, f in enumerate(chunk):
+            seq = phred.encoded_sequence_to_string(bases[j])
+            qual = phred.quality_scores_to_string(quals[j])
+            predictions.append(
+                stitch_utils.DCModelOutput(
+                    window_pos=f["window_pos"],
+                    molecule_name=f["name"],
+                    sequence=seq,
+                    quality_string=qual,
+                    ec=f["ec"],
+                    np_num_passes=f["np_num_passes"],
+                    rq=f["rq"],
+                    rg=f["rg"],
+                )
+            )
+    return predictions
+
+
+class Timelog:
+    """Stage wall-clock CSV (quick_inference.py:278-299,777-783)."""
+
+    def __init__(self):
+        self.rows: List[Dict[str, Any]] = []
+
+    def add(self, stage, item, before, num_examples=None, num_subreads=None,
+            num_zmws=None):
+        self.rows.append(
+            dict(
+                item=item, stage=stage, runtime=time.time() - before,
+                num_zmws=num_zmws, num_examples=num_examples,
+                num_subreads=num_subreads,
+            )
+        )
+
+    def save(self, output_prefix: str):
+        cols = ["item", "stage", "runtime", "num_zmws", "num_examples",
+                "num_subreads"]
+        with open(f"{output_prefix}.csv", "w") as f:
+            f.write(",".join(cols) + "\n")
+            for r in self.rows:
+                f.write(",".join(str(r[c]) for c in cols) + "\n")
+
+
+def _write_outputs(
+    predictions: List[stitch_utils.DCModelOutput],
+    output_writer,
+    bam_out,
+    options: InferenceOptions,
+    outcome_counter: stitch_utils.OutcomeCounter,
+):
+    """Sorts, stitches and writes per-ZMW outputs
+    (quick_inference.py:718-760)."""
+    from deepconsensus_amd.dcio import bam as bam_lib
+
+    predictions = sorted(
+        predictions, key=lambda dc: (dc.molecule_name, dc.window_pos)
+    )
+    for zmw, preds in itertools.groupby(
+        predictions, lambda p: p.molecule_name
+    ):
+        preds = list(preds)
+        fastq_string = stitch_utils.stitch_to_fastq(
+            molecule_name=zmw,
+            predictions=preds,
+            max_length=options.max_length,
+            min_quality=options.min_quality,
+            min_length=options.min_length,
+            outcome_counter=outcome_counter,
+        )
+        if not fastq_string:
+            continue
+        if bam_out is None:
+            output_writer.write(fastq_string)
+        else:
+            name, seq, _, qual = fastq_string.splitlines()
+            name = name[1:]
+            record = bam_lib.BamRead(
+                qname=name,
+                flag=4,
+                ref_id=-1,
+                pos=-1,
+                mapq=255,
+                seq=seq,
+                query_qualities=phred.quality_string_to_array(qual),
+                tags={
+                    "ec": preds[0].ec if preds[0].ec is not None else -1.0,
+                    "np": preds[0].np_num_passes or 0,
+                    "rq": preds[0].rq if preds[0].rq is not None else -1.0,
+                    "RG": preds[0].rg or "",
+                    "zm": int(name.split("/")[1]),
+                },
+            )
+            bam_out.write(record)
+
+
+def run(
+    subreads_to_ccs: str,
+    ccs_bam: str,
+    checkpoint: str,
+    output: str,
+    options: Optional[InferenceOptions] = None,
+    limit: int = 0,
+    dc_calibration: Optional[str] = None,
+    ccs_calibration: Optional[str] = None,
+    device: Optional[str] = None,
+) -> stitch_utils.OutcomeCounter:
+    """Performs an inference run (quick_inference.py:794-963)."""
+    t_start = time.time()
+    options = options or InferenceOptions()
+
+    # Load params + model. checkpoint == 'random' builds a random-init model
+    # (testing only); otherwise a directory with params.json + weights.
+    if checkpoint == "random":
+        params = cfg.get_config("transformer_learn_values+custom")
+        params.max_passes = options.max_passes
+        cfg.modify_params(
+            params, max_length=options.max_length, is_training=False
+        )
+        model = get_model(params)
+    else:
+        params = ckpt_lib.load_params(checkpoint)
+        cfg.modify_params(
+            params, max_length=options.max_length, is_training=False
+        )
+        model = get_model(params)
+        ckpt_lib.load_checkpoint(checkpoint, model)
+        options.max_passes = params.max_passes
+        options.use_ccs_bq = bool(params.get("use_ccs_bq", False))
+    options.example_height = cfg.get_total_rows(
+        options.max_passes, options.use_ccs_bq
+    )
+
+    # Calibration: explicit flag > params.json dc_calibration > skip.
+    if dc_calibration is None:
+        dc_calibration = params.get("dc_calibration", "skip")
+    options.dc_calibration_values = calibration_lib.parse_calibration_string(
+        dc_calibration
+    )
+    if ccs_calibration is not None:
+        options.ccs_calibration_values = (
+            calibration_lib.parse_calibration_string(ccs_calibration)
+        )
+
+    calib_str = dc_calibration if dc_calibration else "skip"
+    runner = InferenceRunner(
+        params, model, device=device, calibration=calib_str,
+        max_qual=options.max_base_quality,
+    )
+    log.info("model on %s (native kernels: %s)", runner.device, runner.native)
+
+    # Output writer.
+    bam_out = None
+    if output.endswith(".bam"):
+        from deepconsensus_amd.dcio import bam as bam_lib
+
+        header = bam_lib.BamHeader(
+            text="@HD\tVN:1.6\tSO:unknown", references=[]
+        )
+        bam_out = bam_lib.BamWriter(output, header)
+        output_writer = None
+    else:
+        output_writer = open(output, "w")
+
+    outcome_counter = stitch_utils.OutcomeCounter()
+    stats_counter = collections.Counter()
+    timelog = Timelog()
+    output_prefix = os.path.splitext(output)[0]
+
+    dc_config = DcConfig(
+        options.max_passes, options.max_length, options.use_ccs_bq
+    )
+    proc_feeder, main_counter = pre_feeder.create_proc_feeder(
+        subreads_to_ccs=subreads_to_ccs,
+        ccs_bam=ccs_bam,
+        dc_config=dc_config,
+        ins_trim=options.ins_trim,
+        use_ccs_smart_windows=options.use_ccs_smart_windows,
+        limit=limit,
+    )
+
+    pool = None
+    if options.cpus > 0:
+        pool = concurrent.futures.ProcessPoolExecutor(options.cpus)
+
+    def zmw_batches():
+        batch = []
+        for input_data in proc_feeder():
+            subreads, zmw, dcc, split, window_widths = input_data
+            batch.append((zmw, subreads, dcc, window_widths))
+            if len(batch) >= options.batch_zmws:
+                yield batch
+                batch = []
+        if batch:
+            yield batch
+
+    def preprocess_batch(inputs):
+        if pool is not None:
+            return list(pool.map(preprocess_one_zmw, inputs))
+        return [preprocess_one_zmw(z) for z in inputs]
+
+    def infer_batch(inputs, outputs, batch_name):
+        before = time.time()
+        feature_dicts_for_zmws, counters = (
+            zip(*outputs) if outputs else ([], [])
+        )
+        for counter in counters:
+            stats_counter.update(counter)
+        n_examples = sum(len(z) for z in feature_dicts_for_zmws)
+        n_subreads = sum(len(s) for _, s, _, _ in inputs)
+        timelog.add("preprocess", batch_name, before, n_examples,
+                    n_subreads, len(inputs))
+
+        before = time.time()
+        for_model: List[Dict[str, Any]] = []
+        skipped: List[stitch_utils.DCModelOutput] = []
+        for one_zmw in feature_dicts_for_zmws:
+            for window in one_zmw:
+                skip = False
+                if window["overflow"]:
+                    skipped.append(process_skipped_window(window, options))
+                    skip = True
+                if options.skip_windows_above and not skip:
+                    avg_q = phred.avg_phred(
+                        window["ccs_base_quality_scores"]
+                    )
+                    if avg_q > options.skip_windows_above:
+                        skipped.append(
+                            process_skipped_window(window, options)
+                        )
+                        skip = True
+                if not skip:
+                    for_model.append(window)
+        preds = run_model_on_examples(for_model, runner, options)
+        preds.extend(skipped)
+        timelog.add("run_model", batch_name, before, n_examples,
+                    n_subreads, len(inputs))
+
+        before = time.time()
+        _write_outputs(preds, output_writer, bam_out, options,
+                       outcome_counter)
+        timelog.add("stitch_and_write_fastq", batch_name, before,
+                    n_examples, n_subreads, len(inputs))
+
+    # Pipelined loop: preprocess batch N+1 while batch N runs the model.
+    batch_iter = zmw_batches()
+    pending: Optional[Tuple[Sequence, Any]] = None
+    n_batches = 0
+    with concurrent.futures.ThreadPoolExecutor(1) as prefetcher:
+        for inputs in batch_iter:
+            fut = prefetcher.submit(preprocess_batch, inputs)
+            if pending is not None:
+                prev_inputs, prev_fut = pending
+                infer_batch(prev_inputs, prev_fut.result(),
+                            f"batch {n_batches}")
+                n_batches += 1
+            pending = (inputs, fut)
+        if pending is not None:
+            prev_inputs, prev_fut = pending
+            infer_batch(prev_inputs, prev_fut.result(), f"batch {n_batches}")
+            n_batches += 1
+
+    if pool is not None:
+        pool.shutdown()
+    if bam_out is not None:
+        bam_out.close()
+    if output_writer is not None:
+        output_writer.close()
+
+    stats_counter.update(main_counter)
+    timelog.save(f"{output_prefix}.runtime")
+    with open(f"{output_prefix}.inference.json", "w") as f:
+        json.dump(dict(stats_counter), f, indent=True)
+
+    log.info(
+        "Processed %d ZMWs in %.3f seconds",
+        main_counter["n_zmw_processed"], time.time() - t_start,
+    )
+    log.info("Outcome counts: %s", outcome_counter)
+    return outcome_counter

@@ -1,0 +1,297 @@
+"""I/O round-trips + end-to-end pipeline tests (`run` and `preprocess` on
+synthetic BAMs), mirroring reference preprocess_test.py's E2E style with
+in-repo fixtures."""
+import json
+import os
+
+import numpy as np
+import pytest
+
+from deepconsensus_amd.dcio import bam as bam_lib
+from deepconsensus_amd.dcio import example_codec, tfrecord
+from deepconsensus_amd.postprocess import stitch
+from deepconsensus_amd.utils import phred
+
+
+def _random_seq(rng, n):
+    return "".join(rng.choice(list("ATCG"), size=n))
+
+
+def make_test_bams(tmp_path, n_zmws=3, length=220, n_subreads=4, seed=7):
+    """Writes synthetic subreads_to_ccs.bam + ccs.bam; returns paths."""
+    rng = np.random.default_rng(seed)
+    refs = []
+    zmw_seqs = {}
+    for z in range(n_zmws):
+        name = f"m000/{z + 10}/ccs"
+        seq = _random_seq(rng, length)
+        refs.append((name, length))
+        zmw_seqs[name] = seq
+    header = bam_lib.BamHeader(text="@HD\tVN:1.6", references=refs)
+
+    sub_path = str(tmp_path / "subreads_to_ccs.bam")
+    with bam_lib.BamWriter(sub_path, header) as w:
+        for rid, (name, ln) in enumerate(refs):
+            zm = int(name.split("/")[1])
+            seq = zmw_seqs[name]
+            for i in range(n_subreads):
+                # Introduce a small mutation region per subread.
+                s = list(seq)
+                p = int(rng.integers(0, ln - 1))
+                s[p] = rng.choice(list("ATCG"))
+                read = bam_lib.BamRead(
+                    qname=f"m000/{zm}/{i * 100}_{i * 100 + ln}",
+                    flag=16 if i % 2 else 0,
+                    ref_id=rid,
+                    pos=0,
+                    mapq=60,
+                    cigartuples=[(0, ln)],
+                    seq="".join(s),
+                    query_qualities=[30] * ln,
+                    tags={
+                        "zm": zm,
+                        "pw": rng.integers(0, 60, ln).astype(np.uint8),
+                        "ip": rng.integers(0, 60, ln).astype(np.uint8),
+                        "sn": np.array([6.0, 7.0, 5.5, 9.1],
+                                       dtype=np.float32),
+                    },
+                )
+                w.write(read)
+
+    ccs_path = str(tmp_path / "ccs.bam")
+    with bam_lib.BamWriter(ccs_path, header) as w:
+        for rid, (name, ln) in enumerate(refs):
+            zm = int(name.split("/")[1])
+            read = bam_lib.BamRead(
+                qname=name,
+                flag=4,
+                ref_id=-1,
+                pos=-1,
+                cigartuples=[],
+                seq=zmw_seqs[name],
+                query_qualities=rng.integers(20, 40, ln),
+                tags={"zm": zm, "ec": 11.5, "np": n_subreads, "rq": 0.998,
+                      "RG": "rg0"},
+            )
+            w.write(read)
+    return sub_path, ccs_path
+
+
+def test_bam_round_trip(tmp_path):
+    sub, ccs = make_test_bams(tmp_path)
+    reads = list(bam_lib.BamReader(sub))
+    assert len(reads) == 12
+    r = reads[0]
+    assert r.qname.startswith("m000/10/")
+    assert r.cigartuples == [(0, 220)]
+    assert len(r.seq) == 220
+    assert r.get_tag("zm") == 10
+    assert r.get_tag("sn").shape == (4,)
+    assert r.reference_name == "m000/10/ccs"
+    assert not r.is_unmapped
+    reads2 = list(bam_lib.BamReader(ccs))
+    assert reads2[0].is_unmapped
+    assert abs(reads2[0].get_tag("rq") - 0.998) < 1e-6
+    # fetch_index groups by reference name.
+    idx = bam_lib.fetch_index(sub)
+    assert len(idx) == 3
+    assert len(idx["m000/10/ccs"]) == 4
+
+
+def test_example_codec_round_trip():
+    feats = {
+        "subreads/encoded": (example_codec.BYTES, [b"\x00\x01\x02\x03"]),
+        "subreads/shape": (example_codec.INT64, [85, 100, 1]),
+        "name": (example_codec.BYTES, [b"m0/1/ccs"]),
+        "floats": (example_codec.FLOAT, [1.5, -2.25]),
+    }
+    enc = example_codec.encode_example(feats)
+    dec = example_codec.decode_example(enc)
+    assert dec["subreads/shape"] == (example_codec.INT64, [85, 100, 1])
+    assert dec["name"][1][0] == b"m0/1/ccs"
+    assert dec["floats"][0] == example_codec.FLOAT
+    np.testing.assert_allclose(dec["floats"][1], [1.5, -2.25])
+
+
+def test_tfrecord_round_trip(tmp_path):
+    path = str(tmp_path / "x.tfrecord.gz")
+    records = [b"alpha", b"beta" * 100, b""]
+    with tfrecord.TFRecordWriter(path, compression="gzip") as w:
+        for r in records:
+            w.write(r)
+    got = list(tfrecord.read_tfrecords(path, check_crc=True))
+    assert got == records
+
+
+def test_stitch_basic():
+    outs = [
+        stitch.DCModelOutput(
+            molecule_name="m/1/ccs", window_pos=0,
+            sequence="AAAA TT", quality_string="IIIIIII",
+        ),
+        stitch.DCModelOutput(
+            molecule_name="m/1/ccs", window_pos=7,
+            sequence="GGGG", quality_string="IIII",
+        ),
+    ]
+    counter = stitch.OutcomeCounter()
+    fq = stitch.stitch_to_fastq("m/1/ccs", outs, max_length=7,
+                                min_quality=10, min_length=1,
+                                outcome_counter=counter)
+    lines = fq.splitlines()
+    assert lines[0] == "@m/1/ccs"
+    assert lines[1] == "AAAATTGGGG"  # gap removed
+    assert len(lines[3]) == 10
+    assert counter.success == 1
+
+
+def test_stitch_missing_window_discards():
+    outs = [
+        stitch.DCModelOutput(
+            molecule_name="m/1/ccs", window_pos=100,
+            sequence="GGGG", quality_string="IIII",
+        ),
+    ]
+    counter = stitch.OutcomeCounter()
+    fq = stitch.stitch_to_fastq("m/1/ccs", outs, max_length=100,
+                                min_quality=10, min_length=1,
+                                outcome_counter=counter)
+    assert fq is None
+    assert counter.empty_sequence == 1
+
+
+def test_stitch_quality_rounding():
+    # all-Q10 read passes a min_quality=10 filter (rounding caveat,
+    # stitch_utils.py:101-109).
+    q10 = phred.quality_scores_to_string(np.full(50, 10))
+    outs = [stitch.DCModelOutput(molecule_name="m", window_pos=0,
+                                 sequence="A" * 50, quality_string=q10)]
+    counter = stitch.OutcomeCounter()
+    fq = stitch.stitch_to_fastq("m", outs, max_length=50, min_quality=10,
+                                min_length=1, outcome_counter=counter)
+    assert fq is not None
+
+
+def test_quick_inference_e2e_fastq(tmp_path):
+    """Full `run` on synthetic BAMs with a random-init model (CPU)."""
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=3, length=220)
+    out = str(tmp_path / "out.fastq")
+    options = qi.InferenceOptions(
+        batch_size=16, batch_zmws=2, cpus=0, min_quality=0,
+        skip_windows_above=0,
+    )
+    counter = qi.run(
+        subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+        output=out, options=options, device="cpu",
+    )
+    assert counter.total == 3
+    # Runtime CSV + stats JSON written.
+    assert os.path.exists(str(tmp_path / "out.runtime.csv"))
+    stats = json.load(open(tmp_path / "out.inference.json"))
+    assert stats["n_zmw_processed"] == 3
+    # Every ZMW produced output (min_quality=0 disables filtering).
+    from deepconsensus_amd.dcio.fastq import read_fastq
+
+    recs = list(read_fastq(out))
+    assert len(recs) == 3
+    for r in recs:
+        assert len(r.sequence) == len(r.quality)
+        assert len(r.sequence) > 0
+
+
+def test_quick_inference_skip_windows(tmp_path):
+    """skip_windows_above adopts CCS bases for high-quality windows."""
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=2, length=150, seed=3)
+    out = str(tmp_path / "out.fastq")
+    options = qi.InferenceOptions(
+        batch_size=16, batch_zmws=2, cpus=0, min_quality=0,
+        skip_windows_above=20,  # CCS quals are 20-40 -> most windows skip
+    )
+    counter = qi.run(
+        subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+        output=out, options=options, device="cpu",
+    )
+    assert counter.total == 2
+    stats = json.load(open(tmp_path / "out.inference.json"))
+    assert stats["n_zmw_processed"] == 2
+
+
+def test_quick_inference_bam_output(tmp_path):
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=2, length=120, seed=5)
+    out = str(tmp_path / "out.bam")
+    options = qi.InferenceOptions(batch_size=8, batch_zmws=2, cpus=0,
+                                  min_quality=0, skip_windows_above=0)
+    qi.run(subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+           output=out, options=options, device="cpu")
+    reads = list(bam_lib.BamReader(out))
+    assert len(reads) == 2
+    r = reads[0]
+    assert r.flag == 4
+    assert r.has_tag("zm") and r.has_tag("rq") and r.has_tag("RG")
+    assert len(r.seq) > 0
+
+
+def test_preprocess_cli_e2e(tmp_path):
+    """`preprocess` inference mode: serial + parallel produce same count."""
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models import data as data_lib
+    from deepconsensus_amd.preprocess import preprocess_cli
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=3, length=220)
+    out0 = str(tmp_path / "serial" / "ex.tfrecord.gz")
+    preprocess_cli.main([
+        "--subreads_to_ccs", sub, "--ccs_bam", ccs,
+        "--output", out0, "--cpus", "0",
+    ])
+    recs = list(tfrecord.read_tfrecords(out0))
+    assert len(recs) == 9  # 3 ZMWs x ceil(220/100) windows
+    summary = json.load(
+        open(str(tmp_path / "serial" / "ex.inference.json"))
+    )
+    assert summary["n_zmw_processed"] == 3
+    assert summary["max_passes"] == "20"
+
+    out2 = str(tmp_path / "par" / "ex.tfrecord.gz")
+    preprocess_cli.main([
+        "--subreads_to_ccs", sub, "--ccs_bam", ccs,
+        "--output", out2, "--cpus", "2",
+    ])
+    recs2 = list(tfrecord.read_tfrecords(out2))
+    assert len(recs2) == 9
+
+    # Records parse through the data layer.
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=False)
+    ex = data_lib.process_input(recs[0], params, inference=True)
+    assert ex["rows"].shape == (85, 100, 1)
+    # Window positions monotonic within one ZMW.
+    by_name = {}
+    for r in recs:
+        d = data_lib.process_input(r, params, inference=True)
+        by_name.setdefault(d["name"], []).append(int(d["window_pos"]))
+    for name, poss in by_name.items():
+        assert poss == sorted(poss)
+
+
+def test_filter_reads_fastq(tmp_path):
+    from deepconsensus_amd.calibration import filter_reads as fr
+    from deepconsensus_amd.dcio.fastq import FastqRecord, read_fastq, write_fastq
+
+    inp = str(tmp_path / "in.fastq")
+    write_fastq(inp, [
+        FastqRecord("good", "ACGT" * 10, "I" * 40),   # Q40
+        FastqRecord("bad", "ACGT" * 10, "+" * 40),    # Q10
+    ])
+    outp = str(tmp_path / "out.fastq")
+    fr.main(["-i", inp, "-o", outp, "-q", "20"])
+    recs = list(read_fastq(outp))
+    assert [r.name for r in recs] == ["good"]
+    # Threshold exactly at the quality keeps the read (rounding).
+    fr.main(["-i", inp, "-o", outp, "-q", "10"])
+    assert len(list(read_fastq(outp))) == 2
