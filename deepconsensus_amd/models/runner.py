@@ -201,6 +201,29 @@ class InferenceRunner:
                 lw["wout_t_a"] = lw["wout_t"] * lw["alpha_attn"]
                 lw["w2_t_a"] = lw["w2_t"] * lw["alpha_ffn"]
                 lw["b2_a"] = lw["b2"] * lw["alpha_ffn"]
+            # Padded weights for the fused FFN kernel (in-bounds 16B frags):
+            # W1 [2048, 288], W2 [320, 2048], b1 fp32, b2 [320] fp32.
+            self.ffn_fused_ok = (
+                model.params["hidden_size"] == 280
+                and model.params["filter_size"] == 2048
+            )
+            if self.ffn_fused_ok:
+                for i, l in enumerate(model.layers):
+                    lw = self.layer_w[i]
+                    w1 = l.ffn.filter_layer.weight.detach().float()
+                    w1p = torch.zeros(2048, 288)
+                    w1p[:, :280] = w1
+                    lw["w1_pad"] = w1p.to(bf16).contiguous().to(dev)
+                    lw["b1_f32"] = (
+                        l.ffn.filter_layer.bias.detach().float().to(dev)
+                    )
+                    w2 = l.ffn.output_layer.weight.detach().float()
+                    w2p = torch.zeros(320, 2048)
+                    w2p[:280] = w2
+                    lw["w2_pad"] = w2p.to(bf16).contiguous().to(dev)
+                    b2p = torch.zeros(320)
+                    b2p[:280] = l.ffn.output_layer.bias.detach().float()
+                    lw["b2_f32"] = b2p.to(dev)
             # hipBLASLt fused bias+ReLU epilogue, when this torch exposes it.
             self._addmm_act = hasattr(torch, "_addmm_activation")
             if self._addmm_act:
@@ -250,11 +273,17 @@ class InferenceRunner:
                 a = self._attn(qkv)
                 # Residual fused into the GEMM epilogue (alpha pre-folded).
                 flat = torch.addmm(flat, a.view(b * l, h), lw["wout_t_a"])
-                if self._addmm_act:
+                if self.ffn_fused_ok:
+                    flat = self.ext.fused_ffn(
+                        flat, lw["w1_pad"], lw["b1_f32"], lw["w2_pad"],
+                        lw["b2_f32"], lw["alpha_ffn"],
+                    )
+                elif self._addmm_act:
                     ff = torch._addmm_activation(lw["b1"], flat, lw["w1_t"])
+                    flat = torch.addmm(flat, ff, lw["w2_t_a"]).add_(lw["b2_a"])
                 else:
                     ff = torch.addmm(lw["b1"], flat, lw["w1_t"]).relu_()
-                flat = torch.addmm(flat, ff, lw["w2_t_a"]).add_(lw["b2_a"])
+                    flat = torch.addmm(flat, ff, lw["w2_t_a"]).add_(lw["b2_a"])
             x = flat.view(b, l, h)
         else:
             for layer in self.layers_bf16:

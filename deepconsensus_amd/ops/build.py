@@ -16,6 +16,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "banded_attn.hip"),
     os.path.join(_OPS_DIR, "hip", "banded_attn_mfma.hip"),
     os.path.join(_OPS_DIR, "hip", "embed_gather.hip"),
+    os.path.join(_OPS_DIR, "hip", "fused_ffn.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

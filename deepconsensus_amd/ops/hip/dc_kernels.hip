@@ -176,6 +176,8 @@ at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
 at::Tensor embed_gather(at::Tensor rows, at::Tensor table_flat,
                         at::Tensor row_shift, at::Tensor row_vocab,
                         at::Tensor chunk_cnt, at::Tensor chunk_entries);
+at::Tensor fused_ffn(at::Tensor x, at::Tensor w1, at::Tensor b1,
+                     at::Tensor w2, at::Tensor b2, double alpha);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
@@ -186,4 +188,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Banded MHA forward on MFMA (D=140, L<=104)");
   m.def("embed_gather", &embed_gather,
         "Subread-stack embedding gather (K2)");
+  m.def("fused_ffn", &fused_ffn,
+        "Fused FFN + ReZero residual (K8+K9), hidden tensor LDS-resident");
 }

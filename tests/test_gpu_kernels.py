@@ -215,3 +215,30 @@ def test_banded_attn_mfma_matches_torch(setup):
         err = (out - ref).abs()
         assert err.max().item() < 0.06, (L, win, err.max().item())
         assert err.mean().item() < 0.006, (L, win, err.mean().item())
+
+
+def test_fused_ffn_matches_torch(setup):
+    """Fused FFN kernel vs fp32 torch reference (includes relu + residual)."""
+    params, model, runner, rows = setup
+    assert runner.ffn_fused_ok
+    torch.manual_seed(21)
+    for M in (1600, 4096, 129):  # non-multiples of the 128-row tile too
+        x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+        lw = runner.layer_w[0]
+        out = runner.ext.fused_ffn(
+            x, lw["w1_pad"], lw["b1_f32"], lw["w2_pad"], lw["b2_f32"], 0.7,
+        ).float()
+        l = model.layers[0]
+        xf = x.float()
+        ref = xf + 0.7 * (
+            torch.relu(
+                xf @ l.ffn.filter_layer.weight.t().cuda().float()
+                + l.ffn.filter_layer.bias.cuda().float()
+            )
+            @ l.ffn.output_layer.weight.t().cuda().float()
+            + l.ffn.output_layer.bias.cuda().float()
+        )
+        err = (out - ref).abs()
+        scale = ref.abs().mean().clamp_min(1e-3)
+        assert (err.mean() / scale).item() < 0.02, (M, err.mean().item())
+        assert err.max().item() < 0.3, (M, err.max().item())
