@@ -5,21 +5,21 @@
 // activation never touches HBM (the eager path writes + re-reads ~3.3 GB of
 // it per layer at batch 4096, plus a full ReLU pass).
 //
-// Structure (512 threads = 8 waves; one workgroup = 128 rows):
-//  * x tile staged once through LDS, then each wave holds its 18 A-fragments
-//    (32 rows x 288 k) in registers; the x LDS region is then REUSED as the
-//    weight staging buffer — full-cacheline coalesced staging, never
-//    fragment-shaped global reads (cdna_hip_programming.md section 5,
-//    "x through LDS in full lines");
-//  * 16 chunks of 128 hidden columns: stage W1 slice [128][288] -> barrier
-//    -> h_chunk = relu(x @ W1^T + b1) via v_mfma_f32_32x32x16_bf16, written
-//    bf16 row-major to the h region -> stage W2 slice [280][128] over the
-//    same weight region -> barrier -> out accumulators (5 col-tiles/wave,
-//    fp32) consume h_chunk;
-//  * epilogue folds b2, the ReZero alpha and the residual (x re-read from
-//    global).
+// Structure (512 threads = 8 waves; one workgroup = 128 rows; 32 chunks of
+// 64 hidden columns):
+//  * x staged once through LDS, then each wave's 18 A-fragments (32 rows x
+//    288 k) live in registers; the x image region is re-carved into W1/W2
+//    staging buffers;
+//  * weight staging is the T14 async-STAGE split
+//    (cdna_hip_programming.md G15): each phase ISSUES the next slice's
+//    global loads into ~5 uint4 registers before its MFMA cluster and
+//    ds_writes them after it, so HBM/L2 latency hides under the matrix
+//    work; all LDS images are padded to conflict-free b128 strides;
+//  * B1: h_chunk = relu(x @ W1^T + b1) -> bf16 row-major LDS;
+//    B2: fp32 out accumulators (5 col-tiles/wave) consume h_chunk vs W2;
+//  * epilogue folds b2, the ReZero alpha and the residual.
 // Weights arrive host-padded (W1 [2048, 288], W2 [320, 2048]) so every 16-B
-// fragment read is in bounds.
+// read is in bounds.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -36,103 +36,139 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 constexpr int BM = 128;        // rows per workgroup
 constexpr int K1 = 280;        // model width
 constexpr int K1P = 288;       // padded k (W1 second dim)
-constexpr int A_STRIDE = 312;  // region-A row stride: 156 dwords % 64 = 28,
-                               // gcd(28,64)=4 -> conflict-free b128 groups
-constexpr int NC = 128;        // hidden cols per chunk
+constexpr int NC = 64;         // hidden cols per chunk
 constexpr int NHID = 2048;
 constexpr int NCHUNK = NHID / NC;
-constexpr int H_STRIDE = 136;  // h LDS row stride
-constexpr int W2_ROWS = 280;   // W2 rows staged (out cols)
-constexpr int NOUT_PAD = 320;  // padded W2 first dim (host tensor)
+constexpr int W1_STRIDE = 296;  // 148 dw % 64 = 20 -> conflict-free groups
+constexpr int W2_STRIDE = 72;   // 36 dw % 64 = 36 -> conflict-free groups
+constexpr int H_STRIDE = 72;
+constexpr int W2_ROWS = 280;
+constexpr int NOUT_PAD = 320;
+
+constexpr int W1_GRAN = NC * (K1P / 8);       // 16B granules per W1 slice
+constexpr int W2_GRAN = W2_ROWS * (NC / 8);   // granules per W2 slice
+constexpr int G_PER_T = 5;                    // ceil(granules / 512)
 
 __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w1,
     const float* __restrict__ b1, const bf16* __restrict__ w2,
     const float* __restrict__ b2, bf16* __restrict__ out,
     int M, float alpha) {
-  // Region A (77.8 KB): x image, then per-chunk W1 [128][A_STRIDE] and
-  // W2 [280][H_STRIDE] images. Region H (34.8 KB): hidden chunk.
-  __shared__ __attribute__((aligned(16))) bf16 smem_a[BM * A_STRIDE];
+  // W1 buf 37.9 KB + W2 buf 40.3 KB (the x image overlays both at start),
+  // h 18.4 KB.
+  __shared__ __attribute__((aligned(16))) bf16 smem_w[64 * W1_STRIDE
+                                                      + NOUT_PAD * W2_STRIDE];
   __shared__ __attribute__((aligned(16))) bf16 h_lds[BM][H_STRIDE];
+  bf16* w1_lds = smem_w;                       // [64][W1_STRIDE]
+  bf16* w2_lds = smem_w + 64 * W1_STRIDE;      // [<=320][W2_STRIDE]
 
   const int tid = threadIdx.x;
-  const int wave = tid >> 6;      // 0..7
+  const int wave = tid >> 6;
   const int lane = tid & 63;
   const int c = lane & 31;
   const int hi = lane >> 5;
   const int rg = wave >> 1;       // row group (rows 32*rg..+31)
-  const int ch = wave & 1;        // out col half (cols 140*ch..+139)
+  const int ch = wave & 1;        // col half
   const int m0 = blockIdx.x * BM;
 
-  // ---- Stage x tile into region A (zero pad cols/rows), coalesced. ----
-  for (int idx = tid; idx < BM * (A_STRIDE / 2); idx += 512) {
-    const int r = idx / (A_STRIDE / 2), d2 = idx % (A_STRIDE / 2);
+  // ---- Stage x image over smem_w ([128][296] rows), pull A-frags. ----
+  for (int idx = tid; idx < BM * (W1_STRIDE / 2); idx += 512) {
+    const int r = idx / (W1_STRIDE / 2), d2 = idx % (W1_STRIDE / 2);
     unsigned v = 0;
     if (m0 + r < M && 2 * d2 + 1 < K1) {
       v = *reinterpret_cast<const unsigned*>(
           x + (size_t)(m0 + r) * K1 + 2 * d2);
     }
-    *reinterpret_cast<unsigned*>(&smem_a[r * A_STRIDE + 2 * d2]) = v;
+    *reinterpret_cast<unsigned*>(&smem_w[r * W1_STRIDE + 2 * d2]) = v;
   }
   __syncthreads();
-
-  // ---- Pull this wave's 18 x A-fragments into registers. ----
   bf16x8 af[18];
 #pragma unroll
   for (int s = 0; s < 18; ++s) {
     af[s] = *reinterpret_cast<const bf16x8*>(
-        &smem_a[(32 * rg + c) * A_STRIDE + 16 * s + 8 * hi]);
+        &smem_w[(32 * rg + c) * W1_STRIDE + 16 * s + 8 * hi]);
   }
-  __syncthreads();  // region A free for weight staging
+  __syncthreads();
+
+  // ---- T14 staging: issue loads to regs early, ds_write after MFMAs.
+  // One shared register set: the W1 and W2 in-flight windows never overlap
+  // (issue_w2..write_w2 inside B1; issue_w1..write_w1 inside B2). ----
+  uint4 stage_regs[G_PER_T];
+
+  auto issue_w1 = [&](int chunk) {
+#pragma unroll
+    for (int i = 0; i < G_PER_T; ++i) {
+      const int g = min(tid + i * 512, W1_GRAN - 1);
+      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
+      stage_regs[i] = *reinterpret_cast<const uint4*>(
+          w1 + (size_t)(chunk * NC + row) * K1P + 8 * k8);
+    }
+  };
+  auto write_w1 = [&]() {
+#pragma unroll
+    for (int i = 0; i < G_PER_T; ++i) {
+      const int g = min(tid + i * 512, W1_GRAN - 1);
+      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
+      *reinterpret_cast<uint4*>(&w1_lds[row * W1_STRIDE + 8 * k8]) =
+          stage_regs[i];
+    }
+  };
+  auto issue_w2 = [&](int chunk) {
+#pragma unroll
+    for (int i = 0; i < G_PER_T; ++i) {
+      const int g = min(tid + i * 512, W2_GRAN - 1);
+      const int row = g / (NC / 8), k8 = g % (NC / 8);
+      stage_regs[i] = *reinterpret_cast<const uint4*>(
+          w2 + (size_t)row * NHID + chunk * NC + 8 * k8);
+    }
+  };
+  auto write_w2 = [&]() {
+#pragma unroll
+    for (int i = 0; i < G_PER_T; ++i) {
+      const int g = min(tid + i * 512, W2_GRAN - 1);
+      const int row = g / (NC / 8), k8 = g % (NC / 8);
+      *reinterpret_cast<uint4*>(&w2_lds[row * W2_STRIDE + 8 * k8]) =
+          stage_regs[i];
+    }
+  };
+
+  // Prologue: W1_0 staged synchronously.
+  issue_w1(0);
+  write_w1();
+  __syncthreads();
 
   f32x16 oacc[5] = {};
 
   for (int chunk = 0; chunk < NCHUNK; ++chunk) {
-    const int n0 = chunk * NC;
-    // ---- Stage W1 slice [NC rows n0..][K1P] -> region A, coalesced. ----
-    for (int g = tid; g < NC * (K1P / 8); g += 512) {
-      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
-      *reinterpret_cast<uint4*>(&smem_a[row * A_STRIDE + 8 * k8]) =
-          *reinterpret_cast<const uint4*>(
-              w1 + (size_t)(n0 + row) * K1P + 8 * k8);
-    }
-    __syncthreads();
-
-    // ---- B1: h_chunk = relu(x @ W1^T + b1). Wave: rows 32*rg..+31,
-    // hidden cols 64*ch + {0..63} (two 32-col tiles). ----
-#pragma unroll 1
-    for (int t = 0; t < 2; ++t) {
-      const int colt = 64 * ch + 32 * t;  // within chunk
-      f32x16 acc = {};
+    // ---- B1 phase: issue W2_c early; MFMAs; write h + W2_c. ----
+    issue_w2(chunk);
+    const int colt = 32 * ch;             // wave's hidden col tile
+    const int hcol = chunk * NC + colt + c;
+    f32x16 acc = {};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int s = 0; s < 18; ++s) {
-        const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
-            &smem_a[(colt + c) * A_STRIDE + 16 * s + 8 * hi]);
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], bfr, acc,
-                                                      0, 0, 0);
-      }
-      const float bias = b1[n0 + colt + c];
+    for (int s = 0; s < 18; ++s) {
+      const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+          &w1_lds[(colt + c) * W1_STRIDE + 16 * s + 8 * hi]);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], bfr, acc,
+                                                    0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    const float bias = b1[hcol];
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int row = 32 * rg + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float v = acc[r] + bias;
-        v = v > 0.f ? v : 0.f;
-        h_lds[row][colt + c] = __float2bfloat16(v);
-      }
+    for (int r = 0; r < 16; ++r) {
+      const int row = 32 * rg + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      float v = acc[r] + bias;
+      v = v > 0.f ? v : 0.f;
+      h_lds[row][colt + c] = __float2bfloat16(v);
     }
+    write_w2();
     __syncthreads();
 
-    // ---- Stage W2 slice [280 rows][NC k] over region A, coalesced. ----
-    for (int g = tid; g < W2_ROWS * (NC / 8); g += 512) {
-      const int row = g / (NC / 8), k8 = g % (NC / 8);
-      *reinterpret_cast<uint4*>(&smem_a[row * H_STRIDE + 8 * k8]) =
-          *reinterpret_cast<const uint4*>(
-              w2 + (size_t)row * NHID + n0 + 8 * k8);
-    }
-    __syncthreads();
-
-    // ---- B2: oacc += h_chunk @ W2^T (this wave's 140 out cols). ----
-#pragma unroll 1
+    // ---- B2 phase: issue W1_{c+1} early; MFMAs; write W1_{c+1}. ----
+    if (chunk + 1 < NCHUNK) issue_w1(chunk + 1);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
     for (int s = 0; s < NC / 16; ++s) {
       const bf16x8 a = *reinterpret_cast<const bf16x8*>(
           &h_lds[32 * rg + c][16 * s + 8 * hi]);
@@ -140,17 +176,18 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
       for (int ct = 0; ct < 5; ++ct) {
         const int ocol = min(140 * ch + 32 * ct + c, W2_ROWS - 1);
         const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
-            &smem_a[ocol * H_STRIDE + 16 * s + 8 * hi]);
-        oacc[ct] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bfr,
-                                                           oacc[ct], 0, 0, 0);
+            &w2_lds[ocol * W2_STRIDE + 16 * s + 8 * hi]);
+        oacc[ct] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a, bfr, oacc[ct], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
+    if (chunk + 1 < NCHUNK) write_w1();
     __syncthreads();
   }
 
-  // ---- Epilogue: out = x + alpha * (oacc + b2); x re-read from global.
-  // Fully unrolled: a runtime-indexed oacc[ct] would force the accumulators
-  // to scratch for the whole kernel (cdna_hip_programming.md rule 20). ----
+  // ---- Epilogue: out = x + alpha * (oacc + b2). Fully unrolled (a
+  // runtime-indexed oacc[ct] would force scratch, rule 20). ----
 #pragma unroll
   for (int ct = 0; ct < 5; ++ct) {
     const int col = 140 * ch + 32 * ct + c;
