@@ -179,6 +179,16 @@ class AlignmentLoss(torch.nn.Module):
         y_pred = self.preprocess_y_pred(y_pred)
         subs_costs = xentropy_subs_cost_fn(y_true_oh, y_pred)
         ins_costs = xentropy_ins_cost_fn(y_pred)
+        if subs_costs.is_cuda:
+            # HIP wavefront kernel with custom VJP (K13); fails loudly if the
+            # extension is missing on a GPU machine.
+            from deepconsensus_amd import ops as dc_ops
+
+            ext = dc_ops.get_ext(required=True)
+            return _AlignmentDP.apply(
+                subs_costs.float(), ins_costs.float(), seq_lens,
+                self.del_cost, self.loss_reg, self.width, ext,
+            )
         return self.alignment(
             subs_costs, ins_costs, self.del_cost, seq_lens
         )
@@ -190,6 +200,32 @@ class AlignmentLoss(torch.nn.Module):
         if self.reduction == "sum":
             return per_example.sum()
         return per_example.mean()
+
+
+class _AlignmentDP(torch.autograd.Function):
+    """HIP wavefront DP (ops/hip/alignment_dp.hip) with saved soft-min
+    weights driving the backward gather recursion."""
+
+    @staticmethod
+    def forward(ctx, subs, ins, seq_lens, del_cost, loss_reg, width, ext):
+        reg = 0.0 if loss_reg is None else float(loss_reg)
+        w = 0 if width is None else int(width)
+        loss, weights = ext.alignment_dp_fwd(
+            subs, ins, seq_lens.to(torch.int32), float(del_cost), reg, w
+        )
+        ctx.save_for_backward(weights, seq_lens.to(torch.int32))
+        ctx.dims = (subs.shape[1], subs.shape[2], w)
+        ctx.ext = ext
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        weights, seq_lens = ctx.saved_tensors
+        m, n, w = ctx.dims
+        grad_subs, grad_ins = ctx.ext.alignment_dp_bwd(
+            grad_out.contiguous().float(), weights, seq_lens, m, n, w
+        )
+        return grad_subs, grad_ins, None, None, None, None, None
 
 
 # ---------------------------------------------------------------------------

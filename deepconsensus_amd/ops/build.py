@@ -17,6 +17,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "banded_attn_mfma.hip"),
     os.path.join(_OPS_DIR, "hip", "embed_gather.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_ffn.hip"),
+    os.path.join(_OPS_DIR, "hip", "alignment_dp.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

@@ -1,0 +1,245 @@
+// Alignment-loss wavefront DP (K13) for gfx950: forward + custom VJP.
+//
+// The differentiable Needleman-Wunsch-style recursion of
+// losses_and_metrics.py:346-411 (soft-min via -reg*logsumexp(-t/reg)) /
+// :475-547 (banded variant, realized here as an in-band mask):
+//   V[i][j] = minop(V[i-1][j-1] + subs[i-1][j-1],
+//                   V[i][j-1]   + ins[j-1],
+//                   V[i-1][j]   + del_cost)
+// with V[0][j] cumulative insertion costs and V[i][0] = i*del_cost; the loss
+// is V[seq_len][j_end].  The reference backpropagates through ~200 sequential
+// tf graph steps; here one 128-thread workgroup per example runs the m+n-1
+// anti-diagonals in LDS, saving the three soft-min weights per cell, and the
+// backward kernel runs the reverse (gather-form) adjoint recursion:
+//   adj[i][j] = adj[i+1][j+1]*wm[i+1][j+1] + adj[i][j+1]*wi[i][j+1]
+//             + adj[i+1][j]*wd[i+1][j]
+//   d subs[i-1][j-1] = adj[i][j]*wm[i][j],  d ins[j-1] += adj[i][j]*wi[i][j].
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+namespace {
+
+constexpr int MAXW = 160;   // max (m+1): label window <= 159
+constexpr float DPINF = 1e9f;
+
+__device__ __forceinline__ void softmin3(
+    float cm, float ci, float cd, float reg,
+    float* v, float* wm, float* wi, float* wd) {
+  const float tmin = fminf(cm, fminf(ci, cd));
+  if (tmin >= 1e8f) {
+    *v = DPINF;
+    *wm = *wi = *wd = 0.f;
+    return;
+  }
+  if (reg <= 0.f) {  // hard min with argmin one-hot (ties: m, i, d order)
+    *v = tmin;
+    *wm = *wi = *wd = 0.f;
+    if (cm == tmin) *wm = 1.f;
+    else if (ci == tmin) *wi = 1.f;
+    else *wd = 1.f;
+    return;
+  }
+  const float em = __expf(-(cm - tmin) / reg);
+  const float ei = __expf(-(ci - tmin) / reg);
+  const float ed = __expf(-(cd - tmin) / reg);
+  const float denom = em + ei + ed;
+  *v = tmin - reg * __logf(denom);
+  *wm = em / denom;
+  *wi = ei / denom;
+  *wd = ed / denom;
+}
+
+// weights layout: [B][m+1][n+1][3]
+__global__ void alignment_fwd_kernel(
+    const float* __restrict__ subs,   // [B, m, n]
+    const float* __restrict__ ins,    // [B, n]
+    const int* __restrict__ seq_lens, // [B]
+    float* __restrict__ loss,         // [B]
+    float* __restrict__ weights,      // [B, m+1, n+1, 3]
+    int B, int m, int n, float del_cost, float reg, int width) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  __shared__ float vbuf[3][MAXW];   // rotating diagonals, indexed by i
+  float* vp2 = vbuf[0];
+  float* vp1 = vbuf[1];
+  float* vcur = vbuf[2];
+
+  const float* sb = subs + (size_t)b * m * n;
+  const float* ib = ins + (size_t)b * n;
+  float* wb = weights + (size_t)b * (m + 1) * (n + 1) * 3;
+  const int sl = seq_lens[b];
+  const int jend = (width > 0) ? min(n, sl + width) : n;
+  const int kend = sl + jend;
+
+  for (int i = tid; i < MAXW; i += blockDim.x) {
+    vp2[i] = DPINF;
+    vp1[i] = DPINF;
+    vcur[i] = DPINF;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    vp2[0] = 0.f;                       // V[0][0]
+    vp1[0] = ib[0];                     // V[0][1]
+    vp1[1] = del_cost;                  // V[1][0]
+    // weight entries for the k=1 cells
+    float* w01 = wb + (0 * (n + 1) + 1) * 3;
+    w01[0] = 0.f; w01[1] = 1.f; w01[2] = 0.f;
+    float* w10 = wb + (1 * (n + 1) + 0) * 3;
+    w10[0] = 0.f; w10[1] = 0.f; w10[2] = 1.f;
+  }
+  __syncthreads();
+  if (kend <= 1 && tid == 0) {
+    // Degenerate: empty prediction window.
+    loss[b] = (kend == 0) ? 0.f : ((sl == 0) ? ib[0] : del_cost);
+  }
+
+  for (int k = 2; k <= m + n; ++k) {
+    for (int i = tid; i <= min(m, k); i += blockDim.x) {
+      const int j = k - i;
+      float v = DPINF, wm = 0.f, wi = 0.f, wd = 0.f;
+      if (j >= 0 && j <= n &&
+          !(width > 0 && abs(j - i) > width)) {
+        if (i == 0) {
+          v = vp1[0] + ib[j - 1];
+          wi = 1.f;
+        } else if (j == 0) {
+          v = vp1[i - 1] + del_cost;
+          wd = 1.f;
+        } else {
+          const float cm = vp2[i - 1] + sb[(size_t)(i - 1) * n + (j - 1)];
+          const float ci = vp1[i] + ib[j - 1];
+          const float cd = vp1[i - 1] + del_cost;
+          softmin3(cm, ci, cd, reg, &v, &wm, &wi, &wd);
+        }
+        float* w = wb + ((size_t)i * (n + 1) + j) * 3;
+        w[0] = wm; w[1] = wi; w[2] = wd;
+      }
+      vcur[i] = v;
+      if (k == kend && i == sl) loss[b] = v;
+    }
+    __syncthreads();
+    // rotate: vp2 <- vp1 <- vcur <- (reused)
+    float* t = vp2;
+    vp2 = vp1;
+    vp1 = vcur;
+    vcur = t;
+    // vcur (old vp2) cells beyond this diagonal get overwritten next round;
+    // reset to inf the slots we may not touch.
+    for (int i = tid; i <= min(m, k + 1); i += blockDim.x) vcur[i] = DPINF;
+    __syncthreads();
+  }
+}
+
+__global__ void alignment_bwd_kernel(
+    const float* __restrict__ grad_out,  // [B]
+    const float* __restrict__ weights,   // [B, m+1, n+1, 3]
+    const int* __restrict__ seq_lens,    // [B]
+    float* __restrict__ grad_subs,       // [B, m, n] (pre-zeroed)
+    float* __restrict__ grad_ins,        // [B, n]    (pre-zeroed)
+    int B, int m, int n, int width) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  __shared__ float abuf[3][MAXW];   // adj diagonals k+2, k+1, k
+  __shared__ float gins[MAXW * 2];  // per-column ins-grad accumulator
+  float* ak2 = abuf[0];
+  float* ak1 = abuf[1];
+  float* acur = abuf[2];
+
+  const float* wb = weights + (size_t)b * (m + 1) * (n + 1) * 3;
+  float* gs = grad_subs + (size_t)b * m * n;
+  float* gi = grad_ins + (size_t)b * n;
+  const int sl = seq_lens[b];
+  const int jend = (width > 0) ? min(n, sl + width) : n;
+  const int kend = sl + jend;
+  const float seed = grad_out[b];
+
+  for (int i = tid; i < MAXW; i += blockDim.x) {
+    ak2[i] = 0.f;
+    ak1[i] = 0.f;
+    acur[i] = 0.f;
+  }
+  for (int j = tid; j < n; j += blockDim.x) gins[j] = 0.f;
+  __syncthreads();
+
+  for (int k = kend; k >= 0; --k) {
+    for (int i = tid; i <= min(m, k); i += blockDim.x) {
+      const int j = k - i;
+      float a = 0.f;
+      if (j >= 0 && j <= n) {
+        if (k == kend) {
+          a = (i == sl && j == jend) ? seed : 0.f;
+        } else {
+          // successors: (i+1, j+1) [wm, diag k+2], (i, j+1) [wi, k+1],
+          // (i+1, j) [wd, k+1]
+          if (i + 1 <= m && j + 1 <= n)
+            a += ak2[i + 1] *
+                 wb[((size_t)(i + 1) * (n + 1) + (j + 1)) * 3 + 0];
+          if (j + 1 <= n)
+            a += ak1[i] * wb[((size_t)i * (n + 1) + (j + 1)) * 3 + 1];
+          if (i + 1 <= m)
+            a += ak1[i + 1] * wb[((size_t)(i + 1) * (n + 1) + j) * 3 + 2];
+        }
+        if (a != 0.f) {
+          const float* w = wb + ((size_t)i * (n + 1) + j) * 3;
+          if (i > 0 && j > 0)
+            gs[(size_t)(i - 1) * n + (j - 1)] = a * w[0];
+          if (j > 0) {
+            // within a diagonal each j appears once: safe LDS accumulate.
+            gins[j - 1] += a * w[1];
+          }
+        }
+      }
+      acur[i] = a;
+    }
+    __syncthreads();
+    float* t = ak2;
+    ak2 = ak1;
+    ak1 = acur;
+    acur = t;
+    for (int i = tid; i < MAXW; i += blockDim.x) acur[i] = 0.f;
+    __syncthreads();
+  }
+  for (int j = tid; j < n; j += blockDim.x) gi[j] = gins[j];
+}
+
+}  // namespace
+
+std::vector<at::Tensor> alignment_dp_fwd(
+    at::Tensor subs, at::Tensor ins, at::Tensor seq_lens,
+    double del_cost, double reg, int64_t width) {
+  TORCH_CHECK(subs.is_cuda() && subs.dtype() == at::kFloat, "subs fp32 cuda");
+  auto sc = subs.contiguous();
+  auto ic = ins.contiguous();
+  auto lc = seq_lens.to(at::kInt).contiguous();
+  const int B = sc.size(0), m = sc.size(1), n = sc.size(2);
+  TORCH_CHECK(m + 1 < MAXW && n + 1 < MAXW, "window too long");
+  auto loss = at::zeros({B}, sc.options());
+  auto weights = at::zeros({B, m + 1, n + 1, 3}, sc.options());
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(alignment_fwd_kernel, dim3(B), dim3(128), 0, stream,
+                     sc.data_ptr<float>(), ic.data_ptr<float>(),
+                     lc.data_ptr<int>(), loss.data_ptr<float>(),
+                     weights.data_ptr<float>(), B, m, n,
+                     (float)del_cost, (float)reg, (int)width);
+  return {loss, weights};
+}
+
+std::vector<at::Tensor> alignment_dp_bwd(
+    at::Tensor grad_out, at::Tensor weights, at::Tensor seq_lens,
+    int64_t m, int64_t n, int64_t width) {
+  auto gc = grad_out.contiguous();
+  auto wc = weights.contiguous();
+  auto lc = seq_lens.to(at::kInt).contiguous();
+  const int B = wc.size(0);
+  auto grad_subs = at::zeros({B, m, n}, wc.options());
+  auto grad_ins = at::zeros({B, n}, wc.options());
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(alignment_bwd_kernel, dim3(B), dim3(128), 0, stream,
+                     gc.data_ptr<float>(), wc.data_ptr<float>(),
+                     lc.data_ptr<int>(), grad_subs.data_ptr<float>(),
+                     grad_ins.data_ptr<float>(), B, (int)m, (int)n,
+                     (int)width);
+  return {grad_subs, grad_ins};
+}
