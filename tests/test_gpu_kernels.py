@@ -314,13 +314,7 @@ def test_train_step_gpu(setup):
 
     params, model, runner, rows = setup
     m = model  # fp32 weights on cuda already
-    opt, sched = lamb_lib.create_optimizer(
-        type(params)(params, initial_learning_rate=1e-3,
-                     end_learning_rate=1e-5, warmup_steps=10,
-                     weight_decay_rate=0.01, beta_1=0.9, beta_2=0.999,
-                     epsilon=1e-6),
-        100, m,
-    ) if False else lamb_lib.create_optimizer(params, 100, m)
+    opt, sched = lamb_lib.create_optimizer(params, 100, m)
     label = torch.randint(0, 5, (16, 100)).cuda()
     probs = m(rows.cuda(), training=True)
     loss = L.AlignmentLoss(del_cost=10.0, loss_reg=0.1,
