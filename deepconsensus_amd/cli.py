@@ -15,7 +15,7 @@ from typing import List, Optional
 import deepconsensus_amd
 
 COMMANDS = ["run", "preprocess", "calibrate", "filter_reads", "train",
-            "distill", "eval"]
+            "distill", "eval", "export"]
 
 
 def _run_main(argv: List[str]) -> None:
@@ -119,6 +119,10 @@ def main(argv: Optional[List[str]] = None) -> None:
         from deepconsensus_amd.models import infer_eval
 
         infer_eval.main(rest)
+    elif command == "export":
+        from deepconsensus_amd.models import export_model
+
+        export_model.main(rest)
     else:
         print(f"unknown command {command!r}; one of: {', '.join(COMMANDS)}")
         sys.exit(2)

@@ -59,10 +59,6 @@ def test_calculate_baseq_calibration(tmp_path):
         for line in f:
             q, m, x = line.strip().split(",")
             rows[int(q)] = (int(m), int(x))
-    # Interval striping covers positions 0..60 and 60..100 with stop
-    # boundaries inclusive per reference semantics: position 60 double
-    # counted at the seam is avoided... reference uses <= stop, so pos 60
-    # lands in both intervals. Accept 100 or 101 matches accordingly.
     # Position 60 sits in both intervals (inclusive stop, reference
     # semantics), so one duplicate count at the seam: 101 matches.
     assert rows[30][0] == 101
