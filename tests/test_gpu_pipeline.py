@@ -1,0 +1,78 @@
+"""GPU end-to-end pipeline tests: full train loop and full `run` inference
+on the native MI355X path."""
+import glob
+import json
+import os
+
+import pytest
+import torch
+
+from deepconsensus_amd.models import config as cfg
+
+from test_io_and_pipeline import make_test_bams
+from test_train import make_training_data, _tiny_params
+
+pytestmark = pytest.mark.gpu
+
+
+def test_train_e2e_gpu(tmp_path):
+    """Full training loop on cuda: HIP alignment loss + LAMB + checkpoints."""
+    from deepconsensus_amd.models import train as train_lib
+
+    train_file, _ = make_training_data(tmp_path)
+    params = _tiny_params(train_file)
+    params.batch_size = 8
+    cfg.modify_params(params)
+    out_dir = str(tmp_path / "model")
+    summary = train_lib.train_model(
+        out_dir, params, device="cuda", eval_every=2, limit_steps=3,
+    )
+    assert summary["steps"] >= 3
+    assert glob.glob(os.path.join(out_dir, "checkpoint-*.pt"))
+    assert 0.0 <= summary["eval/per_example_accuracy"] <= 1.0
+
+
+def test_quick_inference_e2e_gpu(tmp_path):
+    """`run` end-to-end on cuda with the native kernel path."""
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=4, length=300)
+    out = str(tmp_path / "out.fastq")
+    options = qi.InferenceOptions(
+        batch_size=64, batch_zmws=2, cpus=0, min_quality=0,
+        skip_windows_above=0,
+    )
+    counter = qi.run(
+        subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+        output=out, options=options, device="cuda",
+    )
+    assert counter.total == 4
+    stats = json.load(open(tmp_path / "out.inference.json"))
+    assert stats["n_zmw_processed"] == 4
+    from deepconsensus_amd.dcio.fastq import read_fastq
+
+    recs = list(read_fastq(out))
+    assert len(recs) == 4
+
+
+def test_trained_checkpoint_serves_gpu(tmp_path):
+    """Train -> checkpoint -> quick_inference with that checkpoint (cuda)."""
+    from deepconsensus_amd.inference import quick_inference as qi
+    from deepconsensus_amd.models import train as train_lib
+
+    train_file, _ = make_training_data(tmp_path, n_zmws=3)
+    params = _tiny_params(train_file)
+    cfg.modify_params(params)
+    out_dir = str(tmp_path / "model")
+    train_lib.train_model(out_dir, params, device="cuda", eval_every=100,
+                          limit_steps=1)
+
+    sub, ccs = make_test_bams(tmp_path / "infer", n_zmws=2, length=150)
+    out = str(tmp_path / "served.fastq")
+    options = qi.InferenceOptions(batch_size=32, batch_zmws=2, cpus=0,
+                                  min_quality=0, skip_windows_above=0)
+    counter = qi.run(
+        subreads_to_ccs=sub, ccs_bam=ccs, checkpoint=out_dir,
+        output=out, options=options, device="cuda",
+    )
+    assert counter.total == 2
