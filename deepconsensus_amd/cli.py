@@ -44,6 +44,9 @@ def _run_main(argv: List[str]) -> None:
                     help='"threshold,w,b" or "skip"; default from params.json')
     ap.add_argument("--ccs_calibration", default=None)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--shard", default=None,
+                    help="'i/N': process only ZMWs with index %% N == i "
+                    "(one process per GPU, like the reference's ccs --chunk)")
     args = ap.parse_args(argv)
 
     options = qi.InferenceOptions(
@@ -57,6 +60,9 @@ def _run_main(argv: List[str]) -> None:
         ins_trim=args.ins_trim,
         use_ccs_smart_windows=args.use_ccs_smart_windows,
     )
+    if args.shard:
+        i, n = args.shard.split("/")
+        options.shard_index, options.shard_count = int(i), int(n)
     outcome = qi.run(
         subreads_to_ccs=args.subreads_to_ccs,
         ccs_bam=args.ccs_bam,

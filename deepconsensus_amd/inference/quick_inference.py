@@ -73,6 +73,11 @@ class InferenceOptions:
     batch_zmws: int = 100
     ins_trim: int = 0
     use_ccs_smart_windows: bool = False
+    # ZMW sharding for multi-process / multi-GPU runs ("i/N", like the
+    # reference's `ccs --chunk` sharding, docs/quick_start.md:216-248):
+    # this process handles ZMWs with index % N == i.
+    shard_index: int = 0
+    shard_count: int = 1
 
 
 def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
@@ -329,8 +334,13 @@ def run(
 
     def zmw_batches():
         batch = []
+        idx = 0
         for input_data in proc_feeder():
             subreads, zmw, dcc, split, window_widths = input_data
+            keep = idx % options.shard_count == options.shard_index
+            idx += 1
+            if not keep:
+                continue
             batch.append((zmw, subreads, dcc, window_widths))
             if len(batch) >= options.batch_zmws:
                 yield batch

@@ -295,3 +295,24 @@ def test_filter_reads_fastq(tmp_path):
     # Threshold exactly at the quality keeps the read (rounding).
     fr.main(["-i", inp, "-o", outp, "-q", "10"])
     assert len(list(read_fastq(outp))) == 2
+
+
+def test_quick_inference_sharding(tmp_path):
+    """Two shards cover disjoint ZMWs whose union is the full set."""
+    from deepconsensus_amd.inference import quick_inference as qi
+    from deepconsensus_amd.dcio.fastq import read_fastq
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=4, length=120, seed=11)
+    names = []
+    for i in range(2):
+        out = str(tmp_path / f"out{i}.fastq")
+        options = qi.InferenceOptions(
+            batch_size=8, batch_zmws=2, cpus=0, min_quality=0,
+            skip_windows_above=0, shard_index=i, shard_count=2,
+        )
+        qi.run(subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+               output=out, options=options, device="cpu")
+        names.append({r.name for r in read_fastq(out)})
+    assert names[0] and names[1]
+    assert not (names[0] & names[1])
+    assert len(names[0] | names[1]) == 4
