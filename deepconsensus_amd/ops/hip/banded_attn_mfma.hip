@@ -63,39 +63,6 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
   const size_t RS = (size_t)3 * H * D;  // qkv row stride (elems)
   const bf16* base = qkv + (size_t)b * L * RS + (size_t)h * D;
 
-  // ---- Phase 0: stage the Q image (coalesced) into the Vt region and pull
-  // this wave's 9 B-fragments into registers; the region is then re-staged
-  // with V. Direct per-lane fragment loads from global would be 16-B reads
-  // at a 3*H*D-element stride (uncoalesced HBM traffic, measured 74%
-  // wave-parked). ----
-  bf16* qimg = &vt_lds[0][0];  // [L][K_STRIDE] overlay
-  for (int idx = tid; idx < L * (K_STRIDE / 2); idx += 256) {
-    const int r = idx / (K_STRIDE / 2), d2 = idx % (K_STRIDE / 2);
-    unsigned v = 0;
-    if (2 * d2 + 1 < D) {
-      v = *reinterpret_cast<const unsigned*>(base + r * RS + 2 * d2);
-    }
-    *reinterpret_cast<unsigned*>(&qimg[r * K_STRIDE + 2 * d2]) = v;
-  }
-  __syncthreads();
-
-  const int l0w_ = 32 * wave;
-  const int qrow_ = l0w_ + c;
-  bf16x8 qf[9];
-  {
-    const bool qv = qrow_ < L;
-#pragma unroll
-    for (int s = 0; s < 9; ++s) {
-      bf16x8 t = {};
-      if (qv) {
-        t = *reinterpret_cast<const bf16x8*>(
-            &qimg[qrow_ * K_STRIDE + 16 * s + 8 * hi]);
-      }
-      qf[s] = t;
-    }
-  }
-  __syncthreads();
-
   // ---- Stage K row-major (dims 140..151 zeroed). ----
   for (int idx = tid; idx < L * (K_STRIDE / 2); idx += 256) {
     const int r = idx / (K_STRIDE / 2), d2 = idx % (K_STRIDE / 2);
@@ -127,7 +94,31 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
     }
   }
 
-  const int l0w = l0w_;
+  // ---- Load Q B-fragments (this lane's query row, 9 k-steps of 16). ----
+  const int l0w = 32 * wave;
+  const int qrow = l0w + c;
+  bf16x8 qf[9];
+  {
+    const bool qv = qrow < L;
+    const unsigned short* qp =
+        reinterpret_cast<const unsigned short*>(base + (size_t)qrow * RS);
+#pragma unroll
+    for (int s = 0; s < 9; ++s) {
+      const int d0 = 16 * s + 8 * hi;
+      bf16x8 t = {};
+      if (qv) {
+        if (d0 + 8 <= D) {
+          t = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const bf16*>(qp) + d0);
+        } else {
+          unsigned short* tw = reinterpret_cast<unsigned short*>(&t);
+          for (int j = 0; j < 8; ++j) tw[j] = (d0 + j < D) ? qp[d0 + j] : 0;
+        }
+      }
+      qf[s] = t;
+    }
+  }
+  __syncthreads();
 
   // ---- QK^T (swapped): S^T tiles; lane holds qrow=l0w+c, keys in regs. ----
   const int kw0 = l0w - win;  // key window start (may be negative)
