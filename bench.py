@@ -58,8 +58,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch-size", type=int, default=2048)
-    ap.add_argument("--pool-batches", type=int, default=8)
+    ap.add_argument("--batch-size", type=int, default=8192)
+    ap.add_argument("--pool-batches", type=int, default=4)
+    ap.add_argument("--no-graphs", action="store_true",
+                    help="disable hipGraph capture of the serving step")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -122,11 +124,34 @@ def main():
 
         prefetch(0)
 
+        # hipGraph-capture the whole forward per device buffer: one replay
+        # per step instead of ~50 eager launches.
+        graphs = None
+        if not args.no_graphs and runner.native:
+            try:
+                for b in range(2):  # allocator warmup outside capture
+                    runner.forward_windows(dev_bufs[b])
+                torch.cuda.synchronize()
+                graphs, static_out = [], []
+                for b in range(2):
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        static_out.append(runner.forward_windows(dev_bufs[b]))
+                    graphs.append(g)
+            except Exception as e:  # pragma: no cover - graph support varies
+                print(f"# hipGraph capture unavailable ({e}); eager path",
+                      file=sys.stderr)
+                graphs = None
+
         def one_step(i: int):
             buf = i % 2
             cur = torch.cuda.current_stream()
             cur.wait_event(ready[buf])
-            bases, quals = runner.forward_windows(dev_bufs[buf])
+            if graphs is not None:
+                graphs[buf].replay()
+                bases, quals = static_out[buf]
+            else:
+                bases, quals = runner.forward_windows(dev_bufs[buf])
             consumed[buf].record(cur)
             prefetch(i + 1)
             # D2H of the uint8 calls (the serving step's output contract).
@@ -189,6 +214,7 @@ def main():
                 "windows_per_zmw": WINDOWS_PER_ZMW,
                 "parallelism": f"dp{world}",
                 "native_kernels": bool(runner.native),
+                "hipgraph": bool(have_cuda) and not args.no_graphs,
             },
         }
         print(json.dumps(result))
