@@ -62,6 +62,10 @@ def main():
     ap.add_argument("--pool-batches", type=int, default=4)
     ap.add_argument("--no-graphs", action="store_true",
                     help="disable hipGraph capture of the serving step")
+    ap.add_argument("--max-passes", type=int, default=20,
+                    help="subread stack depth (BASELINE config #5: 32)")
+    ap.add_argument("--windows-per-zmw", type=int, default=None,
+                    help="override (24 kb insert: 240)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -86,7 +90,9 @@ def main():
         )
 
     params = cfg.get_config("transformer_learn_values+custom")
+    params.max_passes = args.max_passes
     cfg.modify_params(params, is_training=False)
+    windows_per_zmw = args.windows_per_zmw or WINDOWS_PER_ZMW
     batch = args.batch_size if have_cuda else 32
 
     torch.manual_seed(1234)
@@ -189,7 +195,7 @@ def main():
         elapsed = float(t.item())
 
     windows_per_sec = args.steps * batch / elapsed * world
-    zmw_per_sec = windows_per_sec / WINDOWS_PER_ZMW
+    zmw_per_sec = windows_per_sec / windows_per_zmw
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
@@ -207,11 +213,12 @@ def main():
             "dtype": "bf16" if have_cuda else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "transformer_learn_values v1.2 (hidden 280, 6 layers, "
-                "heads 2, band +-12, max_passes 20)",
+                "model": "transformer_learn_values v1.2 (hidden 280, "
+                f"6 layers, heads 2, band +-12, max_passes "
+                f"{args.max_passes})",
                 "global_batch": batch * world,
                 "seq_len": 100,
-                "windows_per_zmw": WINDOWS_PER_ZMW,
+                "windows_per_zmw": windows_per_zmw,
                 "parallelism": f"dp{world}",
                 "native_kernels": bool(runner.native),
                 "hipgraph": bool(have_cuda) and not args.no_graphs,

@@ -326,3 +326,38 @@ def test_train_step_gpu(setup):
     sched.apply(opt, 0)
     opt.step()
     m.zero_grad(set_to_none=True)
+
+
+def test_max_passes_32_native(setup):
+    """BASELINE config #5 shape: max_passes=32 (133 rows) native path."""
+    params = cfg.get_config("transformer_learn_values+custom")
+    params.max_passes = 32
+    cfg.modify_params(params, is_training=False)
+    assert params.total_rows == 133
+    torch.manual_seed(3)
+    model = get_model(params)
+    runner = InferenceRunner(params, model, device="cuda:0")
+    assert runner.native
+    rng = np.random.default_rng(1)
+    B, R, L, mp = 8, 133, 100, 32
+    rows = np.zeros((B, R, L), dtype=np.float32)
+    rows[:, 0:mp] = rng.integers(0, 5, size=(B, mp, L))
+    rows[:, mp:2 * mp] = rng.integers(0, 256, size=(B, mp, L))
+    rows[:, 2 * mp:3 * mp] = rng.integers(0, 256, size=(B, mp, L))
+    rows[:, 3 * mp:4 * mp] = rng.integers(0, 3, size=(B, mp, L))
+    rows[:, 4 * mp] = rng.integers(0, 5, size=(B, L))
+    rows[:, -4:] = rng.integers(0, 501, size=(B, 4, 1))
+    t = torch.from_numpy(rows)
+    bases, quals = runner.forward_windows(t)
+    assert bases.shape == (8, 100)
+    # Embedding gather matches the fp32 reference at this depth too.
+    out = runner.ext.embed_gather(
+        t.cuda(), runner.table_flat, runner.row_shift, runner.row_vocab,
+        runner.chunk_cnt, runner.chunk_entries,
+    ).float()
+    with torch.no_grad():
+        saved = model.condense
+        model.condense = False
+        ref = model.embed(model._prepare_inputs(t.cuda()))
+        model.condense = saved
+    torch.testing.assert_close(out, ref, atol=0.05, rtol=0.01)
