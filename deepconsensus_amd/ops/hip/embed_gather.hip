@@ -57,18 +57,36 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
   const int npos = min(TILE_P, L - l0);
   for (int t = tid; t < npos * nchunk; t += 256) {
     const int p = t / nchunk, c = t % nchunk;
-    ushort val[8];
-    int pos = 0;
     const int cnt = chunk_cnt[c];
-    for (int k = 0; k < cnt; ++k) {
-      const int4 e = chunk_entries[c * 4 + k];  // x=row, y=elem_base, z=width
-      const int id = ids[e.x][p];
-      const ushort* src = reinterpret_cast<const ushort*>(
-          table_flat + e.y + (size_t)id * e.z);
-      for (int j = 0; j < e.z; ++j) val[pos++] = src[j];
-    }
     uint4 raw;
-    memcpy(&raw, val, 16);
+    if (cnt == 1) {
+      // Width-8 table row: one 16-B load.
+      const int4 e = chunk_entries[c * 4];
+      const int id = ids[e.x][p];
+      raw = *reinterpret_cast<const uint4*>(
+          table_flat + e.y + (size_t)id * 8);
+    } else {
+      // Narrow rows (e.g. 4 x width-2 strand entries): accumulate into two
+      // 64-bit registers — a local ushort[8] would land in scratch
+      // (runtime-indexed array, cdna_hip_programming.md rule 20).
+      unsigned long long lo = 0, hi = 0;
+      int pos = 0;
+      for (int k = 0; k < cnt; ++k) {
+        const int4 e = chunk_entries[c * 4 + k];
+        const int id = ids[e.x][p];
+        const ushort* src = reinterpret_cast<const ushort*>(
+            table_flat + e.y + (size_t)id * e.z);
+        for (int j = 0; j < e.z; ++j, ++pos) {
+          const unsigned long long v = src[j];
+          if (pos < 4) lo |= v << (16 * pos);
+          else hi |= v << (16 * (pos - 4));
+        }
+      }
+      raw.x = (unsigned)lo;
+      raw.y = (unsigned)(lo >> 32);
+      raw.z = (unsigned)hi;
+      raw.w = (unsigned)(hi >> 32);
+    }
     *reinterpret_cast<uint4*>(
         out + ((size_t)b * L + l0 + p) * H + c * 8) = raw;
   }

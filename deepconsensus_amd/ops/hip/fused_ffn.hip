@@ -92,16 +92,25 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
 
   // ---- T14 staging: issue loads to regs early, ds_write after MFMAs.
   // One shared register set: the W1 and W2 in-flight windows never overlap
-  // (issue_w2..write_w2 inside B1; issue_w1..write_w1 inside B2). ----
-  uint4 stage_regs[G_PER_T];
+  // (issue_w2..write_w2 inside B1; issue_w1..write_w1 inside B2).
+  // NAMED scalars, not an array: AMDGPUPromoteAlloca silently moves a
+  // per-thread uint4[5] into LDS (5*16B*512 threads = 40 KB!), turning the
+  // register pipeline into an LDS round-trip. ----
+  uint4 sr0, sr1, sr2, sr3, sr4;
+  uint4* stage_reg_ptrs[G_PER_T] = {&sr0, &sr1, &sr2, &sr3, &sr4};
+  (void)stage_reg_ptrs;
+#define STAGE_REG(i) (i == 0 ? sr0 : i == 1 ? sr1 : i == 2 ? sr2 \
+                      : i == 3 ? sr3 : sr4)
 
   auto issue_w1 = [&](int chunk) {
 #pragma unroll
     for (int i = 0; i < G_PER_T; ++i) {
       const int g = min(tid + i * 512, W1_GRAN - 1);
       const int row = g / (K1P / 8), k8 = g % (K1P / 8);
-      stage_regs[i] = *reinterpret_cast<const uint4*>(
+      const uint4 v = *reinterpret_cast<const uint4*>(
           w1 + (size_t)(chunk * NC + row) * K1P + 8 * k8);
+      if (i == 0) sr0 = v; else if (i == 1) sr1 = v; else if (i == 2) sr2 = v;
+      else if (i == 3) sr3 = v; else sr4 = v;
     }
   };
   auto write_w1 = [&]() {
@@ -110,7 +119,7 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
       const int g = min(tid + i * 512, W1_GRAN - 1);
       const int row = g / (K1P / 8), k8 = g % (K1P / 8);
       *reinterpret_cast<uint4*>(&w1_lds[row * W1_STRIDE + 8 * k8]) =
-          stage_regs[i];
+          (i == 0 ? sr0 : i == 1 ? sr1 : i == 2 ? sr2 : i == 3 ? sr3 : sr4);
     }
   };
   auto issue_w2 = [&](int chunk) {
@@ -118,8 +127,10 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
     for (int i = 0; i < G_PER_T; ++i) {
       const int g = min(tid + i * 512, W2_GRAN - 1);
       const int row = g / (NC / 8), k8 = g % (NC / 8);
-      stage_regs[i] = *reinterpret_cast<const uint4*>(
+      const uint4 v = *reinterpret_cast<const uint4*>(
           w2 + (size_t)row * NHID + chunk * NC + 8 * k8);
+      if (i == 0) sr0 = v; else if (i == 1) sr1 = v; else if (i == 2) sr2 = v;
+      else if (i == 3) sr3 = v; else sr4 = v;
     }
   };
   auto write_w2 = [&]() {
@@ -128,7 +139,7 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
       const int g = min(tid + i * 512, W2_GRAN - 1);
       const int row = g / (NC / 8), k8 = g % (NC / 8);
       *reinterpret_cast<uint4*>(&w2_lds[row * W2_STRIDE + 8 * k8]) =
-          stage_regs[i];
+          (i == 0 ? sr0 : i == 1 ? sr1 : i == 2 ? sr2 : i == 3 ? sr3 : sr4);
     }
   };
 
