@@ -203,16 +203,16 @@ class InferenceRunner:
                 lw["b2_a"] = lw["b2"] * lw["alpha_ffn"]
             # Padded weights for the fused FFN kernel (in-bounds 16B frags):
             # W1 [2048, 288], W2 [320, 2048], b1 fp32, b2 [320] fp32.
-            # The hand-fused FFN kernel currently measures ~2.2 ms/layer vs
-            # ~1.6 ms for hipBLASLt GEMMs with fused bias+ReLU epilogues at
-            # batch 4096 (profiles/r01_*), so the eager path is the default;
-            # set DC_FUSED_FFN=1 to use the kernel.
+            # The hand-fused FFN kernel measures 1.46 ms/layer vs 1.79 ms
+            # for the hipBLASLt pair with fused epilogues at batch 4096
+            # (profiles/r01_perf_journal.md) - default ON for the production
+            # shape; DC_FUSED_FFN=0 falls back to hipBLASLt.
             import os as _os
 
             self.ffn_fused_ok = (
                 model.params["hidden_size"] == 280
                 and model.params["filter_size"] == 2048
-                and _os.environ.get("DC_FUSED_FFN") == "1"
+                and _os.environ.get("DC_FUSED_FFN") != "0"
             )
             if self.ffn_fused_ok:
                 for i, l in enumerate(model.layers):
