@@ -217,6 +217,19 @@ class InferenceRunner:
             if self.ffn_fused_ok:
                 for i, l in enumerate(model.layers):
                     lw = self.layer_w[i]
+                    # Padded projection weights for fused_linear (K5/K7):
+                    # QKV rows [840->896, 288], out-proj [320, 288].
+                    wq = l.attn.q_proj.weight.detach().float()
+                    wk = l.attn.k_proj.weight.detach().float()
+                    wv = l.attn.v_proj.weight.detach().float()
+                    wqkv = torch.cat([wq, wk, wv], 0)  # [840, 280]
+                    wqkv_pad = torch.zeros(896, 288)
+                    wqkv_pad[:840, :280] = wqkv
+                    lw["wqkv_pad"] = wqkv_pad.to(bf16).contiguous().to(dev)
+                    wo = l.attn.out_proj.weight.detach().float()
+                    wout_pad = torch.zeros(320, 288)
+                    wout_pad[:280, :280] = wo
+                    lw["wout_pad"] = wout_pad.to(bf16).contiguous().to(dev)
                     w1 = l.ffn.filter_layer.weight.detach().float()
                     w1p = torch.zeros(2048, 288)
                     w1p[:, :280] = w1
@@ -275,17 +288,28 @@ class InferenceRunner:
         if self.rezero_fast:
             h = x.shape[-1]
             flat = x.reshape(b * l, h)
+            empty = flat.new_empty(0)
             for lw in self.layer_w:
-                qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
-                a = self._attn(qkv)
-                # Residual fused into the GEMM epilogue (alpha pre-folded).
-                flat = torch.addmm(flat, a.view(b * l, h), lw["wout_t_a"])
                 if self.ffn_fused_ok:
+                    qkv = self.ext.fused_linear(
+                        flat, lw["wqkv_pad"], empty, empty, 840, False, 0.0
+                    ).view(b, l, -1)
+                else:
+                    qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
+                a = self._attn(qkv)
+                if self.ffn_fused_ok:
+                    flat = self.ext.fused_linear(
+                        a.view(b * l, h), lw["wout_pad"], empty, flat,
+                        280, False, lw["alpha_attn"],
+                    )
                     flat = self.ext.fused_ffn(
                         flat, lw["w1_pad"], lw["b1_f32"], lw["w2_pad"],
                         lw["b2_f32"], lw["alpha_ffn"],
                     )
-                elif self._addmm_act:
+                    continue
+                # Residual fused into the GEMM epilogue (alpha pre-folded).
+                flat = torch.addmm(flat, a.view(b * l, h), lw["wout_t_a"])
+                if self._addmm_act:
                     ff = torch._addmm_activation(lw["b1"], flat, lw["w1_t"])
                     flat = torch.addmm(flat, ff, lw["w2_t_a"]).add_(lw["b2_a"])
                 else:

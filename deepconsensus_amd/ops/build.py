@@ -18,6 +18,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "embed_gather.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_ffn.hip"),
     os.path.join(_OPS_DIR, "hip", "alignment_dp.hip"),
+    os.path.join(_OPS_DIR, "hip", "fused_linear.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

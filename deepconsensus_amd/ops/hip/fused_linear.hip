@@ -1,0 +1,225 @@
+// Fused linear projection for gfx950: out = act(x @ W^T + b)[*alpha + x].
+//
+// The K5 QKV projection and K7 output projection (attention_layer.py:78-107)
+// as one tiled MFMA kernel reusing the fused-FFN structure: 512 threads =
+// 8 waves per 128-row tile, K fixed at the model width 280 (padded 288),
+// N processed in 64-column chunks with T14 async weight staging (issue
+// global loads to named registers before the MFMA cluster, ds_write after),
+// outputs staged through LDS for coalesced 16-B global writes. Optional
+// fused bias, ReLU, and ReZero residual (out = x + alpha*y).
+//
+// W arrives host-padded to [Npad, 288] with Npad a multiple of 64.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int BM = 128;
+constexpr int K1 = 280;
+constexpr int K1P = 288;
+constexpr int NC = 64;
+constexpr int W_STRIDE = 296;   // weight image row stride (conflict-free)
+constexpr int O_STRIDE = 72;    // output chunk image stride
+constexpr int WG = NC * (K1P / 8);   // granules per weight slice (2304)
+constexpr int G_PER_T = 5;
+
+template <bool RELU, bool RESIDUAL>
+__global__ __launch_bounds__(512, 2) void fused_linear_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ bias, const bf16* __restrict__ resid,
+    bf16* __restrict__ out, int M, int N, int Npad, float alpha) {
+  __shared__ __attribute__((aligned(16))) bf16 w_lds[NC * W_STRIDE];
+  __shared__ __attribute__((aligned(16))) bf16 o_lds[BM * O_STRIDE];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int c = lane & 31;
+  const int hi = lane >> 5;
+  const int rg = wave >> 1;        // row group
+  const int ch = wave & 1;         // 32-col half of the chunk
+  const int m0 = blockIdx.x * BM;
+
+  // Stage x image over w_lds+o_lds... x needs 128*296 elems = 37,888;
+  // w_lds (18,944) + o_lds (9,216) = 28,160 < that, so stage x in two
+  // passes: pull this wave's A-fragments directly with a strided-but-L2
+  // friendly pattern instead: stage rows through w_lds half at a time.
+  bf16x8 af[18];
+  for (int half = 0; half < 2; ++half) {
+    __syncthreads();
+    for (int idx = tid; idx < 64 * (W_STRIDE / 2); idx += 512) {
+      const int r = idx / (W_STRIDE / 2), d2 = idx % (W_STRIDE / 2);
+      const int row = 64 * half + r;
+      unsigned v = 0;
+      if (m0 + row < M && 2 * d2 + 1 < K1) {
+        v = *reinterpret_cast<const unsigned*>(
+            x + (size_t)(m0 + row) * K1 + 2 * d2);
+      }
+      *reinterpret_cast<unsigned*>(&w_lds[r * W_STRIDE + 2 * d2]) = v;
+    }
+    __syncthreads();
+    // Waves whose row group sits in this half pull their fragments.
+    if ((32 * rg) / 64 == half) {
+      const int r_local = 32 * rg - 64 * half + c;
+#pragma unroll
+      for (int s = 0; s < 18; ++s) {
+        af[s] = *reinterpret_cast<const bf16x8*>(
+            &w_lds[r_local * W_STRIDE + 16 * s + 8 * hi]);
+      }
+    }
+  }
+  __syncthreads();
+
+  // T14 staging registers (named: an array would be promoted to LDS).
+  uint4 sr0, sr1, sr2, sr3, sr4;
+
+  auto issue_w = [&](int chunk) {
+#pragma unroll
+    for (int i = 0; i < G_PER_T; ++i) {
+      const int g = min(tid + i * 512, WG - 1);
+      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
+      const uint4 v = *reinterpret_cast<const uint4*>(
+          w + (size_t)(chunk * NC + row) * K1P + 8 * k8);
+      if (i == 0) sr0 = v; else if (i == 1) sr1 = v;
+      else if (i == 2) sr2 = v; else if (i == 3) sr3 = v; else sr4 = v;
+    }
+  };
+  auto write_w = [&]() {
+#pragma unroll
+    for (int i = 0; i < G_PER_T; ++i) {
+      const int g = min(tid + i * 512, WG - 1);
+      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
+      *reinterpret_cast<uint4*>(&w_lds[row * W_STRIDE + 8 * k8]) =
+          (i == 0 ? sr0 : i == 1 ? sr1 : i == 2 ? sr2
+           : i == 3 ? sr3 : sr4);
+    }
+  };
+
+  issue_w(0);
+  write_w();
+
+  const int nchunk = Npad / NC;
+  for (int chunk = 0; chunk < nchunk; ++chunk) {
+    __syncthreads();  // w_lds chunk ready; o_lds free
+    if (chunk + 1 < nchunk) issue_w(chunk + 1);
+
+    const int colt = 32 * ch;
+    const int ncol = chunk * NC + colt + c;  // this lane's output col
+    f32x16 acc = {};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 18; ++s) {
+      const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+          &w_lds[(colt + c) * W_STRIDE + 16 * s + 8 * hi]);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], bfr, acc,
+                                                    0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    const float bv = (bias != nullptr && ncol < N) ? bias[ncol] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * rg;
+      float v = acc[r] + bv;
+      if (RELU) v = v > 0.f ? v : 0.f;
+      o_lds[row * O_STRIDE + colt + c] = __float2bfloat16(v);
+    }
+    __syncthreads();  // o_lds complete; w_lds consumed
+    if (chunk + 1 < nchunk) write_w();
+
+    // Coalesced copy-out of this 64-col chunk (+ optional residual).
+    const int n0 = chunk * NC;
+    for (int idx = tid; idx < BM * (NC / 8); idx += 512) {
+      const int row = idx / (NC / 8), c8 = idx % (NC / 8);
+      if (m0 + row >= M) continue;
+      const int col = n0 + 8 * c8;
+      if (col >= N) continue;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          &o_lds[row * O_STRIDE + 8 * c8]);
+      if (RESIDUAL) {
+        const bf16x8 res = *reinterpret_cast<const bf16x8*>(
+            resid + (size_t)(m0 + row) * K1 + col);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float rv = (float)res[j];
+          const float vv = (float)v[j];
+          v[j] = (__bf16)(rv + alpha * vv);
+        }
+      }
+      if (col + 8 <= N) {
+        *reinterpret_cast<bf16x8*>(
+            out + (size_t)(m0 + row) * N + col) = v;
+      } else {
+        const unsigned short* vs =
+            reinterpret_cast<const unsigned short*>(&v);
+        for (int j = 0; j < 8 && col + j < N; ++j) {
+          reinterpret_cast<unsigned short*>(
+              out)[(size_t)(m0 + row) * N + col + j] = vs[j];
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
+                        at::Tensor resid, int64_t n_out, bool relu,
+                        double alpha) {
+  const bool residual = resid.defined() && resid.numel() > 0;
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16,
+              "x must be bf16 on device");
+  auto xc = x.contiguous();
+  const int K = xc.size(-1);
+  const int M = xc.numel() / K;
+  TORCH_CHECK(K == K1, "fused_linear requires width-280 input");
+  const int Npad = w.size(0);
+  TORCH_CHECK(w.size(1) == K1P && Npad % NC == 0,
+              "w must be [Npad (mult of 64), 288]");
+  const int N = (int)n_out;
+  TORCH_CHECK(N <= Npad, "n_out exceeds padded weight rows");
+  TORCH_CHECK(!residual || N == K1, "residual requires N == 280");
+  const bf16* resid_ptr = nullptr;
+  at::Tensor rc;
+  if (residual) {
+    rc = resid.contiguous();
+    TORCH_CHECK(rc.dtype() == at::kBFloat16 && rc.numel() == xc.numel(),
+                "resid must match x");
+    resid_ptr = reinterpret_cast<bf16*>(rc.data_ptr());
+  }
+  auto out = at::empty({M, N}, xc.options());
+  const float* bias_ptr = nullptr;
+  at::Tensor bc;
+  if (bias.defined() && bias.numel() > 0) {
+    bc = bias.contiguous();
+    TORCH_CHECK(bc.dtype() == at::kFloat && bc.numel() >= N, "bias fp32");
+    bias_ptr = bc.data_ptr<float>();
+  }
+  dim3 grid((M + BM - 1) / BM);
+  dim3 block(512);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto relu_t, auto res_t) {
+    hipLaunchKernelGGL(
+        (fused_linear_kernel<decltype(relu_t)::value,
+                             decltype(res_t)::value>),
+        grid, block, 0, stream,
+        reinterpret_cast<bf16*>(xc.data_ptr()),
+        reinterpret_cast<bf16*>(w.data_ptr()), bias_ptr, resid_ptr,
+        reinterpret_cast<bf16*>(out.data_ptr()), M, N, Npad,
+        (float)alpha);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  if (relu && residual) launch(T{}, T{});
+  else if (relu) launch(T{}, F{});
+  else if (residual) launch(F{}, T{});
+  else launch(F{}, F{});
+  return out;
+}

@@ -361,3 +361,40 @@ def test_max_passes_32_native(setup):
         ref = model.embed(model._prepare_inputs(t.cuda()))
         model.condense = saved
     torch.testing.assert_close(out, ref, atol=0.05, rtol=0.01)
+
+
+def test_fused_linear_matches_torch(setup):
+    """fused_linear vs fp32 torch: plain / bias+relu / residual variants."""
+    params, model, runner, rows = setup
+    torch.manual_seed(31)
+    empty = torch.empty(0, device="cuda", dtype=torch.bfloat16)
+    for M in (1600, 129):
+        x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+        # N=840 (QKV shape), no bias.
+        w = torch.randn(840, 280, device="cuda") * 0.05
+        w_pad = torch.zeros(896, 288, device="cuda")
+        w_pad[:840, :280] = w
+        w_pad = w_pad.to(torch.bfloat16).contiguous()
+        out = runner.ext.fused_linear(
+            x, w_pad, empty, empty, 840, False, 0.0
+        ).float()
+        ref = x.float() @ w.t()
+        assert (out - ref).abs().max().item() < 0.05
+        # N=320-padded 280 with bias + relu.
+        w2 = torch.randn(280, 280, device="cuda") * 0.05
+        b2 = torch.randn(280, device="cuda")
+        w2_pad = torch.zeros(320, 288, device="cuda")
+        w2_pad[:280, :280] = w2
+        w2_pad = w2_pad.to(torch.bfloat16).contiguous()
+        out2 = runner.ext.fused_linear(
+            x, w2_pad, b2, empty, 280, True, 0.0
+        ).float()
+        ref2 = torch.relu(x.float() @ w2.t() + b2)
+        assert (out2 - ref2).abs().max().item() < 0.05
+        # Residual + alpha.
+        resid = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+        out3 = runner.ext.fused_linear(
+            x, w2_pad, empty, resid, 280, False, 0.6
+        ).float()
+        ref3 = resid.float() + 0.6 * (x.float() @ w2.t())
+        assert (out3 - ref3).abs().max().item() < 0.06
