@@ -19,7 +19,6 @@ COMMANDS = ["run", "preprocess", "calibrate", "filter_reads", "train",
 
 
 def _run_main(argv: List[str]) -> None:
-    from deepconsensus_amd.calibration import calibration as cal
     from deepconsensus_amd.inference import quick_inference as qi
 
     ap = argparse.ArgumentParser("deepconsensus run")
