@@ -230,10 +230,29 @@ def main(argv: Optional[List[str]] = None) -> None:
     device = args.device or (
         "cuda" if torch.cuda.is_available() else "cpu"
     )
-    train_model(
-        args.out_dir, params, device=device, eval_every=args.eval_every,
-        limit_steps=args.limit_steps, warm_start=args.checkpoint,
-    )
+    # Parity with the reference's retry-on-UnavailableError loop
+    # (model_train_custom_loop.py:333-347): transient device/communicator
+    # failures resume from the latest checkpoint, bounded at 5 attempts.
+    attempts = 0
+    while True:
+        try:
+            train_model(
+                args.out_dir, params, device=device,
+                eval_every=args.eval_every, limit_steps=args.limit_steps,
+                warm_start=args.checkpoint,
+            )
+            break
+        except RuntimeError as e:
+            attempts += 1
+            transient = any(
+                k in str(e).lower()
+                for k in ("nccl", "rccl", "connection", "timed out",
+                          "unavailable")
+            )
+            if not transient or attempts >= 5:
+                raise
+            log.warning("transient failure (%s); retrying (%d/5)",
+                        e, attempts)
 
 
 if __name__ == "__main__":
