@@ -78,3 +78,76 @@ def test_examples_bit_exact(golden_run, split, n):
         assert np.array_equal(la, lb), k
         assert (ref[k]["ccs_base_quality_scores"][1]
                 == mine[k]["ccs_base_quality_scores"][1]), k
+
+
+def test_inference_mode_counters(tmp_path):
+    """Preprocess inference-mode summary matches the reference golden."""
+    from deepconsensus_amd.preprocess import preprocess_cli
+
+    out = str(tmp_path / "inf" / "ex.tfrecord.gz")
+    preprocess_cli.main([
+        "--subreads_to_ccs", f"{REF}/subreads_to_ccs.bam",
+        "--ccs_bam", f"{REF}/ccs.bam",
+        "--output", out, "--cpus", "0", "--ins_trim", "5",
+    ])
+    mine = json.load(open(str(tmp_path / "inf" / "ex.inference.json")))
+    ref = json.load(
+        open(f"{REF}/tf_examples/summary/summary.inference.json")
+    )
+    skip = ("subreads_to_ccs", "ccs_bam", "truth_to_ccs", "truth_bed",
+            "truth_split", "version", "ins_trim")
+    for k, v in ref.items():
+        if k.startswith(skip):
+            continue
+        assert str(mine.get(k)) == str(v), (k, v, mine.get(k))
+
+
+@pytest.mark.parametrize("q", [0, 10, 20, 30, 40, 50])
+def test_filter_reads_golden(tmp_path, q):
+    """filter_reads reproduces the reference's golden filtered FASTQs."""
+    from deepconsensus_amd.calibration import filter_reads as fr
+    from deepconsensus_amd.dcio.fastq import read_fastq
+
+    base = "/root/reference/deepconsensus/testdata/filter_fastq"
+    inp = f"{base}/m64062_190806_063919_q0_chr20_100reads.fq.gz"
+    golden = f"{base}/m64062_190806_063919_q0_chr20_100reads.q{q}.fq.gz"
+    out = str(tmp_path / f"out_q{q}.fastq")
+    fr.main(["-i", inp, "-o", out, "-q", str(q)])
+    got = [(r.name, r.sequence, r.quality) for r in read_fastq(out)]
+    want = [(r.name, r.sequence, r.quality) for r in read_fastq(golden)]
+    assert got == want
+
+
+def test_filter_reads_bam_golden(tmp_path):
+    """BAM input path against the q30 golden."""
+    from deepconsensus_amd.calibration import filter_reads as fr
+    from deepconsensus_amd.dcio.fastq import read_fastq
+
+    base = "/root/reference/deepconsensus/testdata/filter_fastq"
+    inp = f"{base}/m64062_190806_063919-chr20.dc.small.bam"
+    golden = f"{base}/m64062_190806_063919-chr20.dc.small.q30.fq.gz"
+    out = str(tmp_path / "out_bam_q30.fastq")
+    fr.main(["-i", inp, "-o", out, "-q", "30"])
+    got = [(r.name, r.sequence, r.quality) for r in read_fastq(out)]
+    want = [(r.name, r.sequence, r.quality) for r in read_fastq(golden)]
+    assert got == want
+
+
+def test_reference_params_json_loads():
+    """The reference's shipped params.json resolves through our config."""
+    from deepconsensus_amd.models import config as cfg
+
+    p = cfg.read_params_from_json(
+        "/root/reference/deepconsensus/testdata/model"
+    )
+    cfg.modify_params(p, is_training=False)
+    assert p.hidden_size == 280
+    assert p.total_rows == 85
+    assert p.dc_calibration == "0,1.197654,-0.99781"
+    from deepconsensus_amd.models.model import get_model
+
+    m = get_model(p)
+    n_params = sum(x.numel() for x in m.parameters())
+    # 8.94 M trainable parameters; the reference's 38.18 MB checkpoint
+    # (docs/quick_start.md:104) implies ~9.5 M fp32 slots incl. bookkeeping.
+    assert 8.5e6 < n_params < 10.1e6, n_params
