@@ -214,6 +214,8 @@ class InferenceRunner:
                 and model.params["filter_size"] == 2048
                 and _os.environ.get("DC_FUSED_FFN") != "0"
             )
+            self.ffn_v2 = _os.environ.get("DC_FFN_V2", "1") != "0"
+
             if self.ffn_fused_ok:
                 for i, l in enumerate(model.layers):
                     lw = self.layer_w[i]
@@ -244,6 +246,12 @@ class InferenceRunner:
                     b2p = torch.zeros(320)
                     b2p[:280] = l.ffn.output_layer.bias.detach().float()
                     lw["b2_f32"] = b2p.to(dev)
+                    # v2 (glds-pipelined) W1 layout: [2048, 296] with b1
+                    # folded into column 287 (constant-1 input column).
+                    w1v2 = torch.zeros(2048, 296)
+                    w1v2[:, :280] = w1
+                    w1v2[:, 287] = l.ffn.filter_layer.bias.detach().float()
+                    lw["w1_v2"] = w1v2.to(bf16).contiguous().to(dev)
             # hipBLASLt fused bias+ReLU epilogue, when this torch exposes it.
             self._addmm_act = hasattr(torch, "_addmm_activation")
             if self._addmm_act:
@@ -302,10 +310,16 @@ class InferenceRunner:
                         a.view(b * l, h), lw["wout_pad"], empty, flat,
                         280, False, lw["alpha_attn"],
                     )
-                    flat = self.ext.fused_ffn(
-                        flat, lw["w1_pad"], lw["b1_f32"], lw["w2_pad"],
-                        lw["b2_f32"], lw["alpha_ffn"],
-                    )
+                    if self.ffn_v2:
+                        flat = self.ext.fused_ffn_v2(
+                            flat, lw["w1_v2"], lw["w2_pad"],
+                            lw["b2_f32"], lw["alpha_ffn"],
+                        )
+                    else:
+                        flat = self.ext.fused_ffn(
+                            flat, lw["w1_pad"], lw["b1_f32"], lw["w2_pad"],
+                            lw["b2_f32"], lw["alpha_ffn"],
+                        )
                     continue
                 # Residual fused into the GEMM epilogue (alpha pre-folded).
                 flat = torch.addmm(flat, a.view(b * l, h), lw["wout_t_a"])

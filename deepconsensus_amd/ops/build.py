@@ -19,6 +19,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "fused_ffn.hip"),
     os.path.join(_OPS_DIR, "hip", "alignment_dp.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_linear.hip"),
+    os.path.join(_OPS_DIR, "hip", "fused_ffn_glds.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

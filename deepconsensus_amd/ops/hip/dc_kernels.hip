@@ -178,6 +178,8 @@ at::Tensor embed_gather(at::Tensor rows, at::Tensor table_flat,
                         at::Tensor chunk_cnt, at::Tensor chunk_entries);
 at::Tensor fused_ffn(at::Tensor x, at::Tensor w1, at::Tensor b1,
                      at::Tensor w2, at::Tensor b2, double alpha);
+at::Tensor fused_ffn_v2(at::Tensor x, at::Tensor w1, at::Tensor w2,
+                        at::Tensor b2, double alpha);
 at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
                         at::Tensor resid, int64_t n_out, bool relu,
                         double alpha);
@@ -200,6 +202,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Subread-stack embedding gather (K2)");
   m.def("fused_ffn", &fused_ffn,
         "Fused FFN + ReZero residual (K8+K9), hidden tensor LDS-resident");
+  m.def("fused_ffn_v2", &fused_ffn_v2,
+        "Fused FFN v2: glds-pipelined weights, bias folded into W1");
   m.def("fused_linear", &fused_linear,
         "Fused linear projection (K5/K7): act(xW^T+b)[*alpha+x]");
   m.def("alignment_dp_fwd", &alignment_dp_fwd,

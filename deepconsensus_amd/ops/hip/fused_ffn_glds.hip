@@ -1,0 +1,232 @@
+// Fused FFN v2 for gfx950: LDS-DMA (global_load_lds) pipelined weights.
+//
+// Same math as fused_ffn.hip, restructured per cdna_hip_programming.md
+// sections 5/T3/T4:
+//  * W1 host-padded to [2048, 296] with b1 FOLDED into column 287 (the x
+//    image carries a constant-1 in that column), so the B1 phase issues no
+//    scalar loads that would make hipcc drain the DMA queue (trap b);
+//  * ALL LDS in ONE __shared__ array (trap a);
+//  * weights stream by global_load_lds in 1-KiB wave chunks (wave-uniform
+//    LDS dst + per-lane source), W1 double-buffered: W1(c+1)'s DMA stays in
+//    flight across B1(c)'s barrier behind a counted asm s_waitcnt (raw
+//    s_barrier; __syncthreads would emit vmcnt(0) and drain it);
+//  * B1: 36 MFMAs -> relu -> h bf16 to LDS; B2: 40 MFMAs into the fp32 out
+//    accumulators; epilogue folds b2 + ReZero alpha + residual.
+// LDS: W1 2x37,888 B + W2 46,080 B + h 18,432 B + 3 KiB DMA overflow
+// scratch = 143,360 B (1 block/CU).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int BM = 128;
+constexpr int K1 = 280;
+constexpr int K1P = 296;        // padded k incl. bias column 287
+constexpr int BIAS_COL = 287;
+constexpr int NC = 64;
+constexpr int NHID = 2048;
+constexpr int NCHUNK = NHID / NC;
+constexpr int W1_STRIDE = K1P;
+constexpr int W2_STRIDE = 72;   // 9 granules/row; granule 8 is a pad slot
+constexpr int H_STRIDE = 72;
+constexpr int W2_ROWS = 280;
+constexpr int NOUT_PAD = 320;
+
+constexpr int W1_BYTES = 64 * W1_STRIDE * 2;        // 37,888 (37 KiB chunks)
+constexpr int W1_CHUNKS = W1_BYTES / 1024;          // 37 -> 5 issues/wave
+constexpr int W2_GRAN = W2_ROWS * (W2_STRIDE / 8);  // 2,520 granules
+constexpr int W2_CHUNKS = (W2_GRAN + 63) / 64;      // 40 -> 5 issues/wave
+
+constexpr int OFF_W1 = 0;                            // elems
+constexpr int OFF_W2 = 2 * 64 * W1_STRIDE;           // 37,888 elems
+constexpr int OFF_H = OFF_W2 + NOUT_PAD * W2_STRIDE; // +23,040
+constexpr int OFF_SCRATCH = OFF_H + BM * H_STRIDE;   // +9,216
+constexpr int LDS_ELEMS = OFF_SCRATCH + 2048;        // 3 overflow chunks + pad
+
+__device__ __forceinline__ void glds16(const bf16* gsrc, bf16* ldst) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned*)gsrc,
+      (__attribute__((address_space(3))) unsigned*)ldst, 16, 0, 0);
+}
+
+__global__ __launch_bounds__(512, 2) void fused_ffn_v2_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w1,
+    const bf16* __restrict__ w2, const float* __restrict__ b2,
+    bf16* __restrict__ out, int M, float alpha) {
+  __shared__ __attribute__((aligned(16))) bf16 smem[LDS_ELEMS];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int c = lane & 31;
+  const int hi = lane >> 5;
+  const int rg = wave >> 1;
+  const int ch = wave & 1;
+  const int m0 = blockIdx.x * BM;
+
+  // ---- Stage x image over the W1 double-buffer region (exact fit):
+  // cols 280..286 zero, col 287 = bf16(1.0), rows >= M zero. ----
+  for (int idx = tid; idx < BM * (K1P / 2); idx += 512) {
+    const int r = idx / (K1P / 2), d2 = idx % (K1P / 2);
+    unsigned v = 0;
+    const bool rv = (m0 + r) < M;
+    if (rv && 2 * d2 + 1 < K1) {
+      v = *reinterpret_cast<const unsigned*>(
+          x + (size_t)(m0 + r) * K1 + 2 * d2);
+    } else if (rv && 2 * d2 + 1 == BIAS_COL) {
+      v = 0x3f800000u;  // upper half bf16(1.0) at col 287, lower col 286 = 0
+    }
+    *reinterpret_cast<unsigned*>(&smem[r * K1P + 2 * d2]) = v;
+  }
+  __syncthreads();
+  bf16x8 af[18];
+#pragma unroll
+  for (int s = 0; s < 18; ++s) {
+    af[s] = *reinterpret_cast<const bf16x8*>(
+        &smem[(32 * rg + c) * K1P + 16 * s + 8 * hi]);
+  }
+  __syncthreads();
+
+  // ---- LDS-DMA issue helpers: per wave, 5 wave-uniform 1-KiB chunks
+  // (clamped overflow chunks land in the scratch region so every wave has
+  // exactly 5 DMAs in flight -> uniform vmcnt counts). ----
+  auto issue_w1 = [&](int chunk) {
+    const bf16* src = w1 + (size_t)chunk * NC * K1P;
+    bf16* dst = &smem[OFF_W1 + (chunk & 1) * 64 * W1_STRIDE];
+#pragma unroll
+    for (int i = 0; i < 5; ++i) {
+      const int ck = wave + i * 8;
+      const bool ok = ck < W1_CHUNKS;
+      bf16* d = ok ? dst + ck * 512 : &smem[OFF_SCRATCH];
+      const bf16* s =
+          src + (ok ? ck * 512 : 0) + lane * 8;
+      glds16(s, d);
+    }
+  };
+  auto issue_w2 = [&](int chunk) {
+#pragma unroll
+    for (int i = 0; i < 5; ++i) {
+      const int ck = wave + i * 8;
+      const int g = min(ck * 64 + lane, W2_GRAN - 1);
+      const int row = g / 9, sub = g % 9;
+      const int k8 = sub == 8 ? 0 : sub;
+      // dst must be wave-uniform; granule (ck*64+lane) maps to LDS bytes
+      // 16*(ck*64) + lane*16 automatically.
+      bf16* d = &smem[OFF_W2 + ck * 512];
+      if (ck * 64 >= W2_GRAN) d = &smem[OFF_SCRATCH];
+      glds16(w2 + (size_t)row * NHID + chunk * NC + 8 * k8, d);
+    }
+  };
+
+  issue_w1(0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  f32x16 oacc[5] = {};
+
+  for (int chunk = 0; chunk < NCHUNK; ++chunk) {
+    // ---- B1: issue W2(c) + W1(c+1) DMAs, then MFMAs vs W1[c&1]. ----
+    issue_w2(chunk);
+    const bool more = chunk + 1 < NCHUNK;
+    if (more) issue_w1(chunk + 1);
+    const bf16* w1buf = &smem[OFF_W1 + (chunk & 1) * 64 * W1_STRIDE];
+    const int colt = 32 * ch;
+    f32x16 acc = {};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 18; ++s) {
+      const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+          &w1buf[(colt + c) * W1_STRIDE + 16 * s + 8 * hi]);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], bfr, acc,
+                                                    0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * rg;
+      float v = acc[r];
+      v = v > 0.f ? v : 0.f;
+      smem[OFF_H + row * H_STRIDE + colt + c] = __float2bfloat16(v);
+    }
+    // W2(c) landed (its 5 DMAs are the oldest); W1(c+1)'s 5 stay in flight.
+    if (more) {
+      asm volatile("s_waitcnt vmcnt(5) lgkmcnt(0)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    // ---- B2: MFMAs vs W2 + h. ----
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < NC / 16; ++s) {
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &smem[OFF_H + (32 * rg + c) * H_STRIDE + 16 * s + 8 * hi]);
+#pragma unroll
+      for (int ct = 0; ct < 5; ++ct) {
+        const int ocol = min(140 * ch + 32 * ct + c, W2_ROWS - 1);
+        const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            &smem[OFF_W2 + ocol * W2_STRIDE + 16 * s + 8 * hi]);
+        oacc[ct] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a, bfr, oacc[ct], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // ---- Epilogue (fully unrolled; rule 20). ----
+#pragma unroll
+  for (int ct = 0; ct < 5; ++ct) {
+    const int col = 140 * ch + 32 * ct + c;
+    if (col >= K1) continue;
+    const float bias = b2[col];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * rg;
+      if (m0 + row < M) {
+        const size_t off = (size_t)(m0 + row) * K1 + col;
+        const float resid = __bfloat162float(x[off]);
+        out[off] = __float2bfloat16(resid + alpha * (oacc[ct][r] + bias));
+      }
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor fused_ffn_v2(at::Tensor x, at::Tensor w1, at::Tensor w2,
+                        at::Tensor b2, double alpha) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16,
+              "x must be bf16 on device");
+  auto xc = x.contiguous();
+  const int K = xc.size(-1);
+  const int M = xc.numel() / K;
+  TORCH_CHECK(K == K1, "fused_ffn_v2 requires width 280");
+  TORCH_CHECK(w1.size(0) == NHID && w1.size(1) == K1P,
+              "w1 must be [2048, 296] with b1 folded into column 287");
+  TORCH_CHECK(w2.size(0) == NOUT_PAD && w2.size(1) == NHID,
+              "w2 must be padded [320, 2048]");
+  TORCH_CHECK(b2.numel() == NOUT_PAD, "b2 must be padded [320]");
+  auto out = at::empty_like(xc);
+  dim3 grid((M + BM - 1) / BM);
+  dim3 block(512);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fused_ffn_v2_kernel, grid, block, 0, stream,
+                     reinterpret_cast<bf16*>(xc.data_ptr()),
+                     reinterpret_cast<bf16*>(w1.data_ptr()),
+                     reinterpret_cast<bf16*>(w2.data_ptr()),
+                     b2.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(out.data_ptr()),
+                     M, (float)alpha);
+  return out;
+}

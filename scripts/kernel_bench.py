@@ -51,6 +51,11 @@ def main():
                     args.iters)
         flops = 2 * M * 280 * 2048 * 2
         print(f"fused_ffn M={M}: {us:.1f} us ({flops / us / 1e6:.0f} GF/s)")
+        w1v2 = torch.randn(2048, 296, device="cuda").to(torch.bfloat16) * 0.05
+        us3 = timeit(lambda: ext.fused_ffn_v2(x, w1v2, w2, b2, 0.5),
+                     args.iters)
+        print(f"fused_ffn_v2 M={M}: {us3:.1f} us "
+              f"({flops / us3 / 1e6:.0f} GF/s)")
         # hipBLASLt pair for comparison.
         w1t = torch.randn(280, 2048, device="cuda").to(torch.bfloat16)
         w2t = torch.randn(2048, 280, device="cuda").to(torch.bfloat16)
