@@ -48,6 +48,7 @@ def train_model(
     limit_steps: int = 0,
     write_checkpoint_metrics: bool = True,
     warm_start: Optional[str] = None,
+    use_bf16: bool = False,
 ) -> dict:
     """Runs the custom training loop; returns summary metrics."""
     rank, world = comm.init_distributed()
@@ -137,7 +138,14 @@ def train_model(
             lr = schedule.apply(optimizer, step)
             rows, label = _prepare_batch(batch, device)
             reducer.zero_()
-            probs = model(rows, training=True)
+            # bf16 autocast (BASELINE config #4): GEMMs/attention run bf16,
+            # LN/softmax and the alignment loss stay fp32; fp32 master
+            # weights and fp32 gradient all-reduce.
+            if use_bf16 and rows.is_cuda:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    probs = model(rows, training=True)
+            else:
+                probs = model(rows, training=True)
             # compute_average_loss: sum / global batch
             # (model_train_custom_loop.py:148-154).
             loss = loss_fn(label, probs.float()) / global_batch
@@ -211,6 +219,8 @@ def main(argv: Optional[List[str]] = None) -> None:
     ap.add_argument("--checkpoint", default=None,
                     help="warm-start checkpoint")
     ap.add_argument("--device", default=None)
+    ap.add_argument("--bf16", action="store_true",
+                    help="autocast forward to bf16 (fp32 loss/LN/softmax)")
     args = ap.parse_args(argv)
 
     params = cfg.get_config(args.params)
@@ -239,7 +249,7 @@ def main(argv: Optional[List[str]] = None) -> None:
             train_model(
                 args.out_dir, params, device=device,
                 eval_every=args.eval_every, limit_steps=args.limit_steps,
-                warm_start=args.checkpoint,
+                warm_start=args.checkpoint, use_bf16=args.bf16,
             )
             break
         except RuntimeError as e:

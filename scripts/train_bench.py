@@ -29,6 +29,7 @@ def main():
     ap.add_argument("--batch-size", type=int, default=256)
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--bf16", action="store_true")
     args = ap.parse_args()
 
     rank, world = comm.init_distributed()
@@ -68,7 +69,11 @@ def main():
     def step(i):
         schedule.apply(optimizer, i)
         reducer.zero_()
-        probs = model(x, training=True)
+        if args.bf16 and have_cuda:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                probs = model(x, training=True)
+        else:
+            probs = model(x, training=True)
         loss = loss_fn(label, probs.float()) / global_batch
         loss.backward()
         reducer.reduce()

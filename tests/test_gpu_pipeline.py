@@ -76,3 +76,19 @@ def test_trained_checkpoint_serves_gpu(tmp_path):
         output=out, options=options, device="cuda",
     )
     assert counter.total == 2
+
+
+def test_train_bf16_gpu(tmp_path):
+    """bf16-autocast training step runs and converges sanely on cuda."""
+    from deepconsensus_amd.models import train as train_lib
+
+    train_file, _ = make_training_data(tmp_path)
+    params = _tiny_params(train_file)
+    out_dir = str(tmp_path / "model_bf16")
+    summary = train_lib.train_model(
+        out_dir, params, device="cuda", eval_every=100, limit_steps=2,
+        use_bf16=True,
+    )
+    assert summary["steps"] >= 2
+    import math
+    assert math.isfinite(summary["eval/loss"])
