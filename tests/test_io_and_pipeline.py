@@ -316,3 +316,21 @@ def test_quick_inference_sharding(tmp_path):
     assert names[0] and names[1]
     assert not (names[0] & names[1])
     assert len(names[0] | names[1]) == 4
+
+
+def test_quick_inference_multiprocess_cpus(tmp_path):
+    """cpus=2 routes preprocessing through a ProcessPoolExecutor (pickling
+    of BamRead/Read across process boundaries)."""
+    from deepconsensus_amd.inference import quick_inference as qi
+    from deepconsensus_amd.dcio.fastq import read_fastq
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=3, length=150, seed=21)
+    out = str(tmp_path / "out.fastq")
+    options = qi.InferenceOptions(
+        batch_size=8, batch_zmws=2, cpus=2, min_quality=0,
+        skip_windows_above=0,
+    )
+    counter = qi.run(subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+                     output=out, options=options, device="cpu")
+    assert counter.total == 3
+    assert len(list(read_fastq(out))) == 3
