@@ -46,6 +46,12 @@ def _run_main(argv: List[str]) -> None:
     ap.add_argument("--shard", default=None,
                     help="'i/N': process only ZMWs with index %% N == i "
                     "(one process per GPU, like the reference's ccs --chunk)")
+    ap.add_argument("--end_after_stage", default="full",
+                    choices=[s.name.lower() for s in qi.DebugStage],
+                    help="stop after this stage (debug/runtime testing)")
+    ap.add_argument("--use_only_gpu_index", type=int, default=None,
+                    help="pin inference to this GPU index "
+                    "(shorthand for --device cuda:N)")
     args = ap.parse_args(argv)
 
     options = qi.InferenceOptions(
@@ -58,10 +64,14 @@ def _run_main(argv: List[str]) -> None:
         skip_windows_above=args.skip_windows_above,
         ins_trim=args.ins_trim,
         use_ccs_smart_windows=args.use_ccs_smart_windows,
+        end_after_stage=qi.DebugStage[args.end_after_stage.upper()],
     )
     if args.shard:
         i, n = args.shard.split("/")
         options.shard_index, options.shard_count = int(i), int(n)
+    device = args.device
+    if args.use_only_gpu_index is not None:
+        device = f"cuda:{args.use_only_gpu_index}"
     outcome = qi.run(
         subreads_to_ccs=args.subreads_to_ccs,
         ccs_bam=args.ccs_bam,
@@ -71,9 +81,9 @@ def _run_main(argv: List[str]) -> None:
         limit=args.limit,
         dc_calibration=args.dc_calibration,
         ccs_calibration=args.ccs_calibration,
-        device=args.device,
+        device=device,
     )
-    if outcome.success == 0:
+    if outcome.success == 0 and options.end_after_stage == qi.DebugStage.FULL:
         sys.exit(1)
 
 

@@ -18,6 +18,7 @@ from __future__ import annotations
 import collections
 import concurrent.futures
 import dataclasses
+import enum
 import itertools
 import json
 import logging
@@ -40,6 +41,17 @@ from deepconsensus_amd.preprocess.windows import DcConfig
 from deepconsensus_amd.utils import constants, phred
 
 log = logging.getLogger(__name__)
+
+
+@enum.unique
+class DebugStage(enum.Enum):
+    """Stage to end after, for debugging and runtime testing
+    (quick_inference.py:68-75)."""
+
+    DC_INPUT = 1
+    TF_EXAMPLES = 2
+    RUN_MODEL = 3
+    FULL = 4
 
 
 @dataclasses.dataclass
@@ -78,17 +90,23 @@ class InferenceOptions:
     # this process handles ZMWs with index % N == i.
     shard_index: int = 0
     shard_count: int = 1
+    # Stop the pipeline early for debugging/runtime testing
+    # (quick_inference.py:501,559,715).
+    end_after_stage: DebugStage = DebugStage.FULL
 
 
 def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
     """Windows + counters for one ZMW (quick_inference.py:535-564)."""
-    zmw, subreads, dc_config, window_widths = one_zmw
+    zmw, subreads, dc_config, window_widths, *rest = one_zmw
+    stage = rest[0] if rest else DebugStage.FULL
     dc_whole = pre_feeder.subreads_to_dc_example(
         subreads=subreads,
         ccs_seqname=zmw,
         dc_config=dc_config,
         window_widths=window_widths,
     )
+    if stage == DebugStage.DC_INPUT:
+        return [], dc_whole.counter
     feature_dicts = [x.to_features_dict() for x in dc_whole.iter_examples()]
     return feature_dicts, dc_whole.counter
 
@@ -341,7 +359,8 @@ def run(
             idx += 1
             if not keep:
                 continue
-            batch.append((zmw, subreads, dcc, window_widths))
+            batch.append((zmw, subreads, dcc, window_widths,
+                          options.end_after_stage))
             if len(batch) >= options.batch_zmws:
                 yield batch
                 batch = []
@@ -361,9 +380,12 @@ def run(
         for counter in counters:
             stats_counter.update(counter)
         n_examples = sum(len(z) for z in feature_dicts_for_zmws)
-        n_subreads = sum(len(s) for _, s, _, _ in inputs)
+        n_subreads = sum(len(z[1]) for z in inputs)
         timelog.add("preprocess", batch_name, before, n_examples,
                     n_subreads, len(inputs))
+        if options.end_after_stage in (DebugStage.DC_INPUT,
+                                       DebugStage.TF_EXAMPLES):
+            return
 
         before = time.time()
         for_model: List[Dict[str, Any]] = []
@@ -389,6 +411,8 @@ def run(
         preds.extend(skipped)
         timelog.add("run_model", batch_name, before, n_examples,
                     n_subreads, len(inputs))
+        if options.end_after_stage == DebugStage.RUN_MODEL:
+            return
 
         before = time.time()
         _write_outputs(preds, output_writer, bam_out, options,

@@ -334,3 +334,26 @@ def test_quick_inference_multiprocess_cpus(tmp_path):
                      output=out, options=options, device="cpu")
     assert counter.total == 3
     assert len(list(read_fastq(out))) == 3
+
+
+@pytest.mark.parametrize("stage,expect_fastq", [
+    ("dc_input", False), ("tf_examples", False), ("run_model", False),
+])
+def test_run_end_after_stage(tmp_path, stage, expect_fastq):
+    """--end_after_stage stops the pipeline early (reference DebugStage)."""
+    from deepconsensus_amd import cli
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=2, length=150, seed=5)
+    out = str(tmp_path / "out.fastq")
+    cli.main(["run", "--subreads_to_ccs", sub, "--ccs_bam", ccs,
+              "--checkpoint", "random", "--output", out,
+              "--batch_size", "8", "--min_quality", "0",
+              "--skip_windows_above", "0", "--device", "cpu",
+              "--end_after_stage", stage])
+    fq = open(out).read()
+    assert bool(fq.strip()) == expect_fastq
+    # runtime CSV is still written with the stages that ran
+    runtime = open(str(tmp_path / "out.runtime.csv")).read()
+    assert "preprocess" in runtime
+    if stage in ("dc_input", "tf_examples"):
+        assert "run_model" not in runtime
