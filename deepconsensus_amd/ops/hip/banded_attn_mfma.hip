@@ -49,8 +49,6 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
     const bf16* __restrict__ qkv, bf16* __restrict__ out,
     int B, int L, int H, int win, float scale) {
   constexpr int D = AM_D;
-  const int b = blockIdx.x / H;
-  const int h = blockIdx.x % H;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -61,146 +59,221 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
   __shared__ __attribute__((aligned(16))) bf16 vt_lds[V_ROWS][V_STRIDE];
 
   const size_t RS = (size_t)3 * H * D;  // qkv row stride (elems)
-  const bf16* base = qkv + (size_t)b * L * RS + (size_t)h * D;
 
-  // ---- Stage K row-major (dims 140..151 zeroed). ----
-  for (int idx = tid; idx < L * (K_STRIDE / 2); idx += 256) {
-    const int r = idx / (K_STRIDE / 2), d2 = idx % (K_STRIDE / 2);
-    unsigned v = 0;
-    if (2 * d2 + 1 < D) {
-      v = *reinterpret_cast<const unsigned*>(base + r * RS + H * D + 2 * d2);
-    }
-    *reinterpret_cast<unsigned*>(&k_lds[r][2 * d2]) = v;
-  }
-  // ---- Stage V transposed: V[key][dim] -> vt_lds[dim][V_OFF + key]. ----
-  for (int idx = tid; idx < L * 70; idx += 256) {
-    const int r = idx / 70, d2 = idx % 70;
-    const unsigned v = *reinterpret_cast<const unsigned*>(
-        base + r * RS + 2 * H * D + 2 * d2);
-    vt_lds[2 * d2][win + r] =
-        __ushort_as_bfloat16((unsigned short)(v & 0xffffu));
-    if (2 * d2 + 1 < D)
-      vt_lds[2 * d2 + 1][win + r] =
-          __ushort_as_bfloat16((unsigned short)(v >> 16));
-  }
-  // Zero-fill: key slots [0, V_OFF) and [V_OFF+L, V_STRIDE) on dims < D,
-  // and all slots on dims >= D (odd dim 141 is also written above w/ zero).
-  for (int d = 0; d < V_ROWS; ++d) {
-    for (int s = tid; s < V_STRIDE; s += 256) {
-      const bool pad_row = d >= D;
-      const bool pad_col = (s < win) || (s >= win + L);
-      if (pad_row || pad_col)
-        vt_lds[d][s] = __ushort_as_bfloat16((unsigned short)0);
-    }
+  // One-time zero of the whole Vt image: staging below only ever writes the
+  // live slots [d < D][win .. win+L), so pad rows/cols stay zero across all
+  // persistent-loop items (L, win are uniform).
+  for (int idx = tid; idx < V_ROWS * V_STRIDE / 8; idx += 256) {
+    *(reinterpret_cast<bf16x8*>(&vt_lds[0][0]) + idx) = bf16x8{};
   }
 
-  // ---- Load Q B-fragments (this lane's query row, 9 k-steps of 16). ----
-  const int l0w = 32 * wave;
-  const int qrow = l0w + c;
-  bf16x8 qf[9];
-  {
-    const bool qv = qrow < L;
-    const unsigned short* qp =
-        reinterpret_cast<const unsigned short*>(base + (size_t)qrow * RS);
+  // K/V staging registers for the item-level prefetch pipeline: the NEXT
+  // item's 16-B granules are issued during the CURRENT item's softmax/PV,
+  // so their latency hides behind compute. NAMED scalars — an array here
+  // would be silently promoted to LDS (AMDGPUPromoteAlloca).
+  uint4 kr0, kr1, kr2, kr3, kr4, kr5, kr6, kr7;
+  uint4 vr0, vr1, vr2, vr3, vr4, vr5, vr6, vr7;
+  const int n_kg = L * 19;  // K granules: 19 x 16 B per row (stride 304 B)
+  const int n_vg = L * 18;  // V granules: 8 dims of one key each
+
+  // Issue the global loads for item `it`'s K and V into the staging
+  // registers. Tail granules (dims 136..139) take the 2-dword path; the
+  // clamped duplicate granule on inactive threads is benign (same value
+  // rewritten). Dims >= D stay zero.
+  auto issue_kv = [&](int it) {
+    const bf16* bs = qkv + (size_t)(it / H) * L * RS + (size_t)(it % H) * D;
 #pragma unroll
-    for (int s = 0; s < 9; ++s) {
-      const int d0 = 16 * s + 8 * hi;
-      bf16x8 t = {};
-      if (qv) {
-        if (d0 + 8 <= D) {
-          t = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<const bf16*>(qp) + d0);
-        } else {
-          unsigned short* tw = reinterpret_cast<unsigned short*>(&t);
-          for (int j = 0; j < 8; ++j) tw[j] = (d0 + j < D) ? qp[d0 + j] : 0;
-        }
+    for (int i = 0; i < 8; ++i) {
+      const int idx = min(tid + i * 256, n_kg - 1);
+      const int r = idx / 19, q4 = idx % 19;
+      uint4 v = {};
+      const bf16* src = bs + r * RS + H * D + 8 * q4;
+      if (8 * q4 + 8 <= D) {
+        v = *reinterpret_cast<const uint4*>(src);
+      } else if (8 * q4 < D) {
+        const unsigned* p = reinterpret_cast<const unsigned*>(src);
+        v.x = p[0];
+        v.y = p[1];
       }
-      qf[s] = t;
-    }
-  }
-  __syncthreads();
-
-  // ---- QK^T (swapped): S^T tiles; lane holds qrow=l0w+c, keys in regs. ----
-  const int kw0 = l0w - win;  // key window start (may be negative)
-  float st[32];
-#pragma unroll
-  for (int t = 0; t < 2; ++t) {
-    f32x16 acc = {};
-    const int krow = min(max(kw0 + 32 * t + c, 0), L - 1);
-#pragma unroll
-    for (int s = 0; s < 9; ++s) {
-      const bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(&k_lds[krow][16 * s + 8 * hi]);
-      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[s], acc, 0, 0, 0);
+      if (i == 0) kr0 = v; else if (i == 1) kr1 = v;
+      else if (i == 2) kr2 = v; else if (i == 3) kr3 = v;
+      else if (i == 4) kr4 = v; else if (i == 5) kr5 = v;
+      else if (i == 6) kr6 = v; else kr7 = v;
     }
 #pragma unroll
-    for (int r = 0; r < 16; ++r) st[16 * t + r] = acc[r];
-  }
+    for (int i = 0; i < 8; ++i) {
+      const int idx = min(tid + i * 256, n_vg - 1);
+      const int r = idx / 18, d0 = 8 * (idx % 18);
+      uint4 v = {};
+      const bf16* src = bs + r * RS + 2 * H * D + d0;
+      if (d0 + 8 <= D) {
+        v = *reinterpret_cast<const uint4*>(src);
+      } else {
+        const unsigned* p = reinterpret_cast<const unsigned*>(src);
+        v.x = p[0];
+        v.y = p[1];
+      }
+      if (i == 0) vr0 = v; else if (i == 1) vr1 = v;
+      else if (i == 2) vr2 = v; else if (i == 3) vr3 = v;
+      else if (i == 4) vr4 = v; else if (i == 5) vr5 = v;
+      else if (i == 6) vr6 = v; else vr7 = v;
+    }
+  };
+  // Drain the staging registers into LDS. K rows are 16-B-aligned vector
+  // writes; V scatters 8 ds_write_b16 down a Vt column (the tail granule's
+  // zero high half rewrites pad rows 140..143 with zeros, which is their
+  // required value).
+  auto write_kv = [&]() {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int idx = min(tid + i * 256, n_kg - 1);
+      const int r = idx / 19, q4 = idx % 19;
+      *reinterpret_cast<uint4*>(&k_lds[r][8 * q4]) =
+          (i == 0 ? kr0 : i == 1 ? kr1 : i == 2 ? kr2 : i == 3 ? kr3
+           : i == 4 ? kr4 : i == 5 ? kr5 : i == 6 ? kr6 : kr7);
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int idx = min(tid + i * 256, n_vg - 1);
+      const int r = idx / 18, d0 = 8 * (idx % 18);
+      const uint4 v =
+          (i == 0 ? vr0 : i == 1 ? vr1 : i == 2 ? vr2 : i == 3 ? vr3
+           : i == 4 ? vr4 : i == 5 ? vr5 : i == 6 ? vr6 : vr7);
+      const unsigned short* vs = reinterpret_cast<const unsigned short*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[d0 + j][win + r] = __ushort_as_bfloat16(vs[j]);
+    }
+  };
 
-  // ---- Masked softmax, in-lane + one half-merge. ----
-  // Key-local index of st[i]: kl(i) = 32*(i/16) + (i&3) + 8*((i&15)>>2) + 4*hi.
-  float m = -1e30f;
-#pragma unroll
-  for (int i = 0; i < 32; ++i) {
-    const int kl = 32 * (i >> 4) + (i & 3) + 8 * ((i & 15) >> 2) + 4 * hi;
-    const int kg = kw0 + kl;
-    const bool valid =
-        (kl >= c) && (kl <= c + 2 * win) && (kg >= 0) && (kg < L);
-    st[i] = valid ? st[i] * scale : -1e30f;
-    m = fmaxf(m, st[i]);
-  }
-  m = fmaxf(m, __shfl_xor(m, 32, 64));
-  float denom = 0.f;
-#pragma unroll
-  for (int i = 0; i < 32; ++i) {
-    st[i] = (st[i] <= -1e29f) ? 0.f : __expf(st[i] - m);
-    denom += st[i];
-  }
-  denom += __shfl_xor(denom, 32, 64);
-  const float inv = (denom > 0.f) ? 1.f / denom : 0.f;
-#pragma unroll
-  for (int i = 0; i < 32; ++i) st[i] *= inv;
+  // Persistent CTA: each resident block walks many (b, h) items. The
+  // one-shot version (grid = B*H tiny blocks) was launch/drain-bound: 74%
+  // of wave-slots parked, ~20 us wall per ~1 us of math.
+  const int BH = B * H;
+  if (blockIdx.x < BH) issue_kv(blockIdx.x);
+  for (int item = blockIdx.x; item < BH; item += gridDim.x) {
+    const bf16* base =
+        qkv + (size_t)(item / H) * L * RS + (size_t)(item % H) * D;
+    __syncthreads();  // prior item's LDS reads done before re-staging
 
-  // ---- Repack P to A-fragments: 4 k-steps x 4 dwords (T12 pattern).
-  // Lane pair (l, l+32) holds interleaved keys; after cvt_pk + two
-  // permlane32_swaps per k-step, lane l's fragment covers its 8 contiguous
-  // keys 16s+8*hi .. +7.
-  bf16x8 pa[4];
+    // ---- Issue Q B-fragment loads (this lane's query row, 9 k-steps of
+    // 16) before the LDS drain so their latency hides behind it. ----
+    const int l0w = 32 * wave;
+    const int qrow = l0w + c;
+    bf16x8 qf[9];
+    {
+      const bool qv = qrow < L;
+      const unsigned short* qp =
+          reinterpret_cast<const unsigned short*>(base + (size_t)qrow * RS);
 #pragma unroll
-  for (int s = 0; s < 4; ++s) {
-    const unsigned x = cvt_pk_bf16(st[8 * s + 0], st[8 * s + 1]);
-    const unsigned y = cvt_pk_bf16(st[8 * s + 2], st[8 * s + 3]);
-    const unsigned x2 = cvt_pk_bf16(st[8 * s + 4], st[8 * s + 5]);
-    const unsigned y2 = cvt_pk_bf16(st[8 * s + 6], st[8 * s + 7]);
-    const auto rx = __builtin_amdgcn_permlane32_swap(x, x2, false, false);
-    const auto ry = __builtin_amdgcn_permlane32_swap(y, y2, false, false);
-    unsigned u[4] = {(unsigned)rx[0], (unsigned)ry[0], (unsigned)rx[1],
-                     (unsigned)ry[1]};
-    pa[s] = *reinterpret_cast<const bf16x8*>(u);
-  }
+      for (int s = 0; s < 9; ++s) {
+        const int d0 = 16 * s + 8 * hi;
+        bf16x8 t = {};
+        if (qv) {
+          if (d0 + 8 <= D) {
+            t = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const bf16*>(qp) + d0);
+          } else {
+            unsigned short* tw = reinterpret_cast<unsigned short*>(&t);
+            for (int j = 0; j < 8; ++j) tw[j] = (d0 + j < D) ? qp[d0 + j] : 0;
+          }
+        }
+        qf[s] = t;
+      }
+    }
+    write_kv();
+    __syncthreads();
 
-  // ---- PV: out[32 qrows x 140] in 5 col tiles. ----
-  // B-frag: Vt[vdim][win + kw0 + 16s + 8*hi ... +7] = Vt[vdim][32w + 16s
-  // + 8*hi ...], which is >= 0 and == 0 (mod 8): reads stay 16-B aligned.
-  bf16* ob = out + ((size_t)b * L) * (H * D) + (size_t)h * D;
+    // ---- QK^T (swapped): S^T tiles; lane holds qrow=l0w+c, keys in regs. ----
+    const int kw0 = l0w - win;  // key window start (may be negative)
+    float st[32];
 #pragma unroll
-  for (int ct = 0; ct < 5; ++ct) {
-    f32x16 acc = {};
-    const int vdim = min(32 * ct + c, V_ROWS - 1);
+    for (int t = 0; t < 2; ++t) {
+      f32x16 acc = {};
+      const int krow = min(max(kw0 + 32 * t + c, 0), L - 1);
+#pragma unroll
+      for (int s = 0; s < 9; ++s) {
+        const bf16x8 a =
+            *reinterpret_cast<const bf16x8*>(&k_lds[krow][16 * s + 8 * hi]);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[s], acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[16 * t + r] = acc[r];
+    }
+
+    // Prefetch the next item's K/V now: the loads fly during softmax + PV
+    // and are drained into LDS only after the top-of-loop barrier.
+    const int next_item = item + gridDim.x;
+    if (next_item < BH) issue_kv(next_item);
+
+    // ---- Masked softmax, in-lane + one half-merge. ----
+    // Key-local index of st[i]: kl(i) = 32*(i/16) + (i&3) + 8*((i&15)>>2)
+    // + 4*hi.
+    float m = -1e30f;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) {
+      const int kl = 32 * (i >> 4) + (i & 3) + 8 * ((i & 15) >> 2) + 4 * hi;
+      const int kg = kw0 + kl;
+      const bool valid =
+          (kl >= c) && (kl <= c + 2 * win) && (kg >= 0) && (kg < L);
+      st[i] = valid ? st[i] * scale : -1e30f;
+      m = fmaxf(m, st[i]);
+    }
+    m = fmaxf(m, __shfl_xor(m, 32, 64));
+    float denom = 0.f;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) {
+      st[i] = (st[i] <= -1e29f) ? 0.f : __expf(st[i] - m);
+      denom += st[i];
+    }
+    denom += __shfl_xor(denom, 32, 64);
+    const float inv = (denom > 0.f) ? 1.f / denom : 0.f;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) st[i] *= inv;
+
+    // ---- Repack P to A-fragments: 4 k-steps x 4 dwords (T12 pattern).
+    // Lane pair (l, l+32) holds interleaved keys; after cvt_pk + two
+    // permlane32_swaps per k-step, lane l's fragment covers its 8 contiguous
+    // keys 16s+8*hi .. +7.
+    bf16x8 pa[4];
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      const int kk = win + kw0 + 16 * s + 8 * hi;
-      const bf16x8 bfrag =
-          *reinterpret_cast<const bf16x8*>(&vt_lds[vdim][kk]);
-      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s], bfrag, acc, 0, 0, 0);
+      const unsigned x = cvt_pk_bf16(st[8 * s + 0], st[8 * s + 1]);
+      const unsigned y = cvt_pk_bf16(st[8 * s + 2], st[8 * s + 3]);
+      const unsigned x2 = cvt_pk_bf16(st[8 * s + 4], st[8 * s + 5]);
+      const unsigned y2 = cvt_pk_bf16(st[8 * s + 6], st[8 * s + 7]);
+      const auto rx = __builtin_amdgcn_permlane32_swap(x, x2, false, false);
+      const auto ry = __builtin_amdgcn_permlane32_swap(y, y2, false, false);
+      unsigned u[4] = {(unsigned)rx[0], (unsigned)ry[0], (unsigned)rx[1],
+                       (unsigned)ry[1]};
+      pa[s] = *reinterpret_cast<const bf16x8*>(u);
     }
-    const int col = 32 * ct + c;
-    if (col < D) {
+
+    // ---- PV: out[32 qrows x 140] in 5 col tiles. ----
+    // B-frag: Vt[vdim][win + kw0 + 16s + 8*hi ... +7] = Vt[vdim][32w + 16s
+    // + 8*hi ...], which is >= 0 and == 0 (mod 8): reads stay 16-B aligned.
+    bf16* ob =
+        out + (size_t)(item / H) * L * (H * D) + (size_t)(item % H) * D;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int qr = l0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        if (qr < L)
-          ob[(size_t)qr * H * D + col] = __float2bfloat16(acc[r]);
+    for (int ct = 0; ct < 5; ++ct) {
+      f32x16 acc = {};
+      const int vdim = min(32 * ct + c, V_ROWS - 1);
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int kk = win + kw0 + 16 * s + 8 * hi;
+        const bf16x8 bfrag =
+            *reinterpret_cast<const bf16x8*>(&vt_lds[vdim][kk]);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s], bfrag, acc,
+                                                      0, 0, 0);
+      }
+      const int col = 32 * ct + c;
+      if (col < D) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qr = l0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          if (qr < L)
+            ob[(size_t)qr * H * D + col] = __float2bfloat16(acc[r]);
+        }
       }
     }
   }
@@ -219,7 +292,8 @@ at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
               "banded_attn_mfma requires D=140, 32<=L<=104");
   TORCH_CHECK(win <= 12 && win >= 1, "win must be in [1, 12]");
   auto out = at::empty({B, L, H * D}, q.options());
-  dim3 grid(B * H);
+  // Persistent grid: 2 blocks resident per CU (80 KB LDS each) on 256 CUs.
+  dim3 grid(std::min(B * (int)H, 512));
   dim3 block(256);
   hipStream_t stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(banded_attn_mfma_kernel, grid, block, 0, stream,
