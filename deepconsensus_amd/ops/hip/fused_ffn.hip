@@ -72,14 +72,16 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_kernel(
   const int m0 = blockIdx.x * BM;
 
   // ---- Stage x image over smem_w ([128][296] rows), pull A-frags. ----
-  for (int idx = tid; idx < BM * (W1_STRIDE / 2); idx += 512) {
-    const int r = idx / (W1_STRIDE / 2), d2 = idx % (W1_STRIDE / 2);
-    unsigned v = 0;
-    if (m0 + r < M && 2 * d2 + 1 < K1) {
-      v = *reinterpret_cast<const unsigned*>(
-          x + (size_t)(m0 + r) * K1 + 2 * d2);
+  // 16-B granules (37/row; row stride 592 B = 37 x 16 B): keeps the x loads
+  // in flight instead of serializing ~37 scalar-load latencies.
+  for (int idx = tid; idx < BM * 37; idx += 512) {
+    const int r = idx / 37, q4 = idx % 37;
+    uint4 v = {};
+    if (m0 + r < M && 8 * q4 + 8 <= K1) {
+      v = *reinterpret_cast<const uint4*>(
+          x + (size_t)(m0 + r) * K1 + 8 * q4);
     }
-    *reinterpret_cast<unsigned*>(&smem_w[r * W1_STRIDE + 2 * d2]) = v;
+    *reinterpret_cast<uint4*>(&smem_w[r * W1_STRIDE + 8 * q4]) = v;
   }
   __syncthreads();
   bf16x8 af[18];

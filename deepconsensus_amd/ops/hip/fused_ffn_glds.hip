@@ -74,17 +74,19 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_v2_kernel(
 
   // ---- Stage x image over the W1 double-buffer region (exact fit):
   // cols 280..286 zero, col 287 = bf16(1.0), rows >= M zero. ----
-  for (int idx = tid; idx < BM * (K1P / 2); idx += 512) {
-    const int r = idx / (K1P / 2), d2 = idx % (K1P / 2);
-    unsigned v = 0;
+  // 16-B granules (37/row; row stride 592 B): granule 35 carries the
+  // bias-column constant — elems 280..287 with bf16(1.0) at col 287.
+  for (int idx = tid; idx < BM * 37; idx += 512) {
+    const int r = idx / 37, q4 = idx % 37;
+    uint4 v = {};
     const bool rv = (m0 + r) < M;
-    if (rv && 2 * d2 + 1 < K1) {
-      v = *reinterpret_cast<const unsigned*>(
-          x + (size_t)(m0 + r) * K1 + 2 * d2);
-    } else if (rv && 2 * d2 + 1 == BIAS_COL) {
-      v = 0x3f800000u;  // upper half bf16(1.0) at col 287, lower col 286 = 0
+    if (rv && 8 * q4 + 8 <= K1) {
+      v = *reinterpret_cast<const uint4*>(
+          x + (size_t)(m0 + r) * K1 + 8 * q4);
+    } else if (rv && 8 * q4 + 7 == BIAS_COL) {
+      v.w = 0x3f800000u;  // upper half bf16(1.0) at col 287
     }
-    *reinterpret_cast<unsigned*>(&smem[r * K1P + 2 * d2]) = v;
+    *reinterpret_cast<uint4*>(&smem[r * K1P + 8 * q4]) = v;
   }
   __syncthreads();
   bf16x8 af[18];

@@ -55,15 +55,16 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
   bf16x8 af[18];
   for (int half = 0; half < 2; ++half) {
     __syncthreads();
-    for (int idx = tid; idx < 64 * (W_STRIDE / 2); idx += 512) {
-      const int r = idx / (W_STRIDE / 2), d2 = idx % (W_STRIDE / 2);
+    // 16-B granules (37/row; W_STRIDE 296 elems = 592 B = 37 x 16 B).
+    for (int idx = tid; idx < 64 * 37; idx += 512) {
+      const int r = idx / 37, q4 = idx % 37;
       const int row = 64 * half + r;
-      unsigned v = 0;
-      if (m0 + row < M && 2 * d2 + 1 < K1) {
-        v = *reinterpret_cast<const unsigned*>(
-            x + (size_t)(m0 + row) * K1 + 2 * d2);
+      uint4 v = {};
+      if (m0 + row < M && 8 * q4 + 8 <= K1) {
+        v = *reinterpret_cast<const uint4*>(
+            x + (size_t)(m0 + row) * K1 + 8 * q4);
       }
-      *reinterpret_cast<unsigned*>(&w_lds[r * W_STRIDE + 2 * d2]) = v;
+      *reinterpret_cast<uint4*>(&w_lds[r * W_STRIDE + 8 * q4]) = v;
     }
     __syncthreads();
     // Waves whose row group sits in this half pull their fragments.

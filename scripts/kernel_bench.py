@@ -70,6 +70,37 @@ def main():
         print(f"hipblaslt pair M={M}: {us2:.1f} us "
               f"({flops / us2 / 1e6:.0f} GF/s)")
 
+    if args.kernel in ("all", "linear"):
+        M = B * L
+        x = torch.randn(M, 280, device="cuda").to(torch.bfloat16) * 0.3
+        wqkv = torch.randn(896, 288, device="cuda").to(torch.bfloat16) * 0.05
+        wout = torch.randn(320, 288, device="cuda").to(torch.bfloat16) * 0.05
+        empty = x.new_empty(0)
+        us = timeit(lambda: ext.fused_linear(x, wqkv, empty, empty, 840,
+                                             False, 0.0), args.iters)
+        flops = 2 * M * 280 * 840
+        print(f"fused_linear qkv M={M}: {us:.1f} us "
+              f"({flops / us / 1e6:.0f} GF/s)")
+        a = torch.randn(M, 280, device="cuda").to(torch.bfloat16) * 0.3
+        us = timeit(lambda: ext.fused_linear(a, wout, empty, x, 280,
+                                             False, 0.07), args.iters)
+        flops = 2 * M * 280 * 280
+        print(f"fused_linear out+resid M={M}: {us:.1f} us "
+              f"({flops / us / 1e6:.0f} GF/s)")
+
+    if args.kernel in ("all", "lnhead"):
+        M = B * L
+        x = torch.randn(M, 280, device="cuda").to(torch.bfloat16)
+        g = torch.randn(280, device="cuda")
+        be = torch.randn(280, device="cuda")
+        wh = torch.randn(5, 280, device="cuda") * 0.05
+        bh = torch.randn(5, device="cuda")
+        us = timeit(lambda: ext.fused_ln_head_qv(
+            x, g, be, wh, bh, 0.0, 1.197654, -0.99781, 93.0, False),
+            args.iters)
+        gb = M * (280 * 2 + 2) / 1e9
+        print(f"fused_ln_head_qv M={M}: {us:.1f} us ({gb / (us / 1e6):.2f} GB/s)")
+
     if args.kernel in ("all", "embed"):
         import numpy as np
 
