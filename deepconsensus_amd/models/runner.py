@@ -215,6 +215,9 @@ class InferenceRunner:
                 and _os.environ.get("DC_FUSED_FFN") != "0"
             )
             self.ffn_v2 = _os.environ.get("DC_FFN_V2", "1") != "0"
+            # v3 (256-row tiles, register-resident h) measures fastest;
+            # DC_FFN_V3=0 drops back to v2/v1.
+            self.ffn_v3 = _os.environ.get("DC_FFN_V3", "1") != "0"
 
             if self.ffn_fused_ok:
                 for i, l in enumerate(model.layers):
@@ -310,7 +313,12 @@ class InferenceRunner:
                         a.view(b * l, h), lw["wout_pad"], empty, flat,
                         280, False, lw["alpha_attn"],
                     )
-                    if self.ffn_v2:
+                    if self.ffn_v3:
+                        flat = self.ext.fused_ffn_v3(
+                            flat, lw["w1_v2"], lw["w2_pad"],
+                            lw["b2_f32"], lw["alpha_ffn"],
+                        )
+                    elif self.ffn_v2:
                         flat = self.ext.fused_ffn_v2(
                             flat, lw["w1_v2"], lw["w2_pad"],
                             lw["b2_f32"], lw["alpha_ffn"],

@@ -180,6 +180,8 @@ at::Tensor fused_ffn(at::Tensor x, at::Tensor w1, at::Tensor b1,
                      at::Tensor w2, at::Tensor b2, double alpha);
 at::Tensor fused_ffn_v2(at::Tensor x, at::Tensor w1, at::Tensor w2,
                         at::Tensor b2, double alpha);
+at::Tensor fused_ffn_v3(at::Tensor x, at::Tensor w1, at::Tensor w2,
+                        at::Tensor b2, double alpha);
 at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
                         at::Tensor resid, int64_t n_out, bool relu,
                         double alpha);
@@ -204,6 +206,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused FFN + ReZero residual (K8+K9), hidden tensor LDS-resident");
   m.def("fused_ffn_v2", &fused_ffn_v2,
         "Fused FFN v2: glds-pipelined weights, bias folded into W1");
+  m.def("fused_ffn_v3", &fused_ffn_v3,
+        "Fused FFN v3: 256-row tiles, register-resident h (swapped B1)");
   m.def("fused_linear", &fused_linear,
         "Fused linear projection (K5/K7): act(xW^T+b)[*alpha+x]");
   m.def("alignment_dp_fwd", &alignment_dp_fwd,

@@ -56,6 +56,10 @@ def main():
                      args.iters)
         print(f"fused_ffn_v2 M={M}: {us3:.1f} us "
               f"({flops / us3 / 1e6:.0f} GF/s)")
+        us4 = timeit(lambda: ext.fused_ffn_v3(x, w1v2, w2, b2, 0.5),
+                     args.iters)
+        print(f"fused_ffn_v3 M={M}: {us4:.1f} us "
+              f"({flops / us4 / 1e6:.0f} GF/s)")
         # hipBLASLt pair for comparison.
         w1t = torch.randn(280, 2048, device="cuda").to(torch.bfloat16)
         w2t = torch.randn(2048, 280, device="cuda").to(torch.bfloat16)

@@ -430,3 +430,36 @@ def test_fused_ffn_v2_matches_torch(setup):
         scale = ref.abs().mean().clamp_min(1e-3)
         assert (err.mean() / scale).item() < 0.02, (M, err.mean().item())
         assert err.max().item() < 0.3, (M, err.max().item())
+
+
+@pytest.mark.gpu
+def test_fused_ffn_v3_matches_torch(setup):
+    """256-row-tile register-resident-h FFN v3 vs fp32 torch reference."""
+    params, model, runner, rows = setup
+    torch.manual_seed(29)
+    l = model.layers[2]
+    w1 = l.ffn.filter_layer.weight.detach().float()
+    b1 = l.ffn.filter_layer.bias.detach().float()
+    w1v2 = torch.zeros(2048, 296)
+    w1v2[:, :280] = w1
+    w1v2[:, 287] = b1
+    w1v2 = w1v2.to(torch.bfloat16).cuda()
+    w2 = l.ffn.output_layer.weight.detach().float()
+    w2_pad = torch.zeros(320, 2048)
+    w2_pad[:280] = w2
+    w2_pad = w2_pad.to(torch.bfloat16).cuda()
+    b2p = torch.zeros(320)
+    b2p[:280] = l.ffn.output_layer.bias.detach().float()
+    b2p = b2p.cuda()
+    for M in (2048, 4096, 300, 257):
+        x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+        out = runner.ext.fused_ffn_v3(x, w1v2, w2_pad, b2p, 0.7).float()
+        xf = x.float()
+        ref = xf + 0.7 * (
+            torch.relu(xf @ w1.t().cuda() + b1.cuda())
+            @ w2.t().cuda() + l.ffn.output_layer.bias.cuda().float()
+        )
+        err = (out - ref).abs()
+        scale = ref.abs().mean().clamp_min(1e-3)
+        assert (err.mean() / scale).item() < 0.02, (M, err.mean().item())
+        assert err.max().item() < 0.3, (M, err.max().item())
