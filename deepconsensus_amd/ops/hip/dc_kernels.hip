@@ -45,9 +45,9 @@ __global__ __launch_bounds__(256) void fused_ln_head_qv_kernel(
     float max_qual) {
   const int wave_in_block = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int n = blockIdx.x * 4 + wave_in_block;
-  if (n >= N) return;
-
+  // Persistent waves: one-shot wave-per-position launched N/4 tiny blocks
+  // and was launch/drain-bound (~11x the HBM time at N=409600).
+  for (int n = blockIdx.x * 4 + wave_in_block; n < N; n += gridDim.x * 4) {
   const T* xi = x + (size_t)n * H;
 
   // Per-lane partial sums over dims lane, lane+64, ...
@@ -120,6 +120,7 @@ __global__ __launch_bounds__(256) void fused_ln_head_qv_kernel(
     bases_out[n] = (uint8_t)arg;
     quals_out[n] = (uint8_t)q;
   }
+  }
 }
 
 }  // namespace
@@ -142,7 +143,7 @@ std::vector<at::Tensor> fused_ln_head_qv(
     probs = at::empty({N, 5}, opts.dtype(at::kFloat));
     probs_ptr = probs.data_ptr<float>();
   }
-  dim3 grid((N + 3) / 4);
+  dim3 grid(std::min((N + 3) / 4, 2048));
   dim3 block(256);
   hipStream_t stream = at::hip::getCurrentHIPStream();
   auto gc = gamma.contiguous(), bc = beta.contiguous();

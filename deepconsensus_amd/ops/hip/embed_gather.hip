@@ -37,11 +37,16 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
   __shared__ short ids[MAX_R][TILE_P];
 
   const int tiles_per_b = (L + TILE_P - 1) / TILE_P;
-  const int b = blockIdx.x / tiles_per_b;
-  const int l0 = (blockIdx.x % tiles_per_b) * TILE_P;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+
+  // Persistent tiles: one-shot B*tiles_per_b tiny blocks were
+  // launch/drain-bound at production batch sizes.
+  for (int tile = blockIdx.x; tile < B * tiles_per_b; tile += gridDim.x) {
+  const int b = tile / tiles_per_b;
+  const int l0 = (tile % tiles_per_b) * TILE_P;
+  __syncthreads();  // prior tile's id reads done before re-staging
 
   const TIn* rows_b = rows + (size_t)b * R * L;
   for (int r = wave; r < R; r += 4) {
@@ -90,6 +95,7 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
     *reinterpret_cast<uint4*>(
         out + ((size_t)b * L + l0 + p) * H + c * 8) = raw;
   }
+  }
 }
 
 }  // namespace
@@ -108,7 +114,7 @@ at::Tensor embed_gather(
   auto out = at::empty({B, L, nchunk * 8},
                        rc.options().dtype(at::kBFloat16));
   const int tiles_per_b = (L + TILE_P - 1) / TILE_P;
-  dim3 grid(B * tiles_per_b);
+  dim3 grid(std::min(B * tiles_per_b, 2048));
   dim3 block(256);
   hipStream_t stream = at::hip::getCurrentHIPStream();
   if (rc.dtype() == at::kFloat) {
