@@ -53,6 +53,9 @@ def make_synthetic_windows(params, batch: int, seed: int) -> np.ndarray:
     return rows
 
 
+used_graphs = False
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -133,6 +136,7 @@ def main():
         # hipGraph-capture the whole forward per device buffer: one replay
         # per step instead of ~50 eager launches.
         graphs = None
+        global used_graphs
         if not args.no_graphs and runner.native:
             try:
                 for b in range(2):  # allocator warmup outside capture
@@ -144,6 +148,7 @@ def main():
                     with torch.cuda.graph(g):
                         static_out.append(runner.forward_windows(dev_bufs[b]))
                     graphs.append(g)
+                used_graphs = True
             except Exception as e:  # pragma: no cover - graph support varies
                 print(f"# hipGraph capture unavailable ({e}); eager path",
                       file=sys.stderr)
@@ -221,7 +226,7 @@ def main():
                 "windows_per_zmw": windows_per_zmw,
                 "parallelism": f"dp{world}",
                 "native_kernels": bool(runner.native),
-                "hipgraph": bool(have_cuda) and not args.no_graphs,
+                "hipgraph": used_graphs,
             },
         }
         print(json.dumps(result))
