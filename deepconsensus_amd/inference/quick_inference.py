@@ -409,6 +409,14 @@ def run(
                     for_model.append(window)
         preds = run_model_on_examples(for_model, runner, options)
         preds.extend(skipped)
+        # Per-batch model/skip split (quick_inference.py:688-705).
+        total = max(len(preds), 1)
+        log.info(
+            "Example summary: ran model=%d (%.2f%%) skip=%d (%.2f%%) "
+            "total=%d.",
+            len(for_model), 100.0 * len(for_model) / total,
+            len(skipped), 100.0 * len(skipped) / total, len(preds),
+        )
         timelog.add("run_model", batch_name, before, n_examples,
                     n_subreads, len(inputs))
         if options.end_after_stage == DebugStage.RUN_MODEL:
