@@ -15,7 +15,6 @@ import argparse
 import collections
 import logging
 import multiprocessing
-import time
 from typing import Dict, List, Optional
 
 import numpy as np

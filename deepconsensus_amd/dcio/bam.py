@@ -20,7 +20,6 @@ name" for truth-to-CCS lookups is provided by a sequential scan into a dict
 """
 from __future__ import annotations
 
-import io
 import struct
 import zlib
 from typing import Any, Dict, Iterator, List, Optional, Sequence, Tuple

@@ -3,7 +3,7 @@ from __future__ import annotations
 
 import dataclasses
 import gzip
-from typing import Iterator, List, Optional
+from typing import Iterator, List
 
 
 @dataclasses.dataclass

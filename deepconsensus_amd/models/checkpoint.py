@@ -21,7 +21,6 @@ from deepconsensus_amd.models.config import (
     read_params_from_json,
     save_params_as_json,
 )
-from deepconsensus_amd.utils import constants
 
 
 def save_checkpoint(

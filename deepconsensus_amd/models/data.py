@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import glob as globlib
 import random
-from typing import Dict, Iterator, List, Optional, Tuple
+from typing import Dict, Iterator, List
 
 import numpy as np
 

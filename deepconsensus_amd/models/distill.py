@@ -14,7 +14,6 @@ import os
 import time
 from typing import List, Optional
 
-import numpy as np
 import torch
 
 from deepconsensus_amd.models import checkpoint as ckpt_lib

@@ -15,7 +15,6 @@ import os
 import shutil
 from typing import List, Optional
 
-import numpy as np
 import torch
 
 from deepconsensus_amd.models import checkpoint as ckpt_lib

@@ -11,7 +11,6 @@ import logging
 import os
 from typing import List, Optional
 
-import numpy as np
 import torch
 
 from deepconsensus_amd.models import checkpoint as ckpt_lib

@@ -8,7 +8,7 @@ linear warmup over warmup_steps.
 """
 from __future__ import annotations
 
-from typing import Iterable, List, Optional, Tuple
+from typing import Tuple
 
 import torch
 

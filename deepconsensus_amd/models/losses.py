@@ -19,7 +19,7 @@ Re-implements the reference's losses_and_metrics.py on torch/numpy:
 """
 from __future__ import annotations
 
-from typing import Dict, Mapping, Optional, Tuple
+from typing import Mapping, Optional, Tuple
 
 import numpy as np
 import torch

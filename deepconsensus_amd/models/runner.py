@@ -16,7 +16,6 @@ import copy
 import math
 from typing import List, Optional, Tuple
 
-import numpy as np
 import torch
 
 from deepconsensus_amd import ops as dc_ops

@@ -7,7 +7,6 @@ PyTorch. CPU paths always use the torch reference implementations.
 from __future__ import annotations
 
 import os
-import sys
 
 _ext = None
 _ext_error: str | None = None

@@ -9,7 +9,6 @@ training later requires for step counts (model_utils.py:182-205).
 from __future__ import annotations
 
 import argparse
-import collections
 import json
 import logging
 import multiprocessing
@@ -17,7 +16,6 @@ import os
 import time
 from typing import Dict, List, Optional
 
-import numpy as np
 
 import deepconsensus_amd
 from deepconsensus_amd.dcio import tfrecord

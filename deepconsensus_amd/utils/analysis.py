@@ -8,7 +8,6 @@ from __future__ import annotations
 
 from typing import Dict
 
-import numpy as np
 
 
 def edit_distance(a: str, b: str) -> int:

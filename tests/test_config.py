@@ -2,7 +2,6 @@
 import json
 import os
 
-import pytest
 
 from deepconsensus_amd.models import config as cfg
 

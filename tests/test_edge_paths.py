@@ -1,9 +1,7 @@
 """Edge-path tests: CCS smart windows, overflow windows, label overflow,
 ccs_bq examples — the fidelity corners of pre_lib (SURVEY.md hard part #2)."""
-import json
 
 import numpy as np
-import pytest
 
 from deepconsensus_amd.dcio import bam as bam_lib, tfrecord
 from deepconsensus_amd.preprocess import read as R

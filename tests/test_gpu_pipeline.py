@@ -5,7 +5,6 @@ import json
 import os
 
 import pytest
-import torch
 
 from deepconsensus_amd.models import config as cfg
 

@@ -6,7 +6,7 @@ import pytest
 import torch
 
 from deepconsensus_amd.models import losses as L
-from deepconsensus_amd.utils import constants, phred
+from deepconsensus_amd.utils import phred
 
 
 def seq_to_array(s):

@@ -1,8 +1,6 @@
 """Tests for the calibration tool, FASTA IO, export tool, analysis utils."""
 import os
 
-import numpy as np
-import pytest
 import torch
 
 from deepconsensus_amd.dcio import bam as bam_lib

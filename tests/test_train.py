@@ -6,7 +6,6 @@ import glob
 import json
 import os
 
-import numpy as np
 import pytest
 import torch
 
