@@ -74,7 +74,13 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
   uint4 kr0, kr1, kr2, kr3, kr4, kr5, kr6, kr7;
   uint4 vr0, vr1, vr2, vr3, vr4, vr5, vr6, vr7;
   const int n_kg = L * 19;  // K granules: 19 x 16 B per row (stride 304 B)
-  const int n_vg = L * 18;  // V granules: 8 dims of one key each
+  // V granules: 8 dims of one key each, indexed key-major so consecutive
+  // lanes WRITE consecutive key slots of one Vt row (conflict-free 2-B
+  // writes) — the dim-major order made lanes stride 8 Vt rows apart,
+  // hitting 2 LDS banks 32-ways (PMC: conflict cycles ~= active cycles).
+  // The now-uncoalesced 16-B global reads stay in L2 (104-row working
+  // set) and their latency rides the cross-item prefetch.
+  const int n_vg = L * 18;
 
   // Issue the global loads for item `it`'s K and V into the staging
   // registers. Tail granules (dims 136..139) take the 2-dword path; the
@@ -103,7 +109,7 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       const int idx = min(tid + i * 256, n_vg - 1);
-      const int r = idx / 18, d0 = 8 * (idx % 18);
+      const int r = idx % L, d0 = 8 * (idx / L);
       uint4 v = {};
       const bf16* src = bs + r * RS + 2 * H * D + d0;
       if (d0 + 8 <= D) {
@@ -135,7 +141,7 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       const int idx = min(tid + i * 256, n_vg - 1);
-      const int r = idx / 18, d0 = 8 * (idx % 18);
+      const int r = idx % L, d0 = 8 * (idx / L);
       const uint4 v =
           (i == 0 ? vr0 : i == 1 ? vr1 : i == 2 ? vr2 : i == 3 ? vr3
            : i == 4 ? vr4 : i == 5 ? vr5 : i == 6 ? vr6 : vr7);
