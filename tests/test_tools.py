@@ -114,3 +114,40 @@ def test_cli_dispatch_help(capsys):
     cli.main([])
     out = capsys.readouterr().out
     assert "run" in out and "preprocess" in out
+
+
+def test_parse_calibration_string():
+    from deepconsensus_amd.calibration import calibration as cal
+
+    skip = cal.parse_calibration_string("skip")
+    assert not skip.enabled and skip.w == 1.0 and skip.b == 0.0
+    v = cal.parse_calibration_string("10,0.99,0.15")
+    assert v.enabled and v.threshold == 10 and v.w == 0.99 and v.b == 0.15
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        cal.parse_calibration_string("1,2")
+    with _pytest.raises(ValueError):
+        cal.parse_calibration_string("not,a,number")
+
+
+def test_calibrate_quality_scores_threshold_semantics():
+    import numpy as np
+
+    from deepconsensus_amd.calibration import calibration as cal
+
+    q = np.array([5.0, 10.0, 20.0, 40.0])
+    # threshold 0: applied everywhere.
+    v0 = cal.parse_calibration_string("0,2,1")
+    np.testing.assert_allclose(
+        cal.calibrate_quality_scores(q, v0), q * 2 + 1
+    )
+    # threshold 10: strictly-above only; at/below unchanged.
+    v10 = cal.parse_calibration_string("10,2,1")
+    np.testing.assert_allclose(
+        cal.calibrate_quality_scores(q, v10), [5.0, 10.0, 41.0, 81.0]
+    )
+    # v1.2 production string maps Q30 to ~Q35 (monotone, finite).
+    vp = cal.parse_calibration_string("0,1.197654,-0.99781")
+    out = cal.calibrate_quality_scores(np.array([30.0]), vp)
+    assert 34.5 < out[0] < 35.1
