@@ -252,3 +252,19 @@ def test_calibrate_tool_real_alignment(tmp_path):
     m_lo = sum(rows[q][0] for q in range(2, 7))
     x_lo = sum(rows[q][1] for q in range(2, 7))
     assert x_lo / (m_lo + x_lo) > 0.1
+
+
+def test_reference_params_json_bq_variant():
+    """The ccs_bq model's shipped params.json also resolves (86 rows)."""
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models.model import get_model
+
+    p = cfg.read_params_from_json(
+        "/root/reference/deepconsensus/testdata/model_bq"
+    )
+    cfg.modify_params(p, is_training=False)
+    assert p.use_ccs_bq
+    assert p.total_rows == 86
+    m = get_model(p)
+    out = m(__import__("torch").zeros(2, 86, 100))
+    assert out.shape == (2, 100, 5)
