@@ -153,6 +153,13 @@ def run_model_on_examples(
     """Batched model execution emitting per-window DCModelOutput."""
     predictions: List[stitch_utils.DCModelOutput] = []
     params = runner.params
+    # Native path: features are non-negative and <= SN_MAX (500) after
+    # clipping, and the embed kernel casts to int anyway, so int16 staging
+    # halves H2D traffic with bit-identical results (numpy's truncation
+    # toward zero == the kernel's (int)v == the torch model's .long()).
+    use_i16 = bool(getattr(runner, "native", False))
+    np_dtype = np.int16 if use_i16 else np.float32
+    pinned: Dict[Tuple[int, ...], torch.Tensor] = {}
     for i in range(0, len(feature_dicts), options.batch_size):
         chunk = feature_dicts[i : i + options.batch_size]
         rows = np.stack(
@@ -162,8 +169,16 @@ def run_model_on_examples(
                 ]
                 for f in chunk
             ]
-        ).astype(np.float32)
-        bases_t, quals_t = runner.forward_windows(torch.from_numpy(rows))
+        ).astype(np_dtype)
+        rows_t = torch.from_numpy(rows)
+        if use_i16:
+            buf = pinned.get(rows_t.shape)
+            if buf is None:
+                buf = torch.empty_like(rows_t).pin_memory()
+                pinned[rows_t.shape] = buf
+            buf.copy_(rows_t)
+            rows_t = buf
+        bases_t, quals_t = runner.forward_windows(rows_t)
         bases = bases_t.cpu().numpy()
         quals = quals_t.cpu().numpy()
         for j, f in enumerate(chunk):

@@ -5,6 +5,7 @@ import json
 import os
 
 import pytest
+import torch
 
 from deepconsensus_amd.models import config as cfg
 
@@ -91,3 +92,35 @@ def test_train_bf16_gpu(tmp_path):
     assert summary["steps"] >= 2
     import math
     assert math.isfinite(summary["eval/loss"])
+
+
+@pytest.mark.gpu
+def test_native_int16_staging_matches_fp32(tmp_path):
+    """int16 feature staging is bit-identical to fp32 through the native
+    path (the embed kernel casts to int either way)."""
+    import numpy as np
+
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models.model import get_model
+    from deepconsensus_amd.models.runner import InferenceRunner
+
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=False)
+    torch.manual_seed(11)
+    runner = InferenceRunner(params, get_model(params), device="cuda")
+    assert runner.native
+    rng = np.random.default_rng(0)
+    rows = np.zeros((64, params.total_rows, 100), dtype=np.float32)
+    mp = params.max_passes
+    rows[:, 0:mp] = rng.integers(0, 5, size=(64, mp, 100))
+    rows[:, mp:3 * mp] = rng.integers(0, 256, size=(64, 2 * mp, 100))
+    rows[:, 3 * mp:4 * mp] = rng.integers(0, 3, size=(64, mp, 100))
+    rows[:, 4 * mp] = rng.integers(0, 5, size=(64, 100))
+    # fractional SN values exercise the truncation equivalence
+    rows[:, -4:] = rng.uniform(3.2, 29.8, size=(64, 4, 1))
+    b32, q32 = runner.forward_windows(torch.from_numpy(rows))
+    b16, q16 = runner.forward_windows(
+        torch.from_numpy(rows.astype(np.int16))
+    )
+    assert torch.equal(b32.cpu(), b16.cpu())
+    assert torch.equal(q32.cpu(), q16.cpu())
