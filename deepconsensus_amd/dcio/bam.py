@@ -32,6 +32,7 @@ BAM_MAGIC = b"BAM\x01"
 _CIGAR_CHARS = "MIDNSHP=XB"
 _SEQ_NT16 = "=ACMGRSVTWYHKDBN"
 _NT16_OF = {c: i for i, c in enumerate(_SEQ_NT16)}
+_NT16_U8 = np.frombuffer(_SEQ_NT16.encode("ascii"), np.uint8)
 
 # Flag bits.
 FUNMAP = 0x4
@@ -263,6 +264,34 @@ class BamRead:
             # CHARD_CLIP / CPAD consume nothing.
         return pairs
 
+    def aligned_index_arrays(self) -> Tuple[np.ndarray, np.ndarray]:
+        """Vectorized get_aligned_pairs: (read_idx, ccs_idx) int32 arrays
+        per aligned column, -1 where the side is absent. Loops over cigar
+        OPS, not bases."""
+        total = 0
+        for op, n in self.cigartuples:
+            if op not in (constants.CHARD_CLIP, constants.CPAD):
+                total += n
+        read_idx = np.full(total, -1, np.int32)
+        ccs_idx = np.full(total, -1, np.int32)
+        qpos, rpos, out = 0, self.pos, 0
+        for op, n in self.cigartuples:
+            if op in (constants.CMATCH, constants.CEQUAL, constants.CDIFF):
+                read_idx[out:out + n] = np.arange(qpos, qpos + n)
+                ccs_idx[out:out + n] = np.arange(rpos, rpos + n)
+                qpos += n
+                rpos += n
+            elif op in (constants.CINS, constants.CSOFT_CLIP):
+                read_idx[out:out + n] = np.arange(qpos, qpos + n)
+                qpos += n
+            elif op in (constants.CDEL, constants.CREF_SKIP):
+                ccs_idx[out:out + n] = np.arange(rpos, rpos + n)
+                rpos += n
+            else:  # CHARD_CLIP / CPAD consume nothing
+                continue
+            out += n
+        return read_idx, ccs_idx
+
     def infer_query_length(self) -> int:
         return len(self.seq)
 
@@ -407,18 +436,17 @@ class BamReader:
         off = 32
         qname = buf[off:off + l_read_name - 1].decode()
         off += l_read_name
-        cigartuples = []
-        for i in range(n_cigar):
-            (v,) = struct.unpack_from("<I", buf, off)
-            cigartuples.append((v & 0xF, v >> 4))
-            off += 4
+        # Vectorized cigar + 4-bit seq decode (the per-base python loops
+        # were the hottest lines of the whole serial feeder path).
+        cig = np.frombuffer(buf, "<u4", n_cigar, off)
+        cigartuples = list(zip((cig & 0xF).tolist(), (cig >> 4).tolist()))
+        off += 4 * n_cigar
         nbytes = (l_seq + 1) // 2
-        seq_chars = []
-        for i in range(l_seq):
-            b = buf[off + i // 2]
-            code = (b >> 4) if i % 2 == 0 else (b & 0xF)
-            seq_chars.append(_SEQ_NT16[code])
-        seq = "".join(seq_chars)
+        packed = np.frombuffer(buf, np.uint8, nbytes, off)
+        chars = np.empty(2 * nbytes, np.uint8)
+        chars[0::2] = _NT16_U8[packed >> 4]
+        chars[1::2] = _NT16_U8[packed & 0xF]
+        seq = chars[:l_seq].tobytes().decode("ascii")
         off += nbytes
         quals = np.frombuffer(buf, np.uint8, l_seq, off).astype(np.int16)
         if l_seq and quals.size and quals[0] == 0xFF:

@@ -73,20 +73,16 @@ def expand_clip_indent(
     if ins_trim > 0:
         read = trim_insertions(read, ins_trim, counter)
 
-    aligned_pairs = read.get_aligned_pairs()
-    read_idx = np.array(
-        [x[0] if x[0] is not None else -1 for x in aligned_pairs]
-    )
-    ccs_idx = np.array(
-        [x[1] if x[1] is not None else -1 for x in aligned_pairs]
-    )
+    read_idx, ccs_idx = read.aligned_index_arrays()
     aln_len = len(read_idx)
 
     new_seq = np.full(aln_len, constants.GAP, dtype="<U1")
     new_pw = np.zeros(aln_len, dtype=np.uint8)
     new_ip = np.zeros(aln_len, dtype=np.uint8)
 
-    new_seq[read_idx >= 0] = list(read.seq)
+    new_seq[read_idx >= 0] = np.frombuffer(
+        read.seq.encode("ascii"), dtype="S1"
+    ).astype("<U1")
 
     strand = (
         constants.Strand.REVERSE
