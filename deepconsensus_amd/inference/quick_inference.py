@@ -96,15 +96,25 @@ class InferenceOptions:
 
 
 def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
-    """Windows + counters for one ZMW (quick_inference.py:535-564)."""
+    """Windows + counters for one ZMW (quick_inference.py:535-564).
+
+    ``subreads`` may be a deferred ZmwJob (raw BAM records): expansion then
+    runs HERE — i.e. in the worker pool — instead of the serial feeder.
+    """
     zmw, subreads, dc_config, window_widths, *rest = one_zmw
     stage = rest[0] if rest else DebugStage.FULL
+    expand_counter = None
+    if isinstance(subreads, pre_feeder.ZmwJob):
+        expand_counter = collections.Counter()
+        subreads = subreads.materialize(expand_counter)
     dc_whole = pre_feeder.subreads_to_dc_example(
         subreads=subreads,
         ccs_seqname=zmw,
         dc_config=dc_config,
         window_widths=window_widths,
     )
+    if expand_counter:
+        dc_whole.counter.update(expand_counter)
     if stage == DebugStage.DC_INPUT:
         return [], dc_whole.counter
     feature_dicts = [x.to_features_dict() for x in dc_whole.iter_examples()]
@@ -359,6 +369,7 @@ def run(
         ins_trim=options.ins_trim,
         use_ccs_smart_windows=options.use_ccs_smart_windows,
         limit=limit,
+        defer_expansion=True,
     )
 
     pool = None

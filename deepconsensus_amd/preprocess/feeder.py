@@ -149,6 +149,34 @@ def read_truth_split(split_fname: str) -> Dict[str, str]:
     return contig_split
 
 
+class ZmwJob:
+    """Deferred per-ZMW expansion: carries the raw BAM records so the
+    CPU-heavy expand_clip_indent runs in the worker pool instead of the
+    serial feeder thread (inference mode only — training filters on the
+    expanded label feeder-side)."""
+
+    __slots__ = ("read_set", "ccs_bam_read", "ins_trim")
+
+    def __init__(self, read_set, ccs_bam_read, ins_trim):
+        self.read_set = read_set
+        self.ccs_bam_read = ccs_bam_read
+        self.ins_trim = ins_trim
+
+    def __len__(self):
+        return len(self.read_set) + 1
+
+    def materialize(self, counter) -> List[Read]:
+        expand = functools.partial(
+            expand_clip_indent,
+            truth_range=None,
+            ins_trim=self.ins_trim,
+            counter=counter,
+        )
+        subreads = list(map(expand, self.read_set))
+        subreads.append(construct_ccs_read(self.ccs_bam_read))
+        return subreads
+
+
 def create_proc_feeder(
     subreads_to_ccs: str,
     ccs_bam: str,
@@ -160,6 +188,7 @@ def create_proc_feeder(
     truth_split: Optional[str] = None,
     limit: int = 0,
     bam_reader_threads: int = 1,
+    defer_expansion: bool = False,
 ):
     """Generator feeding per-ZMW jobs (pre_lib.py:1279-1367)."""
     main_counter = collections.Counter()
@@ -171,17 +200,13 @@ def create_proc_feeder(
         truth_index = bam.fetch_index(truth_to_ccs)
         truth_ref_coords = read_truth_bedfile(truth_bed)
         truth_split_dict = read_truth_split(truth_split)
+    assert not (defer_expansion and is_training), (
+        "defer_expansion is an inference-mode optimization"
+    )
 
     def proc_feeder():
         for read_set in subread_grouper:
             main_counter["n_zmw_processed"] += 1
-            expand = functools.partial(
-                expand_clip_indent,
-                truth_range=None,
-                ins_trim=ins_trim,
-                counter=main_counter,
-            )
-            subreads = list(map(expand, read_set))
             ccs_seqname = read_set[0].reference_name
             while True:
                 ccs_bam_read = next(ccs_bam_h)
@@ -189,11 +214,27 @@ def create_proc_feeder(
                     break
             if ccs_bam_read.qname != ccs_seqname:
                 raise ValueError(f"ccs bam does not contain {ccs_seqname}")
-
-            ccs_read = construct_ccs_read(ccs_bam_read)
             window_widths = None
             if use_ccs_smart_windows:
                 window_widths = np.array(ccs_bam_read.get_tag("wl"))
+
+            if defer_expansion:
+                main_counter["n_zmw_inference"] += 1
+                main_counter["n_zmw_pass"] += 1
+                yield (ZmwJob(read_set, ccs_bam_read, ins_trim),
+                       ccs_seqname, dc_config, "inference", window_widths)
+                if limit and main_counter["n_zmw_pass"] >= limit:
+                    break
+                continue
+
+            expand = functools.partial(
+                expand_clip_indent,
+                truth_range=None,
+                ins_trim=ins_trim,
+                counter=main_counter,
+            )
+            subreads = list(map(expand, read_set))
+            ccs_read = construct_ccs_read(ccs_bam_read)
             subreads.append(ccs_read)
 
             if is_training:
