@@ -118,6 +118,12 @@ def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
     if stage == DebugStage.DC_INPUT:
         return [], dc_whole.counter
     feature_dicts = [x.to_features_dict() for x in dc_whole.iter_examples()]
+    # Features are integral-after-truncation for the model (the embedding
+    # casts to int; SN fractions truncate identically) and <= SN_MAX, so
+    # ship them across the process boundary as int16 — halves the pickle
+    # volume coming back from the worker pool.
+    for f in feature_dicts:
+        f["subreads"] = np.asarray(f["subreads"]).astype(np.int16)
     return feature_dicts, dc_whole.counter
 
 
