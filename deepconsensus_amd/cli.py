@@ -52,7 +52,15 @@ def _run_main(argv: List[str]) -> None:
     ap.add_argument("--use_only_gpu_index", type=int, default=None,
                     help="pin inference to this GPU index "
                     "(shorthand for --device cuda:N)")
+    ap.add_argument("--ccs_fasta", default=None,
+                    help="deprecated; use --ccs_bam")
     args = ap.parse_args(argv)
+    if args.ccs_fasta:
+        # Parity with the reference's deprecation (quick_inference.py:968).
+        raise NotImplementedError(
+            "The --ccs_fasta flag has been deprecated. "
+            "Please use --ccs_bam instead."
+        )
 
     options = qi.InferenceOptions(
         max_length=args.max_length,

@@ -151,3 +151,14 @@ def test_calibrate_quality_scores_threshold_semantics():
     vp = cal.parse_calibration_string("0,1.197654,-0.99781")
     out = cal.calibrate_quality_scores(np.array([30.0]), vp)
     assert 34.5 < out[0] < 35.1
+
+
+def test_run_ccs_fasta_deprecated():
+    import pytest as _pytest
+
+    from deepconsensus_amd import cli
+
+    with _pytest.raises(NotImplementedError, match="deprecated"):
+        cli.main(["run", "--subreads_to_ccs", "x.bam", "--ccs_bam", "y.bam",
+                  "--checkpoint", "random", "--output", "o.fastq",
+                  "--ccs_fasta", "z.fa"])
