@@ -74,6 +74,8 @@ def tf_record_writer_proc(output_fname: str, splits: List[str], queue):
 def main(argv: Optional[List[str]] = None) -> None:
     ap = argparse.ArgumentParser("deepconsensus preprocess")
     ap.add_argument("--subreads_to_ccs", required=True)
+    ap.add_argument("--ccs_fasta", default=None,
+                    help="deprecated; use --ccs_bam")
     ap.add_argument("--ccs_bam", required=True)
     ap.add_argument("--output", required=True,
                     help="must end in .tfrecord.gz; use @split when training")
@@ -90,6 +92,12 @@ def main(argv: Optional[List[str]] = None) -> None:
     ap.add_argument("--max_passes", type=int, default=20)
     ap.add_argument("--max_length", type=int, default=100)
     args = ap.parse_args(argv)
+    if args.ccs_fasta:
+        # Parity with the reference's deprecation (preprocess.py:247).
+        raise NotImplementedError(
+            "The --ccs_fasta flag has been deprecated. "
+            "Please use --ccs_bam instead."
+        )
 
     if args.cpus == 1:
         raise ValueError("Must set cpus to 0 or >=2 for parallel processing.")
