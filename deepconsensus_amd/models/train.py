@@ -221,6 +221,11 @@ def main(argv: Optional[List[str]] = None) -> None:
     ap.add_argument("--device", default=None)
     ap.add_argument("--bf16", action="store_true",
                     help="autocast forward to bf16 (fp32 loss/LN/softmax)")
+    ap.add_argument("--write_checkpoint_metrics", type=lambda v: v != "false",
+                    default=True,
+                    help="append eval metrics to checkpoint_metrics.tsv")
+    ap.add_argument("--eval_and_log_every_step", action="store_true",
+                    help="debug: log every step and eval every step")
     args = ap.parse_args(argv)
 
     params = cfg.get_config(args.params)
@@ -248,7 +253,10 @@ def main(argv: Optional[List[str]] = None) -> None:
         try:
             train_model(
                 args.out_dir, params, device=device,
-                eval_every=args.eval_every, limit_steps=args.limit_steps,
+                eval_every=(1 if args.eval_and_log_every_step
+                            else args.eval_every),
+                limit_steps=args.limit_steps,
+                write_checkpoint_metrics=args.write_checkpoint_metrics,
                 warm_start=args.checkpoint, use_bf16=args.bf16,
             )
             break
