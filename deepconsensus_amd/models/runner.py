@@ -63,8 +63,14 @@ def build_gather_tables(model: EncoderOnlyLearnedValuesTransformer):
     for name in order:
         t = tables[name]
         table_base[name] = off
-        flat_parts.append(t.reshape(-1))
-        off += t.numel()
+        flat = t.reshape(-1)
+        # Pad every table to an 8-element boundary so width-8 gathers stay
+        # 16-B aligned (required for the kernel's LDS ds_read_b128 path).
+        pad = (-flat.numel()) % 8
+        if pad:
+            flat = torch.cat([flat, flat.new_zeros(pad)])
+        flat_parts.append(flat)
+        off += flat.numel()
     table_flat = torch.cat(flat_parts).contiguous()
 
     # Per input row: (table name, shift).
