@@ -208,10 +208,10 @@ class InferenceRunner:
                 lw["b2_a"] = lw["b2"] * lw["alpha_ffn"]
             # Padded weights for the fused FFN kernel (in-bounds 16B frags):
             # W1 [2048, 288], W2 [320, 2048], b1 fp32, b2 [320] fp32.
-            # The hand-fused FFN kernel measures 1.46 ms/layer vs 1.79 ms
-            # for the hipBLASLt pair with fused epilogues at batch 4096
-            # (profiles/r01_perf_journal.md) - default ON for the production
-            # shape; DC_FUSED_FFN=0 falls back to hipBLASLt.
+            # The hand-fused FFN kernels measure 1.29-1.32 ms/layer vs
+            # 1.80 ms for the hipBLASLt pair with fused epilogues at batch
+            # 4096 (profiles/r01_perf_journal.md) - default ON for the
+            # production shape; DC_FUSED_FFN=0 falls back to hipBLASLt.
             import os as _os
 
             self.ffn_fused_ok = (

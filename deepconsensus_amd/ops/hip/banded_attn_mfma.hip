@@ -3,17 +3,22 @@
 // Same math as banded_attn.hip (attention_layer.py:196-218: banded QK^T,
 // fp32 softmax over |i-j| <= win, P V), restructured for the CDNA4 matrix
 // cores:
-//  * one workgroup = one (b, h); 4 waves, each owning 32 query rows;
+//  * PERSISTENT workgroups (grid 512 = 2 blocks/CU): each block of 4
+//    waves (32 query rows each) walks many (b, h) items — the one-shot
+//    grid was launch/drain-bound — and prefetches the NEXT item's K/V
+//    into named staging registers during the current softmax/PV;
 //  * swapped QK^T — mfma(K, Q) — so each lane holds a full query row's band
 //    scores and softmax is in-lane (one shfl_xor(32) pair to merge halves);
 //  * P repacked to MFMA A-fragments with v_cvt_pk_bf16_f32 +
 //    permlane32_swap (the cdna_hip_programming.md T12 pattern);
-//  * K staged row-major in LDS (stride 152 elems: conflict-free b128 lane
-//    groups), V staged TRANSPOSED (Vt[dim][key], stride 168, keys shifted by
-//    V_OFF=12 so every PV fragment read is 16-byte aligned even at the
-//    negative key-window of the first row block);
+//  * K staged row-major in LDS (stride 152 elems) in 16-B granules, V
+//    staged TRANSPOSED (Vt[dim][key], stride 168, keys shifted by win so
+//    every PV fragment read is 16-byte aligned even at the negative
+//    key-window of the first row block) with a key-major conflict-free
+//    scatter;
 //  * 38 v_mfma_f32_32x32x16_bf16 per wave per (b,h), fp32 accumulation;
-//    ~79 KB LDS => 2 workgroups per CU.
+//    ~79 KB LDS => 2 workgroups per CU. 626 -> 287 us at B=4096
+//    (profiles/r01_perf_journal.md).
 //
 // Out-of-band and out-of-range keys are masked to -inf before softmax; the
 // corresponding P entries are exactly 0, and the V slots they multiply are
