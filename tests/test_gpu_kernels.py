@@ -463,3 +463,26 @@ def test_fused_ffn_v3_matches_torch(setup):
         scale = ref.abs().mean().clamp_min(1e-3)
         assert (err.mean() / scale).item() < 0.02, (M, err.mean().item())
         assert err.max().item() < 0.3, (M, err.max().item())
+
+
+@pytest.mark.gpu
+def test_runner_non_rezero_fallback_path():
+    """rezero=False configs take the generic bf16 layer path (pre-LN),
+    not the packed fast path; outputs track the fp32 torch reference."""
+    params = cfg.get_config("transformer_learn_values+custom")
+    params.rezero = False
+    cfg.modify_params(params, is_training=False)
+    torch.manual_seed(13)
+    model = get_model(params)
+    import copy as _copy
+
+    ref_model = _copy.deepcopy(model).float()
+    runner = InferenceRunner(params, model, device="cuda:0")
+    assert runner.native and not runner.rezero_fast
+    rows = _make_rows(params)[:64]
+    bases, quals, probs = runner.forward_windows(rows, want_probs=True)
+    ref = ref_model(rows.float(), training=False)
+    agree = (bases.cpu() == ref.argmax(-1).to(torch.uint8)).float().mean()
+    assert agree > 0.98, float(agree)
+    err = (probs.cpu() - ref).abs().max()
+    assert err < 0.05, float(err)
