@@ -272,9 +272,13 @@ class InferenceRunner:
                 except Exception:
                     self._addmm_act = False
         else:
-            self.layers_bf16 = (
-                copy.deepcopy(model.layers).to(bf16).to(dev)
-            )
+            layers = copy.deepcopy(model.layers).to(bf16)
+            # LN stays fp32 (the model computes it on x.float(); a bf16
+            # LN weight would make torch.layer_norm reject the mix).
+            for m in layers.modules():
+                if isinstance(m, torch.nn.LayerNorm):
+                    m.float()
+            self.layers_bf16 = layers.to(dev)
         self.ln_gamma = model.output_norm.weight.detach().float().to(dev)
         self.ln_beta = model.output_norm.bias.detach().float().to(dev)
         self.w_head = model.fc1.weight.detach().float().contiguous().to(dev)
