@@ -486,3 +486,31 @@ def test_runner_non_rezero_fallback_path():
     assert agree > 0.98, float(agree)
     err = (probs.cpu() - ref).abs().max()
     assert err < 0.05, float(err)
+
+
+@pytest.mark.gpu
+def test_runner_ccs_bq_full_forward():
+    """use_ccs_bq config (86 rows, +1-shift bq embedding) through the full
+    native path vs the fp32 torch reference."""
+    params = cfg.get_config("transformer_learn_values+custom")
+    params.use_ccs_bq = True
+    cfg.modify_params(params, is_training=False)
+    torch.manual_seed(17)
+    model = get_model(params)
+    import copy as _copy
+
+    ref_model = _copy.deepcopy(model).float()
+    runner = InferenceRunner(params, model, device="cuda:0")
+    assert runner.native
+    rows = _make_rows(params)
+    nb = rows.shape[0]
+    # bq row carries CCS base qualities in [-1, 93].
+    mp = params.max_passes
+    rows[:, 4 * mp + 1] = torch.from_numpy(
+        np.random.default_rng(3).integers(-1, 94, size=(nb, 100))
+    ).float()
+    bases, quals, probs = runner.forward_windows(rows, want_probs=True)
+    ref = ref_model(rows.float(), training=False)
+    agree = (bases.cpu() == ref.argmax(-1).to(torch.uint8)).float().mean()
+    assert agree > 0.98, float(agree)
+    assert (probs.cpu() - ref).abs().max() < 0.05
