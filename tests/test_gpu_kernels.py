@@ -514,3 +514,35 @@ def test_runner_ccs_bq_full_forward():
     agree = (bases.cpu() == ref.argmax(-1).to(torch.uint8)).float().mean()
     assert agree > 0.98, float(agree)
     assert (probs.cpu() - ref).abs().max() < 0.05
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("heads,max_len", [(4, 100), (2, 120)])
+def test_runner_generic_attention_shapes(heads, max_len):
+    """Configs off the MFMA fast path (head_dim != 140 or L > 104) run the
+    generic banded-attention kernel through the same native pipeline."""
+    params = cfg.get_config("transformer_learn_values+custom")
+    params.num_heads = heads
+    cfg.modify_params(params, max_length=max_len, is_training=False)
+    torch.manual_seed(19)
+    model = get_model(params)
+    import copy as _copy
+
+    ref_model = _copy.deepcopy(model).float()
+    runner = InferenceRunner(params, model, device="cuda:0")
+    assert runner.native
+    rng = np.random.default_rng(2)
+    R, L, mp = params.total_rows, params.max_length, params.max_passes
+    assert L == max_len
+    rows = np.zeros((32, R, L), dtype=np.float32)
+    rows[:, 0:mp] = rng.integers(0, 5, size=(32, mp, L))
+    rows[:, mp:3 * mp] = rng.integers(0, 256, size=(32, 2 * mp, L))
+    rows[:, 3 * mp:4 * mp] = rng.integers(0, 3, size=(32, mp, L))
+    rows[:, 4 * mp] = rng.integers(0, 5, size=(32, L))
+    rows[:, -4:] = rng.integers(0, 501, size=(32, 4, 1))
+    rows_t = torch.from_numpy(rows)
+    bases, quals, probs = runner.forward_windows(rows_t, want_probs=True)
+    ref = ref_model(rows_t.float(), training=False)
+    agree = (bases.cpu() == ref.argmax(-1).to(torch.uint8)).float().mean()
+    assert agree > 0.98, float(agree)
+    assert (probs.cpu() - ref).abs().max() < 0.05
