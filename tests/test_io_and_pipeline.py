@@ -357,3 +357,66 @@ def test_run_end_after_stage(tmp_path, stage, expect_fastq):
     assert "preprocess" in runtime
     if stage in ("dc_input", "tf_examples"):
         assert "run_model" not in runtime
+
+
+def test_bam_round_trip_fuzz():
+    """Property fuzz: random reads survive BAM write->read bit-exactly."""
+    from hypothesis import given, settings, strategies as st
+
+    header = bam_lib.BamHeader(text="@HD\tVN:1.6", references=[("r0", 5000)])
+
+    @settings(max_examples=40, deadline=None)
+    @given(
+        seq_len=st.integers(1, 300),
+        flag=st.sampled_from([0, 4, 16]),
+        n_ins=st.integers(0, 3),
+        mapq=st.integers(0, 254),
+        data=st.data(),
+    )
+    def check(seq_len, flag, n_ins, mapq, data):
+        rng = np.random.default_rng(data.draw(st.integers(0, 2**31)))
+        seq = "".join(rng.choice(list("ATCGN"), size=seq_len))
+        if flag == 4:
+            cig = []
+        else:
+            cig = [(0, seq_len)]
+            for _ in range(n_ins):
+                cig.append((1, int(rng.integers(1, 5))))
+                seq += "".join(
+                    rng.choice(list("ATCG"), size=cig[-1][1])
+                )
+        read = bam_lib.BamRead(
+            qname="m0/1/0_%d" % len(seq), flag=flag,
+            ref_id=-1 if flag == 4 else 0, pos=-1 if flag == 4 else 3,
+            mapq=mapq, cigartuples=cig, seq=seq,
+            query_qualities=rng.integers(0, 94, len(seq)),
+            tags={
+                "zm": 1,
+                "pw": rng.integers(0, 256, len(seq)).astype(np.uint8),
+                "sn": np.array([1.5, 2.5, 3.5, 4.5], np.float32),
+                "rq": 0.99,
+                "RG": "rg1",
+            },
+        )
+        import io as _io
+        import tempfile
+
+        with tempfile.TemporaryDirectory() as td:
+            p = os.path.join(td, "f.bam")
+            with bam_lib.BamWriter(p, header) as w:
+                w.write(read)
+            got = list(bam_lib.BamReader(p))
+        assert len(got) == 1
+        g = got[0]
+        assert g.qname == read.qname and g.flag == flag
+        assert g.seq == seq and g.mapq == mapq
+        assert g.cigartuples == cig
+        np.testing.assert_array_equal(
+            g.query_qualities, read.query_qualities
+        )
+        np.testing.assert_array_equal(g.get_tag("pw"), read.tags["pw"])
+        np.testing.assert_allclose(g.get_tag("sn"), read.tags["sn"])
+        assert abs(g.get_tag("rq") - 0.99) < 1e-6
+        assert g.get_tag("RG") == "rg1"
+
+    check()
