@@ -546,3 +546,15 @@ def test_runner_generic_attention_shapes(heads, max_len):
     agree = (bases.cpu() == ref.argmax(-1).to(torch.uint8)).float().mean()
     assert agree > 0.98, float(agree)
     assert (probs.cpu() - ref).abs().max() < 0.05
+
+
+@pytest.mark.gpu
+def test_native_forward_deterministic(setup):
+    """The native path is atomics-free: repeated forwards are bitwise
+    identical (base ids, QVs and probabilities)."""
+    params, model, runner, rows = setup
+    b1, q1, p1 = runner.forward_windows(rows, want_probs=True)
+    b2, q2, p2 = runner.forward_windows(rows, want_probs=True)
+    assert torch.equal(b1.cpu(), b2.cpu())
+    assert torch.equal(q1.cpu(), q2.cpu())
+    assert torch.equal(p1.cpu(), p2.cpu())

@@ -420,3 +420,42 @@ def test_bam_round_trip_fuzz():
         assert g.get_tag("RG") == "rg1"
 
     check()
+
+
+def test_example_codec_fuzz():
+    """Random feature dicts survive the tf.Example wire codec."""
+    from hypothesis import given, settings, strategies as st
+
+    keys = st.text(
+        alphabet="abcdefghij_/", min_size=1, max_size=20
+    )
+
+    @settings(max_examples=60, deadline=None)
+    @given(
+        d=st.dictionaries(
+            keys,
+            st.one_of(
+                st.tuples(st.just(example_codec.BYTES),
+                          st.lists(st.binary(max_size=64), max_size=4)),
+                st.tuples(st.just(example_codec.INT64),
+                          st.lists(st.integers(-2**62, 2**62), max_size=8)),
+                st.tuples(st.just(example_codec.FLOAT),
+                          st.lists(st.floats(-1e9, 1e9, width=32),
+                                   max_size=8)),
+            ),
+            max_size=6,
+        )
+    )
+    def check(d):
+        enc = example_codec.encode_example(d)
+        dec = example_codec.decode_example(enc)
+        assert set(dec) == set(d)
+        for k, (kind, vals) in d.items():
+            dkind, dvals = dec[k]
+            assert dkind == kind
+            if kind == example_codec.FLOAT:
+                np.testing.assert_allclose(dvals, vals, rtol=1e-6)
+            else:
+                assert list(dvals) == list(vals)
+
+    check()
