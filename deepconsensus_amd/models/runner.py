@@ -228,16 +228,18 @@ class InferenceRunner:
                 for i, l in enumerate(model.layers):
                     lw = self.layer_w[i]
                     # Padded projection weights for fused_linear (K5/K7):
-                    # QKV rows [840->896, 288], out-proj [320, 288].
+                    # QKV rows [840->896, 296], out-proj [320, 296] — the
+                    # 296-elem row stride matches the kernel's LDS image so
+                    # weights stream by glds (raw row-major copy).
                     wq = l.attn.q_proj.weight.detach().float()
                     wk = l.attn.k_proj.weight.detach().float()
                     wv = l.attn.v_proj.weight.detach().float()
                     wqkv = torch.cat([wq, wk, wv], 0)  # [840, 280]
-                    wqkv_pad = torch.zeros(896, 288)
+                    wqkv_pad = torch.zeros(896, 296)
                     wqkv_pad[:840, :280] = wqkv
                     lw["wqkv_pad"] = wqkv_pad.to(bf16).contiguous().to(dev)
                     wo = l.attn.out_proj.weight.detach().float()
-                    wout_pad = torch.zeros(320, 288)
+                    wout_pad = torch.zeros(320, 296)
                     wout_pad[:280, :280] = wo
                     lw["wout_pad"] = wout_pad.to(bf16).contiguous().to(dev)
                     w1 = l.ffn.filter_layer.weight.detach().float()

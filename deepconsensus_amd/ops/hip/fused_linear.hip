@@ -8,7 +8,8 @@
 // outputs staged through LDS for coalesced 16-B global writes. Optional
 // fused bias, ReLU, and ReZero residual (out = x + alpha*y).
 //
-// W arrives host-padded to [Npad, 288] with Npad a multiple of 64.
+// W arrives host-padded to [Npad, 296] (= the LDS row stride, so glds
+// streams chunks as raw row-major copies) with Npad a multiple of 64.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -79,39 +80,32 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
   }
   __syncthreads();
 
-  // T14 staging registers (named: an array would be promoted to LDS).
-  uint4 sr0, sr1, sr2, sr3, sr4;
-
+  // Weights stream by glds (wave-uniform 1-KiB LDS chunks, per-lane
+  // sources): the T14 register staging this replaces cost ~20 VGPRs and
+  // held the kernel at 148 regs / 3 waves per SIMD. W arrives host-padded
+  // to the LDS row stride (296), so a chunk is one raw row-major copy.
   auto issue_w = [&](int chunk) {
+    const bf16* src = w + (size_t)chunk * NC * W_STRIDE;
 #pragma unroll
-    for (int i = 0; i < G_PER_T; ++i) {
-      const int g = min(tid + i * 512, WG - 1);
-      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
-      const uint4 v = *reinterpret_cast<const uint4*>(
-          w + (size_t)(chunk * NC + row) * K1P + 8 * k8);
-      if (i == 0) sr0 = v; else if (i == 1) sr1 = v;
-      else if (i == 2) sr2 = v; else if (i == 3) sr3 = v; else sr4 = v;
-    }
-  };
-  auto write_w = [&]() {
-#pragma unroll
-    for (int i = 0; i < G_PER_T; ++i) {
-      const int g = min(tid + i * 512, WG - 1);
-      const int row = g / (K1P / 8), k8 = g % (K1P / 8);
-      *reinterpret_cast<uint4*>(&w_lds[row * W_STRIDE + 8 * k8]) =
-          (i == 0 ? sr0 : i == 1 ? sr1 : i == 2 ? sr2
-           : i == 3 ? sr3 : sr4);
+    for (int i = 0; i < 5; ++i) {
+      const int ck = wave + i * 8;
+      if (ck < NC * W_STRIDE * 2 / 1024) {
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned*)(
+                src + ck * 512 + lane * 8),
+            (__attribute__((address_space(3))) unsigned*)(
+                &w_lds[ck * 512]),
+            16, 0, 0);
+      }
     }
   };
 
   issue_w(0);
-  write_w();
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
 
   const int nchunk = Npad / NC;
   for (int chunk = 0; chunk < nchunk; ++chunk) {
-    __syncthreads();  // w_lds chunk ready; o_lds free
-    if (chunk + 1 < nchunk) issue_w(chunk + 1);
-
     const int colt = 32 * ch;
     const int ncol = chunk * NC + colt + c;  // this lane's output col
     f32x16 acc = {};
@@ -133,7 +127,7 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
       o_lds[row * O_STRIDE + colt + c] = __float2bfloat16(v);
     }
     __syncthreads();  // o_lds complete; w_lds consumed
-    if (chunk + 1 < nchunk) write_w();
+    if (chunk + 1 < nchunk) issue_w(chunk + 1);
 
     // Coalesced copy-out of this 64-col chunk (+ optional residual).
     const int n0 = chunk * NC;
@@ -166,6 +160,13 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
         }
       }
     }
+    // Next chunk's weights must have landed (vmcnt also drains this
+    // chunk's copy-out stores, which is harmless) and every wave must be
+    // done with o_lds before the next epilogue rewrites it.
+    if (chunk + 1 < nchunk) {
+      asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
   }
 }
 
@@ -182,8 +183,8 @@ at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int M = xc.numel() / K;
   TORCH_CHECK(K == K1, "fused_linear requires width-280 input");
   const int Npad = w.size(0);
-  TORCH_CHECK(w.size(1) == K1P && Npad % NC == 0,
-              "w must be [Npad (mult of 64), 288]");
+  TORCH_CHECK(w.size(1) == W_STRIDE && Npad % NC == 0,
+              "w must be [Npad (mult of 64), 296]");
   const int N = (int)n_out;
   TORCH_CHECK(N <= Npad, "n_out exceeds padded weight rows");
   TORCH_CHECK(!residual || N == K1, "residual requires N == 280");

@@ -77,8 +77,8 @@ def main():
     if args.kernel in ("all", "linear"):
         M = B * L
         x = torch.randn(M, 280, device="cuda").to(torch.bfloat16) * 0.3
-        wqkv = torch.randn(896, 288, device="cuda").to(torch.bfloat16) * 0.05
-        wout = torch.randn(320, 288, device="cuda").to(torch.bfloat16) * 0.05
+        wqkv = torch.randn(896, 296, device="cuda").to(torch.bfloat16) * 0.05
+        wout = torch.randn(320, 296, device="cuda").to(torch.bfloat16) * 0.05
         empty = x.new_empty(0)
         us = timeit(lambda: ext.fused_linear(x, wqkv, empty, empty, 840,
                                              False, 0.0), args.iters)

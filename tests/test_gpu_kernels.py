@@ -372,7 +372,7 @@ def test_fused_linear_matches_torch(setup):
         x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
         # N=840 (QKV shape), no bias.
         w = torch.randn(840, 280, device="cuda") * 0.05
-        w_pad = torch.zeros(896, 288, device="cuda")
+        w_pad = torch.zeros(896, 296, device="cuda")
         w_pad[:840, :280] = w
         w_pad = w_pad.to(torch.bfloat16).contiguous()
         out = runner.ext.fused_linear(
@@ -383,7 +383,7 @@ def test_fused_linear_matches_torch(setup):
         # N=320-padded 280 with bias + relu.
         w2 = torch.randn(280, 280, device="cuda") * 0.05
         b2 = torch.randn(280, device="cuda")
-        w2_pad = torch.zeros(320, 288, device="cuda")
+        w2_pad = torch.zeros(320, 296, device="cuda")
         w2_pad[:280, :280] = w2
         w2_pad = w2_pad.to(torch.bfloat16).contiguous()
         out2 = runner.ext.fused_linear(
