@@ -14,6 +14,15 @@ import torch
 
 
 class LAMB(torch.optim.Optimizer):
+    """LAMB with a batched (_foreach) step.
+
+    ``capturable=True`` keeps the step count, bias corrections and
+    learning rate on-device (0-dim tensors) so the whole step can live
+    inside a hipGraph capture; the LR schedule then updates each group's
+    ``lr_t`` tensor in place between replays (PolynomialWarmupSchedule
+    .apply does this automatically when the tensor exists).
+    """
+
     def __init__(
         self,
         params,
@@ -21,10 +30,12 @@ class LAMB(torch.optim.Optimizer):
         betas: Tuple[float, float] = (0.9, 0.999),
         eps: float = 1e-6,
         weight_decay: float = 0.0,
+        capturable: bool = False,
     ):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self.capturable = capturable
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -49,7 +60,11 @@ class LAMB(torch.optim.Optimizer):
             if not ps:
                 continue
             grads = [p.grad for p in ps]
-            ms, vs, bc1, bc2 = [], [], [], []
+            ms, vs = [], []
+            if self.capturable and "step_t" not in group:
+                dev = ps[0].device
+                group["step_t"] = torch.zeros((), device=dev)
+                group["lr_t"] = torch.full((), lr, device=dev)
             for p in ps:
                 state = self.state[p]
                 if len(state) == 0:
@@ -57,15 +72,22 @@ class LAMB(torch.optim.Optimizer):
                     state["exp_avg"] = torch.zeros_like(p)
                     state["exp_avg_sq"] = torch.zeros_like(p)
                 state["step"] += 1
-                t = state["step"]
                 ms.append(state["exp_avg"])
                 vs.append(state["exp_avg_sq"])
-                bc1.append(1.0 - beta1**t)
-                bc2.append(1.0 - beta2**t)
             torch._foreach_mul_(ms, beta1)
             torch._foreach_add_(ms, grads, alpha=1 - beta1)
             torch._foreach_mul_(vs, beta2)
             torch._foreach_addcmul_(vs, grads, grads, value=1 - beta2)
+            if self.capturable:
+                group["step_t"] += 1
+                t = group["step_t"]
+                bc1 = [1.0 - beta1**t] * len(ps)
+                bc2 = [1.0 - beta2**t] * len(ps)
+                neg_lr = -group["lr_t"]
+            else:
+                bc1 = [1.0 - beta1 ** self.state[p]["step"] for p in ps]
+                bc2 = [1.0 - beta2 ** self.state[p]["step"] for p in ps]
+                neg_lr = -lr
             m_hat = torch._foreach_div(ms, bc1)
             denom = torch._foreach_div(vs, bc2)
             denom = torch._foreach_sqrt(denom)
@@ -79,7 +101,7 @@ class LAMB(torch.optim.Optimizer):
                 (w_norm > 0) & (u_norm > 0),
                 w_norm / u_norm,
                 torch.ones_like(w_norm),
-            ) * (-lr)
+            ) * neg_lr
             torch._foreach_mul_(update, list(ratio.unbind()))
             torch._foreach_add_(ps, update)
         return loss
@@ -135,10 +157,13 @@ class PolynomialWarmupSchedule:
         lr = self(step)
         for g in optimizer.param_groups:
             g["lr"] = lr
+            if "lr_t" in g:  # capturable mode: graph reads this tensor
+                g["lr_t"].fill_(lr)
         return lr
 
 
-def create_optimizer(params, decay_steps: int, model: torch.nn.Module):
+def create_optimizer(params, decay_steps: int, model: torch.nn.Module,
+                     capturable: bool = False):
     """Factory (model_utils.py:621-669): returns (optimizer, schedule)."""
     groups = build_param_groups(model, params.weight_decay_rate)
     opt = LAMB(
@@ -146,6 +171,7 @@ def create_optimizer(params, decay_steps: int, model: torch.nn.Module):
         lr=params.initial_learning_rate,
         betas=(params.beta_1, params.beta_2),
         eps=params.epsilon,
+        capturable=capturable,
     )
     sched = PolynomialWarmupSchedule(
         initial_learning_rate=params.initial_learning_rate,

@@ -180,3 +180,30 @@ def test_train_dp2_gloo(tmp_path, train_data):
         nprocs=2, join=True,
     )
     assert glob.glob(os.path.join(out_dir, "checkpoint-*.pt"))
+
+
+def test_lamb_capturable_matches_standard():
+    """capturable=True (device-tensor lr/step) matches the standard step."""
+    import torch
+
+    from deepconsensus_amd.models import lamb as lamb_lib
+
+    torch.manual_seed(0)
+    net_a = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+    net_b = __import__("copy").deepcopy(net_a)
+    opt_a = lamb_lib.LAMB(net_a.parameters(), lr=0.01, weight_decay=0.02)
+    opt_b = lamb_lib.LAMB(net_b.parameters(), lr=0.01, weight_decay=0.02,
+                          capturable=True)
+    sched_a = lamb_lib.PolynomialWarmupSchedule(0.01, 0.001, 20, 5)
+    sched_b = lamb_lib.PolynomialWarmupSchedule(0.01, 0.001, 20, 5)
+    x = torch.randn(16, 8)
+    y = torch.randn(16, 4)
+    for step in range(6):
+        for net, opt, sched in ((net_a, opt_a, sched_a),
+                                (net_b, opt_b, sched_b)):
+            sched.apply(opt, step)
+            opt.zero_grad()
+            ((net(x) - y) ** 2).mean().backward()
+            opt.step()
+    for pa, pb in zip(net_a.parameters(), net_b.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), (pa - pb).abs().max()
