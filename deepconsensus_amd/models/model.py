@@ -126,10 +126,9 @@ class BandedSelfAttention(nn.Module):
         q = q * (d**-0.5)
         logits = torch.matmul(q, k.transpose(-1, -2))  # [B,H,T,T]
         mask = self.band_mask[:t, :t]
-        logits = torch.where(
-            mask, logits, torch.tensor(-1e9, dtype=logits.dtype,
-                                       device=logits.device)
-        )
+        # masked_fill with a python scalar (a torch.tensor(-1e9) here would
+        # issue an H2D copy per call and break hipGraph capture).
+        logits = logits.masked_fill(~mask, -1e9)
         # Softmax in fp32 for stability (reference attention_layer.py:208-211).
         weights = torch.softmax(logits.float(), dim=-1).to(x.dtype)
         if training and self.dropout > 0:
