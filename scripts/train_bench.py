@@ -30,8 +30,6 @@ def main():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--bf16", action="store_true")
-    ap.add_argument("--graph", action="store_true",
-                    help="hipGraph-capture the whole training step")
     args = ap.parse_args()
 
     rank, world = comm.init_distributed()
@@ -46,9 +44,7 @@ def main():
     torch.manual_seed(7)
     model = get_model(params).to(device)
     comm.broadcast_parameters(model)
-    optimizer, schedule = lamb_lib.create_optimizer(
-        params, 10000, model, capturable=args.graph and have_cuda
-    )
+    optimizer, schedule = lamb_lib.create_optimizer(params, 10000, model)
     reducer = comm.FlatGradAllreducer(model)
     loss_fn = losses_lib.AlignmentLoss(
         del_cost=params.del_cost, loss_reg=params.loss_reg, reduction="sum"
@@ -86,26 +82,6 @@ def main():
     def step(i):
         schedule.apply(optimizer, i)
         return inner_step()
-
-    graph = None
-    if args.graph and have_cuda and world == 1:
-        # Eager warmup initializes optimizer state + allocator, then the
-        # whole step (fwd + loss + bwd + LAMB) is captured; the LR
-        # schedule keeps updating lr_t between replays.
-        for i in range(3):
-            step(i)
-        torch.cuda.synchronize()
-        schedule.apply(optimizer, 3)
-        graph = torch.cuda.CUDAGraph()
-        # Capture the step WITHOUT schedule.apply: its lr_t.fill_ would
-        # otherwise be recorded as a constant fill inside the graph.
-        with torch.cuda.graph(graph):
-            static_loss = inner_step()
-
-        def step(i):  # noqa: F811 - replay replaces the eager step
-            schedule.apply(optimizer, i)
-            graph.replay()
-            return static_loss
 
     for i in range(args.warmup):
         step(i)
