@@ -61,7 +61,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch-size", type=int, default=8192)
+    ap.add_argument("--batch-size", type=int, default=16384)
     ap.add_argument("--pool-batches", type=int, default=4)
     ap.add_argument("--no-graphs", action="store_true",
                     help="disable hipGraph capture of the serving step")
