@@ -459,3 +459,41 @@ def test_example_codec_fuzz():
                 assert list(dvals) == list(vals)
 
     check()
+
+
+def test_bam_header_larger_than_bgzf_block(tmp_path):
+    """Real subreads_to_ccs.bam headers carry one reference per ZMW and
+    exceed the 64 KB BGZF block payload; the header must span blocks."""
+    refs = [(f"m0/{i}/ccs", 10000) for i in range(20000)]  # ~400 KB header
+    header = bam_lib.BamHeader(text="@HD\tVN:1.6", references=refs)
+    p = str(tmp_path / "big.bam")
+    with bam_lib.BamWriter(p, header) as w:
+        w.write(bam_lib.BamRead(
+            qname="m0/19999/0_100", flag=0, ref_id=19999, pos=0, mapq=60,
+            cigartuples=[(0, 100)], seq="A" * 100,
+            query_qualities=[30] * 100, tags={"zm": 19999},
+        ))
+    rd = bam_lib.BamReader(p)
+    assert len(rd.header.references) == 20000
+    assert next(iter(rd)).reference_name == "m0/19999/ccs"
+
+
+def test_stitch_min_quality_and_min_length_filters():
+    """Reads below min_quality / min_length are filtered with the right
+    outcome counters (stitch_utils.py:101-121)."""
+    lowq = phred.quality_scores_to_string(np.full(50, 5))
+    outs = [stitch.DCModelOutput(molecule_name="m", window_pos=0,
+                                 sequence="A" * 50, quality_string=lowq)]
+    counter = stitch.OutcomeCounter()
+    assert stitch.stitch_to_fastq("m", outs, max_length=50, min_quality=20,
+                                  min_length=1,
+                                  outcome_counter=counter) is None
+    assert counter.failed_quality_filter == 1
+    okq = phred.quality_scores_to_string(np.full(50, 40))
+    outs2 = [stitch.DCModelOutput(molecule_name="m", window_pos=0,
+                                  sequence="A" * 50, quality_string=okq)]
+    counter2 = stitch.OutcomeCounter()
+    assert stitch.stitch_to_fastq("m", outs2, max_length=50, min_quality=20,
+                                  min_length=100,
+                                  outcome_counter=counter2) is None
+    assert counter2.failed_length_filter == 1
