@@ -207,3 +207,23 @@ def test_lamb_capturable_matches_standard():
             opt.step()
     for pa, pb in zip(net_a.parameters(), net_b.parameters()):
         assert torch.allclose(pa, pb, atol=1e-6), (pa - pb).abs().max()
+
+
+def test_train_resume_from_checkpoint(tmp_path, train_data):
+    """A second train_model in the same out_dir resumes from the latest
+    checkpoint and continues the step counter."""
+    from deepconsensus_amd.models import checkpoint as ckpt_lib
+    from deepconsensus_amd.models import train as train_lib
+
+    train_file, _ = train_data
+    params = _tiny_params(train_file)
+    out_dir = str(tmp_path / "model")
+    s1 = train_lib.train_model(out_dir, params, device="cpu",
+                               eval_every=2, limit_steps=2)
+    path1, epoch1, step1 = ckpt_lib.get_checkpoint_and_initial_epoch(out_dir)
+    assert path1 and step1 == s1["steps"] >= 2
+    s2 = train_lib.train_model(out_dir, params, device="cpu",
+                               eval_every=2, limit_steps=2)
+    assert s2["steps"] >= s1["steps"] + 2  # continued, not restarted
+    _, _, step2 = ckpt_lib.get_checkpoint_and_initial_epoch(out_dir)
+    assert step2 == s2["steps"]
