@@ -162,3 +162,40 @@ def test_run_ccs_fasta_deprecated():
         cli.main(["run", "--subreads_to_ccs", "x.bam", "--ccs_bam", "y.bam",
                   "--checkpoint", "random", "--output", "o.fastq",
                   "--ccs_fasta", "z.fa"])
+
+
+def test_run_accepts_exported_bundle(tmp_path):
+    """`deepconsensus run` serves from an exported bundle directory
+    (reference: SavedModel-or-checkpoint detection, quick_inference
+    .py:797-800)."""
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from test_io_and_pipeline import make_test_bams
+
+    from deepconsensus_amd.dcio.fastq import read_fastq
+    from deepconsensus_amd.inference import quick_inference as qi
+    from deepconsensus_amd.models import checkpoint as ckpt_lib
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models import export_model
+    from deepconsensus_amd.models.model import get_model
+
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params)
+    torch.manual_seed(2)
+    model = get_model(params)
+    ckpt_dir = str(tmp_path / "ckpt")
+    ckpt_lib.save_checkpoint(ckpt_dir, 1, 0, model, None, params)
+    bundle = str(tmp_path / "bundle")
+    export_model.export(ckpt_dir, bundle)
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=2, length=150, seed=9)
+    out = str(tmp_path / "out.fastq")
+    counter = qi.run(
+        subreads_to_ccs=sub, ccs_bam=ccs, checkpoint=bundle, output=out,
+        options=qi.InferenceOptions(batch_size=8, batch_zmws=2, cpus=0,
+                                    min_quality=0, skip_windows_above=0),
+        device="cpu",
+    )
+    assert counter.total == 2
+    assert len(list(read_fastq(out))) == 2
