@@ -57,6 +57,10 @@ def export(checkpoint_path: str, out_dir: str, device: str = "cpu") -> str:
         2, params.total_rows, params.max_length, device=device
     )
     with torch.no_grad():
+        # torch.jit.trace is deprecated upstream in favor of torch.compile/
+        # torch.export, but compile is Triton-backed (excluded on this
+        # CDNA4-native build) and jit remains the dependency-free serialized
+        # form loadable with plain torch.jit.load.
         traced = torch.jit.trace(wrapper, example)
     traced_path = os.path.join(out_dir, "serving_model.pt")
     traced.save(traced_path)
