@@ -264,6 +264,13 @@ def test_preprocess_cli_e2e(tmp_path):
     ])
     recs2 = list(tfrecord.read_tfrecords(out2))
     assert len(recs2) == 9
+    # Pool mode produces the same records (order may differ through the
+    # writer queue) and the same summary counters as serial mode.
+    assert sorted(recs) == sorted(recs2)
+    summary2 = json.load(open(str(tmp_path / "par" / "ex.inference.json")))
+    assert {k: v for k, v in summary2.items() if k != "cpus"} == {
+        k: v for k, v in summary.items() if k != "cpus"
+    }
 
     # Records parse through the data layer.
     params = cfg.get_config("transformer_learn_values+custom")
