@@ -504,3 +504,24 @@ def test_stitch_min_quality_and_min_length_filters():
                                   min_length=100,
                                   outcome_counter=counter2) is None
     assert counter2.failed_length_filter == 1
+
+
+def test_quick_inference_pool_matches_serial(tmp_path):
+    """cpus=2 produces byte-identical FASTQ to cpus=0 (same model seed)."""
+    import torch
+
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=3, length=150, seed=33)
+    outs = []
+    for cpus in (0, 2):
+        out = str(tmp_path / f"out{cpus}.fastq")
+        torch.manual_seed(123)
+        qi.run(subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+               output=out,
+               options=qi.InferenceOptions(batch_size=8, batch_zmws=2,
+                                           cpus=cpus, min_quality=0,
+                                           skip_windows_above=0),
+               device="cpu")
+        outs.append(open(out, "rb").read())
+    assert outs[0] == outs[1]
