@@ -525,3 +525,19 @@ def test_quick_inference_pool_matches_serial(tmp_path):
                device="cpu")
         outs.append(open(out, "rb").read())
     assert outs[0] == outs[1]
+
+
+def test_stitch_fill_n_for_missing_windows():
+    """fill_n=True inserts N*max_length with EMPTY_QUAL for gaps
+    (stitch_utils.py:61-74)."""
+    outs = [
+        stitch.DCModelOutput(molecule_name="m", window_pos=0,
+                             sequence="AAAAA", quality_string="IIIII"),
+        stitch.DCModelOutput(molecule_name="m", window_pos=10,
+                             sequence="GGGGG", quality_string="IIIII"),
+    ]
+    seq, qual = stitch.get_full_sequence(outs, max_length=5, fill_n=True)
+    assert seq == "AAAAA" + "N" * 5 + "GGGGG"
+    assert qual[5:10] == phred.quality_scores_to_string(np.zeros(5))
+    seq2, qual2 = stitch.get_full_sequence(outs, max_length=5)
+    assert seq2 is None and qual2 == ""
