@@ -317,3 +317,14 @@ def test_dc_example_tf_example_round_trip():
         feats["subreads"], w.extract_features()
     )
     assert feats["subreads/num_passes"] == 3
+
+
+def test_dc_config_from_shape_invalid_raises():
+    """Decoding an 86-row (ccs_bq) tensor with use_ccs_bq=False raises
+    (preprocess_test.py::test_invalid_tf_examples semantics)."""
+    from deepconsensus_amd.preprocess.windows import dc_config_from_shape
+
+    ok = dc_config_from_shape((86, 100, 1), use_ccs_bq=True)
+    assert ok.max_passes == 20
+    with pytest.raises(ValueError, match="Invalid subreads shape"):
+        dc_config_from_shape((86, 100, 1), use_ccs_bq=False)
