@@ -558,3 +558,28 @@ def test_native_forward_deterministic(setup):
     assert torch.equal(b1.cpu(), b2.cpu())
     assert torch.equal(q1.cpu(), q2.cpu())
     assert torch.equal(p1.cpu(), p2.cpu())
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("env", [
+    {"DC_FUSED_FFN": "0"},                      # hipBLASLt fallback
+    {"DC_FFN_V3": "0"},                         # v2 glds FFN
+    {"DC_FFN_V3": "0", "DC_FFN_V2": "0"},       # v1 T14 FFN
+])
+def test_ffn_knobs_agree(monkeypatch, env):
+    """Every DC_FUSED_FFN / DC_FFN_V* setting produces the same calls as
+    the default path (the knobs select implementations, not semantics)."""
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=False)
+    torch.manual_seed(41)
+    model = get_model(params)
+    rows = _make_rows(params)
+    base = InferenceRunner(params, model, device="cuda:0")
+    b0, q0 = base.forward_windows(rows)
+    for k, v in env.items():
+        monkeypatch.setenv(k, v)
+    alt = InferenceRunner(params, model, device="cuda:0")
+    b1, q1 = alt.forward_windows(rows)
+    agree = (b0.cpu() == b1.cpu()).float().mean()
+    assert agree > 0.995, float(agree)  # bf16 path differences only
+    assert (q0.cpu().float() - q1.cpu().float()).abs().mean() < 0.5
