@@ -3,10 +3,10 @@
 // The K5 QKV projection and K7 output projection (attention_layer.py:78-107)
 // as one tiled MFMA kernel reusing the fused-FFN structure: 512 threads =
 // 8 waves per 128-row tile, K fixed at the model width 280 (padded 288),
-// N processed in 64-column chunks with T14 async weight staging (issue
-// global loads to named registers before the MFMA cluster, ds_write after),
-// outputs staged through LDS for coalesced 16-B global writes. Optional
-// fused bias, ReLU, and ReZero residual (out = x + alpha*y).
+// N processed in 64-column chunks with glds weight streaming (120 VGPRs
+// -> 4 waves/SIMD occupancy), outputs staged through LDS for coalesced
+// 16-B global writes. Optional fused bias, ReLU, and ReZero residual
+// (out = x + alpha*y).
 //
 // W arrives host-padded to [Npad, 296] (= the LDS row stride, so glds
 // streams chunks as raw row-major copies) with Npad a multiple of 64.
@@ -29,8 +29,6 @@ constexpr int K1P = 288;
 constexpr int NC = 64;
 constexpr int W_STRIDE = 296;   // weight image row stride (conflict-free)
 constexpr int O_STRIDE = 72;    // output chunk image stride
-constexpr int WG = NC * (K1P / 8);   // granules per weight slice (2304)
-constexpr int G_PER_T = 5;
 
 template <bool RELU, bool RESIDUAL>
 __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
