@@ -106,6 +106,18 @@ def main():
         print(f"pipeline: {args.zmws / run_s:.2f} ZMW/s wall "
               f"({run_s:.1f}s, success={counter.success}, "
               f"cpus={args.cpus}, batch={args.batch_size})")
+        # Per-stage main-thread time (preprocess overlaps via the prefetch
+        # thread, so its visible share should be ~0 when pipelining works).
+        stage_s = {}
+        with open(os.path.join(td, "out.runtime.csv")) as f:
+            next(f)
+            for line in f:
+                parts = line.strip().split(",")
+                stage_s[parts[1]] = stage_s.get(parts[1], 0.0) + float(
+                    parts[2]
+                )
+        print("stage seconds (main thread): " + ", ".join(
+            f"{k}={v:.2f}" for k, v in sorted(stage_s.items())))
 
 
 if __name__ == "__main__":
