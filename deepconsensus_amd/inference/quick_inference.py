@@ -118,12 +118,23 @@ def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
     if stage == DebugStage.DC_INPUT:
         return [], dc_whole.counter
     feature_dicts = [x.to_features_dict() for x in dc_whole.iter_examples()]
-    # Features are integral-after-truncation for the model (the embedding
-    # casts to int; SN fractions truncate identically) and <= SN_MAX, so
-    # ship them across the process boundary as int16 — halves the pickle
-    # volume coming back from the worker pool.
+    # Worker-side finishing, keeping the serial batch loop thin:
+    #  * format_rows (PW/IP/SN clipping) runs HERE, in parallel — the
+    #    main thread then only stacks ("fmt" marks it done);
+    #  * features are integral-after-truncation for the model (the
+    #    embedding casts to int; SN fractions truncate identically) and
+    #    <= SN_MAX after clipping, so they cross the process boundary as
+    #    int16 — halving the pickle volume.
+    fmt_params = cfg.Params(
+        max_passes=dc_config.max_passes,
+        use_ccs_bq=dc_config.use_ccs_bq,
+        total_rows=dc_config.tensor_height,
+        PW_MAX=255, IP_MAX=255, SN_MAX=500,  # config.py:89-92 defaults
+    )
     for f in feature_dicts:
-        f["subreads"] = np.asarray(f["subreads"]).astype(np.int16)
+        rows = data_lib.format_rows(np.asarray(f["subreads"]), fmt_params)
+        f["subreads"] = rows.astype(np.int16)
+        f["fmt"] = True
     return feature_dicts, dc_whole.counter
 
 
@@ -180,9 +191,11 @@ def run_model_on_examples(
         chunk = feature_dicts[i : i + options.batch_size]
         rows = np.stack(
             [
-                data_lib.format_rows(np.asarray(f["subreads"]), params)[
-                    :, :, 0
-                ]
+                np.asarray(f["subreads"])[:, :, 0]
+                if f.get("fmt")
+                else data_lib.format_rows(
+                    np.asarray(f["subreads"]), params
+                )[:, :, 0]
                 for f in chunk
             ]
         ).astype(np_dtype)
