@@ -478,17 +478,26 @@ def run(
     pending: Optional[Tuple[Sequence, Any]] = None
     n_batches = 0
     with concurrent.futures.ThreadPoolExecutor(1) as prefetcher:
+
+        def drain(prev_inputs, prev_fut, name):
+            # Time spent blocked on the overlapped preprocessing (not a
+            # reference stage; extends the CSV so wall time is accounted).
+            before = time.time()
+            outputs = prev_fut.result()
+            timelog.add("wait_preprocess", name, before, None, None,
+                        len(prev_inputs))
+            infer_batch(prev_inputs, outputs, name)
+
         for inputs in batch_iter:
             fut = prefetcher.submit(preprocess_batch, inputs)
             if pending is not None:
                 prev_inputs, prev_fut = pending
-                infer_batch(prev_inputs, prev_fut.result(),
-                            f"batch {n_batches}")
+                drain(prev_inputs, prev_fut, f"batch {n_batches}")
                 n_batches += 1
             pending = (inputs, fut)
         if pending is not None:
             prev_inputs, prev_fut = pending
-            infer_batch(prev_inputs, prev_fut.result(), f"batch {n_batches}")
+            drain(prev_inputs, prev_fut, f"batch {n_batches}")
             n_batches += 1
 
     if pool is not None:
