@@ -187,8 +187,9 @@ def test_quick_inference_e2e_fastq(tmp_path):
         output=out, options=options, device="cpu",
     )
     assert counter.total == 3
-    # Runtime CSV + stats JSON written.
-    assert os.path.exists(str(tmp_path / "out.runtime.csv"))
+    # Runtime CSV + stats JSON written (incl. the prefetch-wait stage).
+    runtime = open(str(tmp_path / "out.runtime.csv")).read()
+    assert "wait_preprocess" in runtime and "run_model" in runtime
     stats = json.load(open(tmp_path / "out.inference.json"))
     assert stats["n_zmw_processed"] == 3
     # Every ZMW produced output (min_quality=0 disables filtering).
