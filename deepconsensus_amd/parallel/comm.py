@@ -80,16 +80,31 @@ class FlatGradAllreducer:
         total = sum(p.numel() for p in self.params)
         device = self.params[0].device if self.params else "cpu"
         self.flat = torch.zeros(total, dtype=dtype, device=device)
+        self._views = []
         offset = 0
         for p in self.params:
             n = p.numel()
-            p.grad = self.flat[offset:offset + n].view_as(p)
+            view = self.flat[offset:offset + n].view_as(p)
+            p.grad = view
+            self._views.append(view)
             offset += n
 
     def zero_(self):
         self.flat.zero_()
 
+    def _rebind(self):
+        # Defensive: if anything reassigned p.grad (zero_grad(set_to_none),
+        # a library hook), fold the detached gradient back into its flat
+        # view so the bucket reduce never silently misses a tensor.
+        for p, view in zip(self.params, self._views):
+            if p.grad is None:
+                p.grad = view
+            elif p.grad.data_ptr() != view.data_ptr():
+                view.copy_(p.grad)
+                p.grad = view
+
     def reduce(self):
+        self._rebind()
         if dist.is_initialized():
             dist.all_reduce(self.flat, op=dist.ReduceOp.SUM)
             self.flat.div_(dist.get_world_size())

@@ -51,3 +51,19 @@ def _worker(rank, world, port):
 def test_comm_collectives_gloo_world2():
     torch.multiprocessing.spawn(_worker, args=(2, 29881), nprocs=2,
                                 join=True)
+
+
+def test_flat_reducer_rebinds_detached_grads():
+    """A reassigned p.grad is folded back into the bucket on reduce()."""
+    from deepconsensus_amd.parallel import comm
+
+    net = torch.nn.Linear(3, 3)
+    reducer = comm.FlatGradAllreducer(net)
+    reducer.zero_()
+    net.weight.grad = torch.full_like(net.weight, 2.0)  # detached
+    net.bias.grad = None
+    reducer.reduce()  # world 1: no collective, but rebinding runs
+    assert net.weight.grad.data_ptr() == reducer._views[0].data_ptr()
+    assert torch.all(net.weight.grad == 2.0)
+    assert net.bias.grad is not None
+    assert float(reducer.grad_norm()) > 0
