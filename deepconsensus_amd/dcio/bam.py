@@ -398,13 +398,45 @@ def _encode_tags(tags: Dict[str, Any]) -> bytes:
 # ---------------------------------------------------------------------------
 
 
-class BamReader:
-    """Sequential BAM reader."""
+def _threaded_blocks(gen: Iterator[bytes], depth: int = 64
+                     ) -> Iterator[bytes]:
+    """Readahead thread: BGZF read+decompress runs ahead of the record
+    parser (the serial ZMW feeder's floor — profiles/r01_perf_journal)."""
+    import queue
+    import threading
 
-    def __init__(self, path: str):
+    q: "queue.Queue" = queue.Queue(maxsize=depth)
+    _END = object()
+
+    def pump():
+        try:
+            for block in gen:
+                q.put(block)
+            q.put(_END)
+        except BaseException as e:  # surface decode errors to the consumer
+            q.put(e)
+
+    t = threading.Thread(target=pump, daemon=True)
+    t.start()
+    while True:
+        item = q.get()
+        if item is _END:
+            return
+        if isinstance(item, BaseException):
+            raise item
+        yield item
+
+
+class BamReader:
+    """Sequential BAM reader (with a decompression readahead thread)."""
+
+    def __init__(self, path: str, reader_threads: int = 1):
         self.path = path
         self._fh = open(path, "rb")
-        self._stream = _ConcatStream(_read_bgzf_blocks(self._fh))
+        blocks = _read_bgzf_blocks(self._fh)
+        if reader_threads and reader_threads > 0:
+            blocks = _threaded_blocks(blocks)
+        self._stream = _ConcatStream(blocks)
         magic = self._stream.read(4)
         if magic != BAM_MAGIC:
             raise ValueError(f"{path} is not a BAM file")

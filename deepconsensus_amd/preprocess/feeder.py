@@ -27,8 +27,9 @@ class SubreadGrouper:
     """Yields all mapped subreads of one ZMW (zm-tag grouped, streaming)."""
 
     def __init__(self, subreads_to_ccs: str, reader_threads: int = 1):
-        del reader_threads  # the pure-python reader is single-threaded
-        self.bam_reader = iter(bam.BamReader(subreads_to_ccs))
+        self.bam_reader = iter(
+            bam.BamReader(subreads_to_ccs, reader_threads=reader_threads)
+        )
         self.keep_iter = True
         self.subread_group: List[bam.BamRead] = []
         first_read = next(self.bam_reader)
