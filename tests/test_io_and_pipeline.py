@@ -542,3 +542,55 @@ def test_stitch_fill_n_for_missing_windows():
     assert qual[5:10] == phred.quality_scores_to_string(np.zeros(5))
     seq2, qual2 = stitch.get_full_sequence(outs, max_length=5)
     assert seq2 is None and qual2 == ""
+
+
+def test_quick_inference_mixed_skip_and_model_windows(tmp_path):
+    """A ZMW whose windows split between the model path and the CCS skip
+    path still stitches into one complete read (the skipped predictions
+    are appended out of order; _write_outputs must sort before grouping)."""
+    from deepconsensus_amd.dcio import bam as bam_lib
+    from deepconsensus_amd.dcio.fastq import read_fastq
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    rng = np.random.default_rng(55)
+    length = 250  # 3 windows per ZMW
+    name = "m000/10/ccs"
+    seq = _random_seq(rng, length)
+    header = bam_lib.BamHeader(text="@HD\tVN:1.6", references=[(name, length)])
+    sub = str(tmp_path / "subreads_to_ccs.bam")
+    with bam_lib.BamWriter(sub, header) as w:
+        for i in range(3):
+            w.write(bam_lib.BamRead(
+                qname=f"m000/10/{i * 300}_{i * 300 + length}",
+                flag=16 if i % 2 else 0, ref_id=0, pos=0, mapq=60,
+                cigartuples=[(0, length)], seq=seq,
+                query_qualities=[30] * length,
+                tags={"zm": 10,
+                      "pw": rng.integers(0, 60, length).astype(np.uint8),
+                      "ip": rng.integers(0, 60, length).astype(np.uint8),
+                      "sn": np.array([6.0, 7.0, 5.5, 9.1], np.float32)},
+            ))
+    ccs = str(tmp_path / "ccs.bam")
+    # CCS qualities: window 0 (0..99) high (skipped via skip_windows_above),
+    # windows 1-2 low (routed through the model).
+    quals = np.full(length, 10)
+    quals[:100] = 60
+    with bam_lib.BamWriter(ccs, header) as w:
+        w.write(bam_lib.BamRead(
+            qname=name, flag=4, ref_id=-1, pos=-1, cigartuples=[], seq=seq,
+            query_qualities=quals,
+            tags={"zm": 10, "ec": 9.5, "np": 3, "rq": 0.99, "RG": "rg0"},
+        ))
+    out = str(tmp_path / "out.fastq")
+    counter = qi.run(
+        subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random", output=out,
+        options=qi.InferenceOptions(batch_size=8, batch_zmws=2, cpus=0,
+                                    min_quality=0, skip_windows_above=45),
+        device="cpu",
+    )
+    assert counter.success == 1
+    recs = list(read_fastq(out))
+    assert len(recs) == 1
+    # Window 0 adopted the CCS sequence verbatim (skip path).
+    assert recs[0].sequence[:100] == seq[:100]
+    assert len(recs[0].sequence) >= 100
