@@ -345,6 +345,19 @@ class EncoderOnlyLearnedValuesTransformer(EncoderOnlyTransformer):
         # inputs: [B, L, R] float features; slice per-feature row ranges.
         (bi, pwi, ipi, sti, ci, bqi, sni) = self.indices
         x = inputs.transpose(1, 2)  # [B, R, L]
+        if (
+            torch.is_grad_enabled()
+            and self.training
+            and self.bases_embedding.table.requires_grad
+        ):
+            # Training path: same forward math, fused one-kernel backward
+            # for all tables (models/embed_stack.py).
+            from deepconsensus_amd.models.embed_stack import embed_stack
+
+            emb = embed_stack(self, x.float())
+            if self.condense:
+                emb = self.condenser(emb)
+            return emb
         ids = x.long()
         parts = [
             self._block(self.bases_embedding, ids[:, bi[0]:bi[1]]),

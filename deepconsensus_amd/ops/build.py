@@ -21,6 +21,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "fused_linear.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_ffn_glds.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_ffn_v3.hip"),
+    os.path.join(_OPS_DIR, "hip", "embed_grad.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

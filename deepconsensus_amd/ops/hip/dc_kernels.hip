@@ -183,6 +183,10 @@ at::Tensor fused_ffn_v2(at::Tensor x, at::Tensor w1, at::Tensor w2,
                         at::Tensor b2, double alpha);
 at::Tensor fused_ffn_v3(at::Tensor x, at::Tensor w1, at::Tensor w2,
                         at::Tensor b2, double alpha);
+void embed_grad(at::Tensor rows, at::Tensor grad_out, at::Tensor row_shift,
+                at::Tensor row_vocab, at::Tensor row_tbase,
+                at::Tensor row_width, at::Tensor row_col,
+                at::Tensor row_scale, at::Tensor grad_tables);
 at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
                         at::Tensor resid, int64_t n_out, bool relu,
                         double alpha);
@@ -209,6 +213,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused FFN v2: glds-pipelined weights, bias folded into W1");
   m.def("fused_ffn_v3", &fused_ffn_v3,
         "Fused FFN v3: 256-row tiles, register-resident h (swapped B1)");
+  m.def("embed_grad", &embed_grad,
+        "Fused embedding-stack backward (training, all tables, one pass)");
   m.def("fused_linear", &fused_linear,
         "Fused linear projection (K5/K7): act(xW^T+b)[*alpha+x]");
   m.def("alignment_dp_fwd", &alignment_dp_fwd,

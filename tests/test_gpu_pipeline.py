@@ -124,3 +124,42 @@ def test_native_int16_staging_matches_fp32(tmp_path):
     )
     assert torch.equal(b32.cpu(), b16.cpu())
     assert torch.equal(q32.cpu(), q16.cpu())
+
+
+@pytest.mark.gpu
+def test_embed_grad_kernel_matches_cpu_fallback():
+    """The HIP embed_grad kernel produces the same table gradients as the
+    CPU index_add fallback."""
+    import numpy as np
+
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models.embed_stack import EmbedStackFunction, EmbedMeta
+    from deepconsensus_amd.models.model import get_model
+
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params)
+    torch.manual_seed(5)
+    model = get_model(params)
+    meta = EmbedMeta(model)
+    tables = [getattr(model, a).table for a in meta.table_attrs]
+    rng = np.random.default_rng(6)
+    R, L = params.total_rows, params.max_length
+    rows = np.zeros((8, R, L), dtype=np.float32)
+    mp = params.max_passes
+    rows[:, 0:mp] = rng.integers(0, 5, size=(8, mp, L))
+    rows[:, mp:3 * mp] = rng.integers(0, 256, size=(8, 2 * mp, L))
+    rows[:, 3 * mp:4 * mp] = rng.integers(0, 3, size=(8, mp, L))
+    rows[:, 4 * mp] = rng.integers(0, 5, size=(8, L))
+    rows[:, -4:] = rng.uniform(3.0, 30.0, size=(8, 4, 1))
+    x_cpu = torch.from_numpy(rows)
+
+    def grads_on(device):
+        tabs = [t.detach().to(device).requires_grad_(True) for t in tables]
+        out = EmbedStackFunction.apply(x_cpu.to(device), meta, *tabs)
+        out.square().sum().backward()
+        return [t.grad.cpu() for t in tabs]
+
+    g_cpu = grads_on("cpu")
+    g_gpu = grads_on("cuda:0")
+    for a, b in zip(g_cpu, g_gpu):
+        torch.testing.assert_close(a, b, atol=2e-3, rtol=1e-3)

@@ -130,3 +130,72 @@ def test_gradients_flow(prod_params):
         if "alpha" in name:
             continue  # ReZero alphas start at 0: attn branch grads exist.
         assert param.grad is not None, f"no grad for {name}"
+
+
+def test_embed_stack_fused_backward_matches_autograd():
+    """The fused-backward embedding stack produces the same forward and
+    the same table gradients as the plain autograd chain (CPU fallback)."""
+    import copy
+
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models.model import get_model
+
+    for use_bq in (False, True):
+        params = cfg.get_config("transformer_learn_values+custom")
+        params.use_ccs_bq = use_bq
+        cfg.modify_params(params)
+        torch.manual_seed(3)
+        m1 = get_model(params)
+        m2 = copy.deepcopy(m1)
+        rng = np.random.default_rng(5)
+        R, L = params.total_rows, params.max_length
+        rows = np.zeros((3, R, L), dtype=np.float32)
+        mp = params.max_passes
+        rows[:, 0:mp] = rng.integers(0, 5, size=(3, mp, L))
+        rows[:, mp:3 * mp] = rng.integers(0, 256, size=(3, 2 * mp, L))
+        rows[:, 3 * mp:4 * mp] = rng.integers(0, 3, size=(3, mp, L))
+        rows[:, 4 * mp] = rng.integers(0, 5, size=(3, L))
+        if use_bq:
+            rows[:, 4 * mp + 1] = rng.integers(-1, 94, size=(3, L))
+        rows[:, -4:] = rng.uniform(3.0, 30.0, size=(3, 4, 1))
+        x = torch.from_numpy(rows)
+
+        m1.train()
+        torch.manual_seed(77)  # align dropout draws across both paths
+        out1 = m1(x, training=True)
+        out1.sum().backward()
+
+        # m2: force the plain autograd chain by flipping train mode off
+        # during embed (the gate checks self.training).
+        m2.train()
+        m2_embed_gate = m2.bases_embedding.table.requires_grad
+        assert m2_embed_gate
+        # Disable the fused path via eval-mode embed call semantics:
+        # run forward with the original chain by monkeypatching training.
+        import deepconsensus_amd.models.model as model_mod
+
+        orig = model_mod.EncoderOnlyLearnedValuesTransformer.embed
+
+        def plain_embed(self, inputs):
+            was = self.training
+            self.training = False  # gate off -> original chain
+            try:
+                return orig(self, inputs)
+            finally:
+                self.training = was
+
+        model_mod.EncoderOnlyLearnedValuesTransformer.embed = plain_embed
+        try:
+            torch.manual_seed(77)
+            out2 = m2(x, training=True)
+            out2.sum().backward()
+        finally:
+            model_mod.EncoderOnlyLearnedValuesTransformer.embed = orig
+
+        torch.testing.assert_close(out1, out2, atol=1e-6, rtol=1e-6)
+        for n in ("bases", "pw", "ip", "strand", "sn") + (
+            ("ccs_bq",) if use_bq else ()
+        ):
+            g1 = getattr(m1, f"{n}_embedding").table.grad
+            g2 = getattr(m2, f"{n}_embedding").table.grad
+            torch.testing.assert_close(g1, g2, atol=1e-4, rtol=1e-4)
