@@ -77,9 +77,13 @@ def main():
     distributed = world > 1
 
     have_cuda = torch.cuda.is_available()
-    device = f"cuda:{local_rank}" if have_cuda else "cpu"
+    # Modulo lets a smoke run place multiple ranks on one visible GPU
+    # (RCCL permitting); on a full node device_count == nproc and this is
+    # the identity mapping.
+    dev_idx = local_rank % max(torch.cuda.device_count(), 1) if have_cuda else 0
+    device = f"cuda:{dev_idx}" if have_cuda else "cpu"
     if have_cuda:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(dev_idx)
 
     if distributed:
         import torch.distributed as dist
@@ -90,6 +94,7 @@ def main():
             backend="nccl" if have_cuda else "gloo",
             rank=rank,
             world_size=world,
+            device_id=torch.device(device) if have_cuda else None,
         )
 
     params = cfg.get_config("transformer_learn_values+custom")
