@@ -398,6 +398,67 @@ def _encode_tags(tags: Dict[str, Any]) -> bytes:
 # ---------------------------------------------------------------------------
 
 
+def _read_bgzf_blocks_parallel(fh, threads: int = 4, window: int = 64
+                               ) -> Iterator[bytes]:
+    """Decompresses BGZF blocks on a thread pool, yielding in file order.
+
+    zlib.decompress releases the GIL, so block decompression scales with
+    real threads; the serial part left on the caller is only the raw
+    file reads and the yield. This is the feeder-side answer to the
+    single-core BGZF decode ceiling (ROADMAP: host pipeline item 1).
+    """
+    import collections as _collections
+    import concurrent.futures as _futures
+
+    def read_compressed():
+        """Yields (cdata, isize) without decompressing."""
+        while True:
+            header = fh.read(12)
+            if len(header) == 0:
+                return
+            if len(header) < 12:
+                raise ValueError("truncated BGZF header")
+            magic, _mtime, _xfl, _os, xlen = struct.unpack(
+                "<4sLBBH", header
+            )
+            if magic[:2] != b"\x1f\x8b":
+                raise ValueError("not a gzip stream")
+            extra = fh.read(xlen)
+            bsize = None
+            off = 0
+            while off + 4 <= len(extra):
+                si1, si2, slen = struct.unpack_from("<BBH", extra, off)
+                if si1 == 66 and si2 == 67 and slen == 2:
+                    bsize = struct.unpack_from("<H", extra, off + 4)[0]
+                off += 4 + slen
+            if bsize is None:
+                raise ValueError("missing BGZF BC extra field")
+            cdata = fh.read(bsize - xlen - 19)
+            _crc, isize = struct.unpack("<LL", fh.read(8))
+            yield cdata, isize
+
+    def decompress(job):
+        cdata, isize = job
+        data = zlib.decompress(cdata, -15)
+        if len(data) != isize:
+            raise ValueError("BGZF ISIZE mismatch")
+        return data
+
+    with _futures.ThreadPoolExecutor(threads) as pool:
+        pending: "_collections.deque" = _collections.deque()
+        src = read_compressed()
+        for job in src:
+            pending.append(pool.submit(decompress, job))
+            if len(pending) >= window:
+                data = pending.popleft().result()
+                if data:
+                    yield data
+        while pending:
+            data = pending.popleft().result()
+            if data:
+                yield data
+
+
 def _threaded_blocks(gen: Iterator[bytes], depth: int = 64
                      ) -> Iterator[bytes]:
     """Readahead thread: BGZF read+decompress runs ahead of the record
@@ -427,6 +488,22 @@ def _threaded_blocks(gen: Iterator[bytes], depth: int = 64
         yield item
 
 
+def _parse_bam_header(stream: "_ConcatStream", path: str) -> "BamHeader":
+    magic = stream.read(4)
+    if magic != BAM_MAGIC:
+        raise ValueError(f"{path} is not a BAM file")
+    (l_text,) = struct.unpack("<l", stream.read(4))
+    text = stream.read(l_text).decode(errors="replace").rstrip("\x00")
+    (n_ref,) = struct.unpack("<l", stream.read(4))
+    refs = []
+    for _ in range(n_ref):
+        (l_name,) = struct.unpack("<l", stream.read(4))
+        name = stream.read(l_name)[:-1].decode()
+        (l_ref,) = struct.unpack("<l", stream.read(4))
+        refs.append((name, l_ref))
+    return BamHeader(text, refs)
+
+
 class BamReader:
     """Sequential BAM reader (with a decompression readahead thread)."""
 
@@ -437,19 +514,7 @@ class BamReader:
         if reader_threads and reader_threads > 0:
             blocks = _threaded_blocks(blocks)
         self._stream = _ConcatStream(blocks)
-        magic = self._stream.read(4)
-        if magic != BAM_MAGIC:
-            raise ValueError(f"{path} is not a BAM file")
-        (l_text,) = struct.unpack("<l", self._stream.read(4))
-        text = self._stream.read(l_text).decode(errors="replace").rstrip("\x00")
-        (n_ref,) = struct.unpack("<l", self._stream.read(4))
-        refs = []
-        for _ in range(n_ref):
-            (l_name,) = struct.unpack("<l", self._stream.read(4))
-            name = self._stream.read(l_name)[:-1].decode()
-            (l_ref,) = struct.unpack("<l", self._stream.read(4))
-            refs.append((name, l_ref))
-        self.header = BamHeader(text, refs)
+        self.header = _parse_bam_header(self._stream, path)
 
     def __iter__(self) -> Iterator[BamRead]:
         return self
@@ -463,37 +528,154 @@ class BamReader:
         return self._decode(buf)
 
     def _decode(self, buf: bytes) -> BamRead:
-        (ref_id, pos, l_read_name, mapq, _bin, n_cigar, flag, l_seq,
-         next_ref, next_pos, tlen) = struct.unpack_from("<llBBHHHllll", buf, 0)
-        off = 32
-        qname = buf[off:off + l_read_name - 1].decode()
-        off += l_read_name
-        # Vectorized cigar + 4-bit seq decode (the per-base python loops
-        # were the hottest lines of the whole serial feeder path).
-        cig = np.frombuffer(buf, "<u4", n_cigar, off)
-        cigartuples = list(zip((cig & 0xF).tolist(), (cig >> 4).tolist()))
-        off += 4 * n_cigar
-        nbytes = (l_seq + 1) // 2
-        packed = np.frombuffer(buf, np.uint8, nbytes, off)
-        chars = np.empty(2 * nbytes, np.uint8)
-        chars[0::2] = _NT16_U8[packed >> 4]
-        chars[1::2] = _NT16_U8[packed & 0xF]
-        seq = chars[:l_seq].tobytes().decode("ascii")
-        off += nbytes
-        quals = np.frombuffer(buf, np.uint8, l_seq, off).astype(np.int16)
-        if l_seq and quals.size and quals[0] == 0xFF:
-            qual_arr = None
+        return decode_record(buf, self.header)
+
+    def close(self):
+        self._fh.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+
+def decode_record(buf: bytes, header: Optional["BamHeader"]) -> BamRead:
+    """Decodes one raw BAM record buffer (everything after block_size)."""
+    (ref_id, pos, l_read_name, mapq, _bin, n_cigar, flag, l_seq,
+     next_ref, next_pos, tlen) = struct.unpack_from("<llBBHHHllll", buf, 0)
+    off = 32
+    qname = buf[off:off + l_read_name - 1].decode()
+    off += l_read_name
+    # Vectorized cigar + 4-bit seq decode (the per-base python loops
+    # were the hottest lines of the whole serial feeder path).
+    cig = np.frombuffer(buf, "<u4", n_cigar, off)
+    cigartuples = list(zip((cig & 0xF).tolist(), (cig >> 4).tolist()))
+    off += 4 * n_cigar
+    nbytes = (l_seq + 1) // 2
+    packed = np.frombuffer(buf, np.uint8, nbytes, off)
+    chars = np.empty(2 * nbytes, np.uint8)
+    chars[0::2] = _NT16_U8[packed >> 4]
+    chars[1::2] = _NT16_U8[packed & 0xF]
+    seq = chars[:l_seq].tobytes().decode("ascii")
+    off += nbytes
+    quals = np.frombuffer(buf, np.uint8, l_seq, off).astype(np.int16)
+    if l_seq and quals.size and quals[0] == 0xFF:
+        qual_arr = None
+    else:
+        qual_arr = quals
+    off += l_seq
+    tags = _parse_tags(buf[off:])
+    read = BamRead(
+        qname=qname, flag=flag, ref_id=ref_id, pos=pos, mapq=mapq,
+        cigartuples=cigartuples, seq=seq, query_qualities=qual_arr,
+        tags=tags, header=header,
+    )
+    read.next_ref_id, read.next_pos, read.tlen = next_ref, next_pos, tlen
+    return read
+
+
+# ---------------------------------------------------------------------------
+# Raw-record peeks (no full decode)
+# ---------------------------------------------------------------------------
+# Fixed-offset layout of a raw record buffer: ref_id@0, pos@4,
+# l_read_name@8, mapq@9, bin@10, n_cigar@12, flag@14, l_seq@16,
+# next_ref@20, next_pos@24, tlen@28, qname@32.
+
+
+def raw_flag(buf: bytes) -> int:
+    return struct.unpack_from("<H", buf, 14)[0]
+
+
+def raw_ref_id(buf: bytes) -> int:
+    return struct.unpack_from("<l", buf, 0)[0]
+
+
+def raw_qname(buf: bytes) -> str:
+    l_read_name = buf[8]
+    return buf[32:32 + l_read_name - 1].decode()
+
+
+def raw_tag(buf: bytes, name: str, default: Any = None) -> Any:
+    """Minimal tag scan: walks tag headers, decoding only `name`.
+
+    ~10x cheaper than a full decode_record when only the zm group key
+    (or the wl smart-window widths) is needed on the serial feeder path.
+    """
+    l_read_name = buf[8]
+    n_cigar = struct.unpack_from("<H", buf, 12)[0]
+    l_seq = struct.unpack_from("<l", buf, 16)[0]
+    off = 32 + l_read_name + 4 * n_cigar + (l_seq + 1) // 2 + l_seq
+    want = name.encode()
+    n = len(buf)
+    while off + 3 <= n:
+        key = buf[off:off + 2]
+        typ = chr(buf[off + 2])
+        off += 3
+        hit = key == want
+        if typ == "A":
+            if hit:
+                return chr(buf[off])
+            off += 1
+        elif typ in _TAG_FMT:
+            fmt = _TAG_FMT[typ]
+            size = struct.calcsize(fmt)
+            if hit:
+                return struct.unpack_from("<" + fmt, buf, off)[0]
+            off += size
+        elif typ in ("Z", "H"):
+            end = buf.index(0, off)
+            if hit:
+                return buf[off:end].decode()
+            off = end + 1
+        elif typ == "B":
+            sub = chr(buf[off])
+            count = struct.unpack_from("<I", buf, off + 1)[0]
+            dt = _ARRAY_DTYPE[sub]
+            size = count * np.dtype(dt).itemsize
+            if hit:
+                return np.frombuffer(
+                    buf, dtype=dt, count=count, offset=off + 5
+                ).copy()
+            off += 5 + size
+            continue
         else:
-            qual_arr = quals
-        off += l_seq
-        tags = _parse_tags(buf[off:])
-        read = BamRead(
-            qname=qname, flag=flag, ref_id=ref_id, pos=pos, mapq=mapq,
-            cigartuples=cigartuples, seq=seq, query_qualities=qual_arr,
-            tags=tags, header=self.header,
-        )
-        read.next_ref_id, read.next_pos, read.tlen = next_ref, next_pos, tlen
-        return read
+            raise ValueError(f"unknown tag type {typ!r}")
+    return default
+
+
+class RawBamReader:
+    """Sequential reader yielding RAW record buffers (no record decode).
+
+    The serial feeder only needs the zm group key, the flag and the
+    reference id of each subread record; full decode_record runs in the
+    worker pool instead (feeder.RawZmwJob). BGZF decompression runs on
+    a thread pool (decompress_threads) since zlib releases the GIL.
+    """
+
+    def __init__(self, path: str, decompress_threads: int = 8):
+        self.path = path
+        # 4 MB read buffering: a BGZF block is ~64 KB, the default 8 KB
+        # buffer turns every block into several syscalls.
+        self._fh = open(path, "rb", buffering=4 << 20)
+        if decompress_threads > 1:
+            blocks = _read_bgzf_blocks_parallel(
+                self._fh, threads=decompress_threads
+            )
+        else:
+            blocks = _threaded_blocks(_read_bgzf_blocks(self._fh))
+        self._stream = _ConcatStream(blocks)
+        self.header = _parse_bam_header(self._stream, path)
+
+    def __iter__(self) -> Iterator[bytes]:
+        return self
+
+    def __next__(self) -> bytes:
+        head = self._stream.read(4)
+        if len(head) < 4:
+            raise StopIteration
+        (block_size,) = struct.unpack("<l", head)
+        return self._stream.read(block_size)
 
     def close(self):
         self._fh.close()

@@ -98,6 +98,68 @@ def test_bam_round_trip(tmp_path):
     assert len(idx["m000/10/ccs"]) == 4
 
 
+def test_raw_bam_reader_matches_decoded(tmp_path):
+    """RawBamReader buffers + decode_record reproduce BamReader exactly,
+    and the raw_* peeks agree with the decoded fields."""
+    sub, ccs = make_test_bams(tmp_path)
+    decoded = list(bam_lib.BamReader(sub))
+    raw_reader = bam_lib.RawBamReader(sub)
+    raws = list(raw_reader)
+    assert len(raws) == len(decoded)
+    for buf, ref in zip(raws, decoded):
+        assert bam_lib.raw_qname(buf) == ref.qname
+        assert bam_lib.raw_flag(buf) == ref.flag
+        assert bam_lib.raw_ref_id(buf) == ref.ref_id
+        assert bam_lib.raw_tag(buf, "zm") == ref.get_tag("zm")
+        np.testing.assert_array_equal(
+            bam_lib.raw_tag(buf, "pw"), ref.get_tag("pw")
+        )
+        assert bam_lib.raw_tag(buf, "absent", default=-1) == -1
+        got = bam_lib.decode_record(buf, raw_reader.header)
+        assert got.qname == ref.qname
+        assert got.cigartuples == ref.cigartuples
+        assert got.seq == ref.seq
+        np.testing.assert_array_equal(
+            got.query_qualities, ref.query_qualities
+        )
+        assert got.reference_name == ref.reference_name
+        assert set(got.tags) == set(ref.tags)
+
+
+def test_raw_feeder_matches_decoded_feeder(tmp_path):
+    """create_proc_feeder(raw_records=True) materializes to the same
+    expanded Read stacks as the decoded deferred feeder."""
+    from deepconsensus_amd.preprocess import feeder as pre_feeder
+    from deepconsensus_amd.preprocess.windows import DcConfig
+
+    sub, ccs = make_test_bams(tmp_path)
+    dc_config = DcConfig(20, 100, False)
+    jobs = {}
+    for raw in (False, True):
+        pf, counter = pre_feeder.create_proc_feeder(
+            subreads_to_ccs=sub, ccs_bam=ccs, dc_config=dc_config,
+            defer_expansion=True, raw_records=raw,
+        )
+        jobs[raw] = [(z, job) for job, z, *_ in pf()]
+        assert counter["n_zmw_processed"] == 3
+    assert [z for z, _ in jobs[False]] == [z for z, _ in jobs[True]]
+    for (_, a), (_, b) in zip(jobs[False], jobs[True]):
+        assert isinstance(b, pre_feeder.RawZmwJob)
+        assert len(a) == len(b)
+        import collections
+
+        ca, cb = collections.Counter(), collections.Counter()
+        reads_a, reads_b = a.materialize(ca), b.materialize(cb)
+        assert ca == cb
+        assert len(reads_a) == len(reads_b)
+        for ra, rb in zip(reads_a, reads_b):
+            assert ra.name == rb.name
+            np.testing.assert_array_equal(ra.bases, rb.bases)
+            np.testing.assert_array_equal(ra.pw, rb.pw)
+            np.testing.assert_array_equal(ra.ip, rb.ip)
+            np.testing.assert_array_equal(ra.ccs_idx, rb.ccs_idx)
+
+
 def test_example_codec_round_trip():
     feats = {
         "subreads/encoded": (example_codec.BYTES, [b"\x00\x01\x02\x03"]),
