@@ -95,14 +95,46 @@ class InferenceOptions:
     end_after_stage: DebugStage = DebugStage.FULL
 
 
-def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
+@dataclasses.dataclass
+class ZmwWindows:
+    """Worker output for one ZMW: model windows pre-stacked, skipped
+    windows pre-resolved to DCModelOutput.
+
+    The reference crossed the pool boundary with one feature dict per
+    window and ran skip triage + CCS passthrough serially
+    (quick_inference.py:655-677). Here the worker does the triage
+    (avg-Phred + overflow), builds the CCS passthrough outputs, and
+    stacks the model-bound windows into ONE int16 [n, R, L] array — the
+    serial batch loop only concatenates arrays and runs the model.
+    """
+
+    name: str
+    window_pos: np.ndarray  # [n_model] int64, model-bound windows only
+    rows: Optional[np.ndarray]  # [n_model, R, L] int16, format_rows'd
+    skipped: List["stitch_utils.DCModelOutput"]
+    counter: collections.Counter
+    ec: Optional[float] = None
+    np_num_passes: Optional[int] = None
+    rq: Optional[float] = None
+    rg: Optional[str] = None
+
+    @property
+    def n_examples(self) -> int:
+        n = len(self.skipped)
+        if self.rows is not None:
+            n += len(self.rows)
+        return n
+
+
+def preprocess_one_zmw(one_zmw) -> ZmwWindows:
     """Windows + counters for one ZMW (quick_inference.py:535-564).
 
-    ``subreads`` may be a deferred ZmwJob (raw BAM records): expansion then
-    runs HERE — i.e. in the worker pool — instead of the serial feeder.
+    ``subreads`` may be a deferred ZmwJob/RawZmwJob (raw BAM records):
+    record decode and expansion then run HERE — i.e. in the worker
+    pool — instead of the serial feeder.
     """
-    zmw, subreads, dc_config, window_widths, *rest = one_zmw
-    stage = rest[0] if rest else DebugStage.FULL
+    zmw, subreads, dc_config, window_widths, options = one_zmw
+    stage = options.end_after_stage
     expand_counter = None
     if isinstance(subreads, pre_feeder.ZmwJob):
         expand_counter = collections.Counter()
@@ -113,14 +145,14 @@ def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
         dc_config=dc_config,
         window_widths=window_widths,
     )
-    if expand_counter:
-        dc_whole.counter.update(expand_counter)
+    empty = ZmwWindows(name=zmw, window_pos=np.empty(0, np.int64),
+                       rows=None, skipped=[], counter=dc_whole.counter)
     if stage == DebugStage.DC_INPUT:
-        return [], dc_whole.counter
-    feature_dicts = [x.to_features_dict() for x in dc_whole.iter_examples()]
+        if expand_counter:
+            empty.counter.update(expand_counter)
+        return empty
     # Worker-side finishing, keeping the serial batch loop thin:
-    #  * format_rows (PW/IP/SN clipping) runs HERE, in parallel — the
-    #    main thread then only stacks ("fmt" marks it done);
+    #  * format_rows (PW/IP/SN clipping) runs HERE, in parallel;
     #  * features are integral-after-truncation for the model (the
     #    embedding casts to int; SN fractions truncate identically) and
     #    <= SN_MAX after clipping, so they cross the process boundary as
@@ -131,11 +163,33 @@ def preprocess_one_zmw(one_zmw) -> Tuple[List[Dict[str, Any]], Any]:
         total_rows=dc_config.tensor_height,
         PW_MAX=255, IP_MAX=255, SN_MAX=500,  # config.py:89-92 defaults
     )
-    for f in feature_dicts:
+    model_rows: List[np.ndarray] = []
+    window_pos: List[int] = []
+    skipped: List[stitch_utils.DCModelOutput] = []
+    for x in dc_whole.iter_examples():
+        f = x.to_features_dict()
+        skip = bool(f["overflow"])
+        if not skip and options.skip_windows_above:
+            avg_q = phred.avg_phred(f["ccs_base_quality_scores"])
+            skip = avg_q > options.skip_windows_above
+        if skip:
+            skipped.append(process_skipped_window(f, options))
+            continue
         rows = data_lib.format_rows(np.asarray(f["subreads"]), fmt_params)
-        f["subreads"] = rows.astype(np.int16)
-        f["fmt"] = True
-    return feature_dicts, dc_whole.counter
+        model_rows.append(rows[:, :, 0].astype(np.int16))
+        window_pos.append(int(f["window_pos"]))
+    counter = dc_whole.counter
+    if expand_counter:
+        counter.update(expand_counter)
+    ccs = dc_whole.ccs
+    return ZmwWindows(
+        name=zmw,
+        window_pos=np.asarray(window_pos, np.int64),
+        rows=np.stack(model_rows) if model_rows else None,
+        skipped=skipped,
+        counter=counter,
+        ec=ccs.ec, np_num_passes=ccs.np_num_passes, rq=ccs.rq, rg=ccs.rg,
+    )
 
 
 def process_skipped_window(
@@ -172,34 +226,47 @@ def process_skipped_window(
     )
 
 
-def run_model_on_examples(
-    feature_dicts: List[Dict[str, Any]],
+_SEQ_LUT = np.frombuffer(constants.SEQ_VOCAB.encode("ascii"), np.uint8)
+
+
+def run_model_on_zmws(
+    zmws: List[ZmwWindows],
     runner: InferenceRunner,
     options: InferenceOptions,
 ) -> List[stitch_utils.DCModelOutput]:
-    """Batched model execution emitting per-window DCModelOutput."""
+    """Batched model execution emitting per-window DCModelOutput.
+
+    Model-bound windows arrive pre-stacked per ZMW; this concatenates
+    them across the ZMW batch, runs the device model in batch_size
+    chunks, and converts bases/QVs to strings with batch-level numpy
+    (one LUT pass + one +33 pass per chunk).
+    """
     predictions: List[stitch_utils.DCModelOutput] = []
-    params = runner.params
+    # Per-window owner metadata, in concatenation order.
+    owners: List[ZmwWindows] = []
+    pos_list: List[np.ndarray] = []
+    rows_list: List[np.ndarray] = []
+    for z in zmws:
+        predictions.extend(z.skipped)
+        if z.rows is not None and len(z.rows):
+            rows_list.append(z.rows)
+            pos_list.append(z.window_pos)
+            owners.extend([z] * len(z.rows))
+    if not rows_list:
+        return predictions
+    all_rows = np.concatenate(rows_list)
+    all_pos = np.concatenate(pos_list)
     # Native path: features are non-negative and <= SN_MAX (500) after
     # clipping, and the embed kernel casts to int anyway, so int16 staging
     # halves H2D traffic with bit-identical results (numpy's truncation
     # toward zero == the kernel's (int)v == the torch model's .long()).
     use_i16 = bool(getattr(runner, "native", False))
-    np_dtype = np.int16 if use_i16 else np.float32
+    if not use_i16:
+        all_rows = all_rows.astype(np.float32)
     pinned: Dict[Tuple[int, ...], torch.Tensor] = {}
-    for i in range(0, len(feature_dicts), options.batch_size):
-        chunk = feature_dicts[i : i + options.batch_size]
-        rows = np.stack(
-            [
-                np.asarray(f["subreads"])[:, :, 0]
-                if f.get("fmt")
-                else data_lib.format_rows(
-                    np.asarray(f["subreads"]), params
-                )[:, :, 0]
-                for f in chunk
-            ]
-        ).astype(np_dtype)
-        rows_t = torch.from_numpy(rows)
+    for i in range(0, len(all_rows), options.batch_size):
+        chunk = np.ascontiguousarray(all_rows[i : i + options.batch_size])
+        rows_t = torch.from_numpy(chunk)
         if use_i16:
             buf = pinned.get(rows_t.shape)
             if buf is None:
@@ -208,21 +275,20 @@ def run_model_on_examples(
             buf.copy_(rows_t)
             rows_t = buf
         bases_t, quals_t = runner.forward_windows(rows_t)
-        bases = bases_t.cpu().numpy()
-        quals = quals_t.cpu().numpy()
-        for j, f in enumerate(chunk):
-            seq = phred.encoded_sequence_to_string(bases[j])
-            qual = phred.quality_scores_to_string(quals[j])
+        seq_mat = _SEQ_LUT[bases_t.cpu().numpy().astype(np.int64)]
+        qual_mat = (quals_t.cpu().numpy() + 33).astype(np.uint8)
+        for j in range(len(chunk)):
+            z = owners[i + j]
             predictions.append(
                 stitch_utils.DCModelOutput(
-                    window_pos=f["window_pos"],
-                    molecule_name=f["name"],
-                    sequence=seq,
-                    quality_string=qual,
-                    ec=f["ec"],
-                    np_num_passes=f["np_num_passes"],
-                    rq=f["rq"],
-                    rg=f["rg"],
+                    window_pos=int(all_pos[i + j]),
+                    molecule_name=z.name,
+                    sequence=seq_mat[j].tobytes().decode("ascii"),
+                    quality_string=qual_mat[j].tobytes().decode("ascii"),
+                    ec=z.ec,
+                    np_num_passes=z.np_num_passes,
+                    rq=z.rq,
+                    rg=z.rg,
                 )
             )
     return predictions
@@ -404,8 +470,7 @@ def run(
             idx += 1
             if not keep:
                 continue
-            batch.append((zmw, subreads, dcc, window_widths,
-                          options.end_after_stage))
+            batch.append((zmw, subreads, dcc, window_widths, options))
             if len(batch) >= options.batch_zmws:
                 yield batch
                 batch = []
@@ -419,12 +484,9 @@ def run(
 
     def infer_batch(inputs, outputs, batch_name):
         before = time.time()
-        feature_dicts_for_zmws, counters = (
-            zip(*outputs) if outputs else ([], [])
-        )
-        for counter in counters:
-            stats_counter.update(counter)
-        n_examples = sum(len(z) for z in feature_dicts_for_zmws)
+        for z in outputs:
+            stats_counter.update(z.counter)
+        n_examples = sum(z.n_examples for z in outputs)
         n_subreads = sum(len(z[1]) for z in inputs)
         timelog.add("preprocess", batch_name, before, n_examples,
                     n_subreads, len(inputs))
@@ -433,34 +495,18 @@ def run(
             return
 
         before = time.time()
-        for_model: List[Dict[str, Any]] = []
-        skipped: List[stitch_utils.DCModelOutput] = []
-        for one_zmw in feature_dicts_for_zmws:
-            for window in one_zmw:
-                skip = False
-                if window["overflow"]:
-                    skipped.append(process_skipped_window(window, options))
-                    skip = True
-                if options.skip_windows_above and not skip:
-                    avg_q = phred.avg_phred(
-                        window["ccs_base_quality_scores"]
-                    )
-                    if avg_q > options.skip_windows_above:
-                        skipped.append(
-                            process_skipped_window(window, options)
-                        )
-                        skip = True
-                if not skip:
-                    for_model.append(window)
-        preds = run_model_on_examples(for_model, runner, options)
-        preds.extend(skipped)
+        n_model = sum(
+            len(z.rows) for z in outputs if z.rows is not None
+        )
+        n_skipped = sum(len(z.skipped) for z in outputs)
+        preds = run_model_on_zmws(outputs, runner, options)
         # Per-batch model/skip split (quick_inference.py:688-705).
         total = max(len(preds), 1)
         log.info(
             "Example summary: ran model=%d (%.2f%%) skip=%d (%.2f%%) "
             "total=%d.",
-            len(for_model), 100.0 * len(for_model) / total,
-            len(skipped), 100.0 * len(skipped) / total, len(preds),
+            n_model, 100.0 * n_model / total,
+            n_skipped, 100.0 * n_skipped / total, len(preds),
         )
         timelog.add("run_model", batch_name, before, n_examples,
                     n_subreads, len(inputs))
