@@ -420,11 +420,6 @@ def run(
         )
 
     calib_str = dc_calibration if dc_calibration else "skip"
-    runner = InferenceRunner(
-        params, model, device=device, calibration=calib_str,
-        max_qual=options.max_base_quality,
-    )
-    log.info("model on %s (native kernels: %s)", runner.device, runner.native)
 
     # Output writer.
     bam_out = None
@@ -524,6 +519,20 @@ def run(
     pending: Optional[Tuple[Sequence, Any]] = None
     n_batches = 0
     with concurrent.futures.ThreadPoolExecutor(1) as prefetcher:
+        # Prime the pipeline BEFORE device/model setup: the worker pool
+        # preprocesses batch 1 while InferenceRunner loads weights onto
+        # the device and the HIP extension initializes — for short runs
+        # this hides most of the pool spin-up + first-batch latency.
+        first_inputs = next(batch_iter, None)
+        if first_inputs is not None:
+            pending = (first_inputs,
+                       prefetcher.submit(preprocess_batch, first_inputs))
+        runner = InferenceRunner(
+            params, model, device=device, calibration=calib_str,
+            max_qual=options.max_base_quality,
+        )
+        log.info("model on %s (native kernels: %s)", runner.device,
+                 runner.native)
 
         def drain(prev_inputs, prev_fut, name):
             # Time spent blocked on the overlapped preprocessing (not a
