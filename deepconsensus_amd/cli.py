@@ -15,7 +15,27 @@ from typing import List, Optional
 import deepconsensus_amd
 
 COMMANDS = ["run", "preprocess", "calibrate", "filter_reads", "train",
-            "distill", "eval", "export"]
+            "distill", "eval", "export", "index"]
+
+
+def _index_main(argv: List[str]) -> None:
+    from deepconsensus_amd.dcio import bam as bam_lib
+
+    ap = argparse.ArgumentParser(
+        "deepconsensus index",
+        description="Build ZMW byte-range index sidecars (<bam>"
+        f"{bam_lib.ZMW_INDEX_SUFFIX}) so sharded `run --shard i/N` "
+        "jobs seek directly to their ZMW range instead of streaming "
+        "the whole BAM per shard.",
+    )
+    ap.add_argument("bams", nargs="+", help="ZMW-sorted BAM file(s)")
+    args = ap.parse_args(argv)
+    for path in args.bams:
+        out = bam_lib.build_zmw_index(path)
+        idx = bam_lib.load_zmw_index(path)
+        print(f"{out}: {len(idx['zmw'])} ZMW groups"
+              + ("" if idx["sorted_flag"][0] else " (NOT zm-sorted; "
+                 "sharded runs will fall back to full streaming)"))
 
 
 def _run_main(argv: List[str]) -> None:
@@ -44,8 +64,11 @@ def _run_main(argv: List[str]) -> None:
     ap.add_argument("--ccs_calibration", default=None)
     ap.add_argument("--device", default=None)
     ap.add_argument("--shard", default=None,
-                    help="'i/N': process only ZMWs with index %% N == i "
-                    "(one process per GPU, like the reference's ccs --chunk)")
+                    help="'i/N': process only shard i of N (one process "
+                    "per GPU). With `deepconsensus index` sidecars the "
+                    "shard seeks straight to its contiguous ZMW byte "
+                    "range; otherwise it streams the whole BAM keeping "
+                    "every Nth ZMW (like the reference's ccs --chunk)")
     ap.add_argument("--end_after_stage", default="full",
                     choices=[s.name.lower() for s in qi.DebugStage],
                     help="stop after this stage (debug/runtime testing)")
@@ -113,6 +136,7 @@ def main(argv: Optional[List[str]] = None) -> None:
             "  calibrate     empirical base-quality calibration stats\n"
             "  filter_reads  filter FASTQ/BAM by average read quality\n"
             "  train         train a model (see also: distill, eval)\n"
+            "  index         build ZMW byte-range sidecars for sharded runs\n"
         )
         return
     command, rest = argv[0], argv[1:]
@@ -146,6 +170,8 @@ def main(argv: Optional[List[str]] = None) -> None:
         from deepconsensus_amd.models import export_model
 
         export_model.main(rest)
+    elif command == "index":
+        _index_main(rest)
     else:
         print(f"unknown command {command!r}; one of: {', '.join(COMMANDS)}")
         sys.exit(2)

@@ -20,6 +20,7 @@ name" for truth-to-CCS lookups is provided by a sequential scan into a dict
 """
 from __future__ import annotations
 
+import os
 import struct
 import zlib
 from typing import Any, Dict, Iterator, List, Optional, Sequence, Tuple
@@ -651,13 +652,24 @@ class RawBamReader:
     reference id of each subread record; full decode_record runs in the
     worker pool instead (feeder.RawZmwJob). BGZF decompression runs on
     a thread pool (decompress_threads) since zlib releases the GIL.
+
+    ``start=(compressed_offset, within_block_offset)`` resumes reading
+    mid-file at a record boundary taken from a ZMW index
+    (build_zmw_index) — this is how N-shard runs each decompress only
+    1/N of the stream instead of all of it.
     """
 
-    def __init__(self, path: str, decompress_threads: int = 8):
+    def __init__(self, path: str, decompress_threads: int = 8,
+                 start: Optional[Tuple[int, int]] = None):
         self.path = path
         # 4 MB read buffering: a BGZF block is ~64 KB, the default 8 KB
         # buffer turns every block into several syscalls.
         self._fh = open(path, "rb", buffering=4 << 20)
+        if start is not None:
+            # Header lives at file start; fetch it with a tiny separate
+            # read, then jump straight to the shard's first block.
+            self.header = read_bam_header(path)
+            self._fh.seek(start[0])
         if decompress_threads > 1:
             blocks = _read_bgzf_blocks_parallel(
                 self._fh, threads=decompress_threads
@@ -665,7 +677,11 @@ class RawBamReader:
         else:
             blocks = _threaded_blocks(_read_bgzf_blocks(self._fh))
         self._stream = _ConcatStream(blocks)
-        self.header = _parse_bam_header(self._stream, path)
+        if start is not None:
+            if start[1]:
+                self._stream.read(start[1])
+        else:
+            self.header = _parse_bam_header(self._stream, path)
 
     def __iter__(self) -> Iterator[bytes]:
         return self
@@ -685,6 +701,185 @@ class RawBamReader:
 
     def __exit__(self, *a):
         self.close()
+
+
+def read_bam_header(path: str) -> BamHeader:
+    """Reads just the BAM header (decompressing only the leading blocks)."""
+    with open(path, "rb") as fh:
+        stream = _ConcatStream(_read_bgzf_blocks(fh))
+        return _parse_bam_header(stream, path)
+
+
+class _OffsetStream:
+    """Byte stream over (compressed_offset, payload) blocks that can
+    report the BGZF virtual offset of its current read position."""
+
+    def __init__(self, blocks: Iterator[Tuple[int, bytes]]):
+        self._blocks = blocks
+        self._buf = bytearray()
+        self._g = 0  # global (decompressed) position of _buf[0]
+        self._end_g = 0  # global position just past the appended data
+        import collections as _c
+
+        self._spans: "_c.deque" = _c.deque()  # (start_g, coffset, length)
+
+    def _pull(self) -> bool:
+        try:
+            coffset, payload = next(self._blocks)
+        except StopIteration:
+            return False
+        self._spans.append((self._end_g, coffset, len(payload)))
+        self._end_g += len(payload)
+        self._buf += payload
+        return True
+
+    def read(self, n: int) -> bytes:
+        while len(self._buf) < n:
+            if not self._pull():
+                break
+        out = bytes(self._buf[:n])
+        del self._buf[:n]
+        self._g += len(out)
+        return out
+
+    def voffset(self) -> Optional[Tuple[int, int]]:
+        """(compressed_offset, within_block_offset) of the current
+        position, or None at EOF."""
+        while self._spans and (
+            self._spans[0][0] + self._spans[0][2] <= self._g
+        ):
+            self._spans.popleft()
+        if not self._spans:
+            if not self._pull():
+                return None
+        start_g, coffset, _length = self._spans[0]
+        return coffset, self._g - start_g
+
+
+ZMW_INDEX_SUFFIX = ".zmi.npz"
+
+
+def build_zmw_index(path: str, out_path: Optional[str] = None,
+                    decompress_threads: int = 8) -> str:
+    """One fast pass over a ZMW-sorted BAM writing a group index sidecar.
+
+    For every run of consecutive records sharing a zm tag, stores
+    (zmw, compressed_offset, within_block_offset) of the first record.
+    Shard i/N then seeks straight to its contiguous ZMW range and
+    decompresses only ~1/N of the stream (ROADMAP: the coarse index
+    pass that breaks the serial per-rank BGZF floor). The pass itself
+    runs at raw-feeder speed: parallel decompress + zm peeks only.
+
+    Written via np.savez: zmw[G] int64, coffset[G] int64,
+    uoffset[G] uint16, sorted_flag (1 if zmw ids are nondecreasing).
+    """
+    out_path = out_path or path + ZMW_INDEX_SUFFIX
+    zmws: List[int] = []
+    coffs: List[int] = []
+    uoffs: List[int] = []
+    is_sorted = True
+    with open(path, "rb", buffering=4 << 20) as fh:
+        stream = _OffsetStream(
+            _read_bgzf_blocks_parallel_offsets(fh, decompress_threads)
+        )
+        _parse_bam_header(stream, path)
+        prev_zm = None
+        while True:
+            vo = stream.voffset()
+            head = stream.read(4)
+            if len(head) < 4:
+                break
+            (block_size,) = struct.unpack("<l", head)
+            rec = stream.read(block_size)
+            zm = raw_tag(rec, "zm")
+            if zm is None:
+                raise ValueError(
+                    f"{path}: record without zm tag; cannot index"
+                )
+            if zm != prev_zm:
+                if prev_zm is not None and zm < prev_zm:
+                    is_sorted = False
+                zmws.append(int(zm))
+                coffs.append(vo[0])
+                uoffs.append(vo[1])
+                prev_zm = zm
+    tmp = out_path + ".tmp"
+    with open(tmp, "wb") as f:
+        np.savez(
+            f,
+            zmw=np.asarray(zmws, np.int64),
+            coffset=np.asarray(coffs, np.int64),
+            uoffset=np.asarray(uoffs, np.uint16),
+            sorted_flag=np.array([1 if is_sorted else 0], np.int8),
+        )
+    os.replace(tmp, out_path)
+    return out_path
+
+
+def load_zmw_index(path: str) -> Optional[Dict[str, np.ndarray]]:
+    """Loads a sidecar written by build_zmw_index; None if absent."""
+    idx_path = path + ZMW_INDEX_SUFFIX
+    if not os.path.exists(idx_path):
+        return None
+    with np.load(idx_path) as z:
+        return {k: z[k] for k in ("zmw", "coffset", "uoffset",
+                                  "sorted_flag")}
+
+
+def _read_bgzf_blocks_parallel_offsets(
+    fh, threads: int = 8, window: int = 64
+) -> Iterator[Tuple[int, bytes]]:
+    """Like _read_bgzf_blocks_parallel but yields
+    (compressed_file_offset, payload) so callers can build indexes."""
+    import collections as _collections
+    import concurrent.futures as _futures
+
+    def read_compressed():
+        while True:
+            coffset = fh.tell()
+            header = fh.read(12)
+            if len(header) == 0:
+                return
+            if len(header) < 12:
+                raise ValueError("truncated BGZF header")
+            magic, _mtime, _xfl, _os, xlen = struct.unpack(
+                "<4sLBBH", header
+            )
+            if magic[:2] != b"\x1f\x8b":
+                raise ValueError("not a gzip stream")
+            extra = fh.read(xlen)
+            bsize = None
+            off = 0
+            while off + 4 <= len(extra):
+                si1, si2, slen = struct.unpack_from("<BBH", extra, off)
+                if si1 == 66 and si2 == 67 and slen == 2:
+                    bsize = struct.unpack_from("<H", extra, off + 4)[0]
+                off += 4 + slen
+            if bsize is None:
+                raise ValueError("missing BGZF BC extra field")
+            cdata = fh.read(bsize - xlen - 19)
+            _crc, isize = struct.unpack("<LL", fh.read(8))
+            yield coffset, cdata, isize
+
+    def decompress(job):
+        coffset, cdata, isize = job
+        data = zlib.decompress(cdata, -15)
+        if len(data) != isize:
+            raise ValueError("BGZF ISIZE mismatch")
+        return coffset, data
+
+    with _futures.ThreadPoolExecutor(threads) as pool:
+        pending: "_collections.deque" = _collections.deque()
+        for job in read_compressed():
+            pending.append(pool.submit(decompress, job))
+            if len(pending) >= window:
+                coffset, data = pending.popleft().result()
+                if data:
+                    yield coffset, data
+        while pending:
+            coffset, data = pending.popleft().result()
+            if data:
+                yield coffset, data
 
 
 class _ConcatStream:

@@ -450,6 +450,8 @@ def run(
         use_ccs_smart_windows=options.use_ccs_smart_windows,
         limit=limit,
         defer_expansion=True,
+        shard_index=options.shard_index,
+        shard_count=options.shard_count,
     )
 
     pool = None
@@ -457,14 +459,11 @@ def run(
         pool = concurrent.futures.ProcessPoolExecutor(options.cpus)
 
     def zmw_batches():
+        # ZMW sharding happens inside the feeder (byte-range seek when a
+        # `deepconsensus index` sidecar exists, modulo streaming else).
         batch = []
-        idx = 0
         for input_data in proc_feeder():
             subreads, zmw, dcc, split, window_widths = input_data
-            keep = idx % options.shard_count == options.shard_index
-            idx += 1
-            if not keep:
-                continue
             batch.append((zmw, subreads, dcc, window_widths, options))
             if len(batch) >= options.batch_zmws:
                 yield batch

@@ -197,16 +197,30 @@ class RawZmwJob(ZmwJob):
 
 
 class RawSubreadGrouper:
-    """SubreadGrouper over raw record buffers (zm-tag peek grouping)."""
+    """SubreadGrouper over raw record buffers (zm-tag peek grouping).
 
-    def __init__(self, subreads_to_ccs: str, decompress_threads: int = 4):
+    ``start`` is a (compressed_offset, within_block_offset) record
+    boundary from bam.build_zmw_index; ``max_groups`` stops after that
+    many ZMW groups — together they let shard i/N stream only its own
+    contiguous byte range.
+    """
+
+    def __init__(self, subreads_to_ccs: str, decompress_threads: int = 4,
+                 start=None, max_groups: Optional[int] = None):
         self.reader = bam.RawBamReader(
-            subreads_to_ccs, decompress_threads=decompress_threads
+            subreads_to_ccs, decompress_threads=decompress_threads,
+            start=start,
         )
         self.header = self.reader.header
         self._iter = iter(self.reader)
         self.keep_iter = True
+        self.max_groups = max_groups
+        self.groups_out = 0
         self.subread_group: List[bytes] = []
+        if max_groups is not None and max_groups <= 0:
+            self.keep_iter = False
+            self.zmw = None
+            return
         first = next(self._iter)
         self.zmw = bam.raw_tag(first, "zm")
         if not bam.raw_flag(first) & bam.FUNMAP:
@@ -214,6 +228,12 @@ class RawSubreadGrouper:
 
     def __iter__(self):
         return self
+
+    def _emit(self, group: List[bytes]) -> List[bytes]:
+        self.groups_out += 1
+        if self.max_groups is not None and self.groups_out >= self.max_groups:
+            self.keep_iter = False
+        return group
 
     def __next__(self) -> List[bytes]:
         if not self.keep_iter:
@@ -234,11 +254,11 @@ class RawSubreadGrouper:
                 self.subread_group = [buf]
                 self.zmw = read_zmw
                 if subreads_set:
-                    return subreads_set
+                    return self._emit(subreads_set)
         if self.subread_group:
             out = self.subread_group
             self.subread_group = []
-            return out
+            return self._emit(out)
         raise StopIteration
 
 
@@ -255,6 +275,8 @@ def create_proc_feeder(
     bam_reader_threads: int = 1,
     defer_expansion: bool = False,
     raw_records: bool = True,
+    shard_index: int = 0,
+    shard_count: int = 1,
 ):
     """Generator feeding per-ZMW jobs (pre_lib.py:1279-1367).
 
@@ -262,6 +284,13 @@ def create_proc_feeder(
     serial thread never decodes a BAM record: BGZF blocks decompress on
     a thread pool, subreads group by a zm-tag peek on raw buffers, and
     full record decode + expansion happen in the worker pool.
+
+    With shard_count > 1, this feeder yields only shard_index's ZMWs.
+    If ZMW index sidecars exist for both BAMs (bam.build_zmw_index /
+    `deepconsensus index`), the shard is a contiguous byte range and the
+    feeder decompresses only ~1/N of each stream — otherwise it streams
+    everything and keeps every Nth group (each rank then pays the full
+    decompress, like the reference's `ccs --chunk` sharding).
     """
     main_counter = collections.Counter()
     is_training = truth_bed and truth_to_ccs and truth_split
@@ -269,9 +298,39 @@ def create_proc_feeder(
         "defer_expansion is an inference-mode optimization"
     )
     raw = defer_expansion and raw_records
+    modulo_shard = shard_count > 1
     if raw:
-        subread_grouper = RawSubreadGrouper(subreads_to_ccs)
-        ccs_bam_h = iter(bam.RawBamReader(ccs_bam))
+        sub_start = None
+        sub_max_groups = None
+        ccs_start = None
+        if shard_count > 1:
+            sub_idx = bam.load_zmw_index(subreads_to_ccs)
+            ccs_idx = bam.load_zmw_index(ccs_bam)
+            if (
+                sub_idx is not None and ccs_idx is not None
+                and sub_idx["sorted_flag"][0] and ccs_idx["sorted_flag"][0]
+            ):
+                n_groups = len(sub_idx["zmw"])
+                lo = n_groups * shard_index // shard_count
+                hi = n_groups * (shard_index + 1) // shard_count
+                sub_max_groups = hi - lo
+                if lo < n_groups:
+                    sub_start = (int(sub_idx["coffset"][lo]),
+                                 int(sub_idx["uoffset"][lo]))
+                    j = int(np.searchsorted(ccs_idx["zmw"],
+                                            sub_idx["zmw"][lo]))
+                    j = min(j, max(len(ccs_idx["zmw"]) - 1, 0))
+                    ccs_start = (int(ccs_idx["coffset"][j]),
+                                 int(ccs_idx["uoffset"][j]))
+                modulo_shard = False
+                logging.info(
+                    "byte-range shard %d/%d: ZMW groups [%d, %d) of %d",
+                    shard_index, shard_count, lo, hi, n_groups,
+                )
+        subread_grouper = RawSubreadGrouper(
+            subreads_to_ccs, start=sub_start, max_groups=sub_max_groups
+        )
+        ccs_bam_h = iter(bam.RawBamReader(ccs_bam, start=ccs_start))
         header = subread_grouper.header
     else:
         subread_grouper = SubreadGrouper(subreads_to_ccs, bam_reader_threads)
@@ -283,8 +342,12 @@ def create_proc_feeder(
         truth_split_dict = read_truth_split(truth_split)
 
     def proc_feeder():
+        group_idx = -1
         for read_set in subread_grouper:
             main_counter["n_zmw_processed"] += 1
+            group_idx += 1
+            if modulo_shard and group_idx % shard_count != shard_index:
+                continue
             if raw:
                 ccs_seqname = header.references[
                     bam.raw_ref_id(read_set[0])
