@@ -42,13 +42,24 @@ def setup_writers(
 
 def process_subreads(subreads, ccs_seqname, dc_config, split,
                      window_widths, queue=None, local=False):
-    """Subread processing worker (preprocess.py:200-223)."""
+    """Subread processing worker (preprocess.py:200-223).
+
+    ``subreads`` may be a deferred ZmwJob/RawZmwJob (inference mode):
+    BAM record decode + expansion then run here, in the worker."""
+    expand_counter = None
+    if isinstance(subreads, pre_feeder.ZmwJob):
+        import collections
+
+        expand_counter = collections.Counter()
+        subreads = subreads.materialize(expand_counter)
     tf_out = []
     dc_example = pre_feeder.subreads_to_dc_example(
         subreads, ccs_seqname, dc_config, window_widths
     )
     for example in dc_example.iter_examples():
         tf_out.append(example.tf_example())
+    if expand_counter:
+        dc_example.counter.update(expand_counter)
     dc_example.counter[f"n_examples_{split}"] += len(tf_out)
     dc_example.counter["n_examples"] += len(tf_out)
     if local:
@@ -136,6 +147,11 @@ def main(argv: Optional[List[str]] = None) -> None:
         truth_split=args.truth_split,
         limit=args.limit,
         bam_reader_threads=args.bam_reader_threads,
+        # Inference mode defers record decode + expansion to the workers
+        # (raw feeder). Training mode keeps feeder-side expansion: label
+        # fetch/filtering needs the expanded reads and the reference
+        # counts expansion stats for ZMWs it later drops.
+        defer_expansion=not is_training,
     )
 
     if args.cpus == 0:
