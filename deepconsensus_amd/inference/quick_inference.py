@@ -166,8 +166,7 @@ def preprocess_one_zmw(one_zmw) -> ZmwWindows:
     model_rows: List[np.ndarray] = []
     window_pos: List[int] = []
     skipped: List[stitch_utils.DCModelOutput] = []
-    for x in dc_whole.iter_examples():
-        f = x.to_features_dict()
+    for f in dc_whole.iter_feature_dicts():
         skip = bool(f["overflow"])
         if not skip and options.skip_windows_above:
             avg_q = phred.avg_phred(f["ccs_base_quality_scores"])

@@ -232,6 +232,105 @@ class DcExample:
                 self.name, reads, self.config, _overflow=self._overflow
             )
 
+    def iter_feature_dicts(self):
+        """Vectorized inference twin of
+        ``(x.to_features_dict() for x in iter_examples())``.
+
+        The per-window path re-encodes bases and re-stacks features for
+        every (window x read) pair — the worker-pool hot spot. This
+        encodes the whole spaced ZMW once into [n_reads, W] planes and
+        emits each window as column slices of them. Training ZMWs (label
+        handling) and irregular read widths fall back to the per-window
+        path; counters and dict contents are identical either way
+        (tests/test_preprocess.py::test_iter_feature_dicts_matches_slow).
+        """
+        if self.is_training or any(
+            len(r.bases) != self.width for r in self.reads
+        ):
+            for x in self.iter_examples():
+                yield x.to_features_dict()
+            return
+        config = self.config
+        max_length = config.max_length
+        dtype = constants.NP_DATA_TYPE
+        subs = self.subreads[: config.max_passes]
+        n = len(subs)
+        ccs = self.ccs
+        W = self.width
+        if n:
+            sub_bases = np.stack([r.bases for r in subs])
+            sub_enc = np.zeros(sub_bases.shape, dtype)
+            for k, base in enumerate(constants.SEQ_VOCAB):
+                if k:
+                    sub_enc[sub_bases == base] = k
+            pw_all = np.stack([r.pw for r in subs]).astype(dtype)
+            ip_all = np.stack([r.ip for r in subs]).astype(dtype)
+            strand_col = np.array(
+                [int(r.strand) for r in subs], dtype
+            )[:, None]
+            sn_col = np.asarray(subs[0].sn, dtype)[:, None]
+        ccs_enc = np.zeros(W, dtype)
+        for k, base in enumerate(constants.SEQ_VOCAB):
+            if k:
+                ccs_enc[ccs.bases == base] = k
+        ccs_bq_all = np.asarray(ccs.base_quality_scores)
+        ccs_idx_all = ccs.ccs_idx
+        n_rows = config.tensor_height
+        scalars = dict(
+            name=self.name,
+            ec=ccs.ec, np_num_passes=ccs.np_num_passes,
+            rq=ccs.rq, rg=ccs.rg,
+        )
+        keep = self.keep_subreads
+
+        self.counter = collections.Counter()
+        start_pos = 0
+        for window_width in self.calculate_windows(max_length):
+            self.counter[f"example_width_bucket_{window_width}"] += 1
+            if start_pos > self.ccs_width:
+                break
+            s, e = start_pos, min(start_pos + window_width, W)
+            start_pos += window_width
+            w_idx = ccs_idx_all[s:e]
+            valid = w_idx >= 0
+            if not valid.any():
+                self.counter["n_examples_no_ccs_idx"] += 1
+                continue
+            overflow = window_width > max_length
+            if overflow:
+                self.counter["n_examples_overflow"] += 1
+                # Read.pad still applies: an overflow window clipped at
+                # the ZMW end shorter than max_length pads back up to it.
+                width = max(e - s, max_length)
+            else:
+                self.counter["n_examples_skip_large_windows_keep"] += 1
+                width = max_length
+            w = e - s
+            data = np.zeros((n_rows, width), dtype)
+            if n:
+                data[config.indices("bases", n), :w] = sub_enc[:, s:e]
+                data[config.indices("pw", n), :w] = pw_all[:, s:e]
+                data[config.indices("ip", n), :w] = ip_all[:, s:e]
+                data[config.indices("strand", n)] = strand_col
+                data[config.indices("sn")] = sn_col
+            data[config.indices("ccs"), :w] = ccs_enc[s:e]
+            bq = ccs_bq_all[s:e]
+            if w < width:  # Read.pad pads quality scores with -1
+                bq_full = np.full(width, -1, ccs_bq_all.dtype)
+                bq_full[:w] = bq
+                bq = bq_full
+            if config.use_ccs_bq:
+                data[config.indices("ccs_bq"), :w] = bq[:w]
+                data[config.indices("ccs_bq"), w:] = -1
+            yield {
+                "subreads": data[:, :, None],
+                "subreads/num_passes": keep,
+                "window_pos": int(w_idx[valid].min()),
+                "ccs_base_quality_scores": bq,
+                "overflow": overflow,
+                **scalars,
+            }
+
     def stack_subread_feature(self, name: str) -> np.ndarray:
         max_passes = self.config.max_passes
         return np.stack(

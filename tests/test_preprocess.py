@@ -283,6 +283,77 @@ def _make_zmw(n_subreads=3, length=250, seed=0):
     return DcExample("m/7/ccs", spaced, DcConfig(20, 100))
 
 
+def _make_zmw_with_insertions(length=250, seed=3, use_ccs_bq=False,
+                              window_widths=None):
+    """Spaced DcExample whose subreads carry insertions (real gap
+    columns) and varied strands/pw/ip."""
+    rng = np.random.default_rng(seed)
+    seq = "".join(rng.choice(list("ATCG"), size=length))
+    reads = []
+    for i in range(4):
+        ins_pos = int(rng.integers(1, length - 1))
+        ins_len = int(rng.integers(1, 4))
+        ins = "".join(rng.choice(list("ATCG"), size=ins_len))
+        full = seq[:ins_pos] + ins + seq[ins_pos:]
+        cig = f"{ins_pos}M{ins_len}I{length - ins_pos}M"
+        n = len(full)
+        seg = make_segment(
+            full, cig,
+            ip=rng.integers(0, 60, n).tolist(),
+            pw=rng.integers(0, 60, n).tolist(),
+            reverse=bool(i % 2),
+        )
+        seg.qname = f"m/7/{i}"
+        reads.append(expand_clip_indent(seg))
+    ccs = R.Read(
+        name="m/7/ccs",
+        bases=np.array(list(seq), dtype="<U1"),
+        cigar=np.repeat(np.uint8(C["M"]), length),
+        pw=np.zeros(length, dtype=np.uint8),
+        ip=np.zeros(length, dtype=np.uint8),
+        sn=np.zeros(4),
+        strand=constants.Strand.UNKNOWN,
+        base_quality_scores=rng.integers(10, 50, length).astype(np.int16),
+        ccs_idx=np.arange(length),
+        ec=12.0, np_num_passes=10, rq=0.999, rg="rg1",
+    )
+    reads.append(ccs)
+    spaced = R.space_out_subreads(reads)
+    return DcExample("m/7/ccs", spaced, DcConfig(20, 100, use_ccs_bq),
+                     window_widths=window_widths)
+
+
+@pytest.mark.parametrize("case", [
+    "plain", "ccs_bq", "smart_overflow", "smart_mixed",
+])
+def test_iter_feature_dicts_matches_slow(case):
+    """The vectorized inference fast path emits dicts identical to the
+    per-window iter_examples()/to_features_dict() path, counters too.
+    (CCS-only ZMWs are not covered: the per-window path itself rejects
+    n_subreads=0, and the grouper never yields such a ZMW.)"""
+    kwargs = {}
+    if case == "ccs_bq":
+        kwargs["use_ccs_bq"] = True
+    elif case == "smart_overflow":
+        kwargs["window_widths"] = np.array([120, 80, 50])
+    elif case == "smart_mixed":
+        kwargs["window_widths"] = np.array([90, 150, 10])
+    ex_fast = _make_zmw_with_insertions(**kwargs)
+    ex_slow = _make_zmw_with_insertions(**kwargs)
+    fast = list(ex_fast.iter_feature_dicts())
+    slow = [x.to_features_dict() for x in ex_slow.iter_examples()]
+    assert ex_fast.counter == ex_slow.counter
+    assert len(fast) == len(slow) and fast
+    for f, s in zip(fast, slow):
+        assert set(f) == set(s)
+        for key in s:
+            if isinstance(s[key], np.ndarray):
+                assert f[key].shape == s[key].shape, key
+                np.testing.assert_array_equal(f[key], s[key], err_msg=key)
+            else:
+                assert f[key] == s[key], key
+
+
 def test_dc_example_windows_and_features():
     ex = _make_zmw(length=250)
     assert ex.n_subreads == 3
