@@ -32,7 +32,6 @@ import torch
 from deepconsensus_amd.calibration import calibration as calibration_lib
 from deepconsensus_amd.models import checkpoint as ckpt_lib
 from deepconsensus_amd.models import config as cfg
-from deepconsensus_amd.models import data as data_lib
 from deepconsensus_amd.models.model import get_model
 from deepconsensus_amd.models.runner import InferenceRunner
 from deepconsensus_amd.postprocess import stitch as stitch_utils
@@ -151,22 +150,18 @@ def preprocess_one_zmw(one_zmw) -> ZmwWindows:
         if expand_counter:
             empty.counter.update(expand_counter)
         return empty
-    # Worker-side finishing, keeping the serial batch loop thin:
-    #  * format_rows (PW/IP/SN clipping) runs HERE, in parallel;
-    #  * features are integral-after-truncation for the model (the
-    #    embedding casts to int; SN fractions truncate identically) and
-    #    <= SN_MAX after clipping, so they cross the process boundary as
-    #    int16 — halving the pickle volume.
-    fmt_params = cfg.Params(
-        max_passes=dc_config.max_passes,
-        use_ccs_bq=dc_config.use_ccs_bq,
-        total_rows=dc_config.tensor_height,
-        PW_MAX=255, IP_MAX=255, SN_MAX=500,  # config.py:89-92 defaults
-    )
+    # Worker-side finishing, keeping the serial batch loop thin: the
+    # fast path applies format_rows' PW/IP/SN clipping at ZMW level and
+    # emits int16 matrices directly (integral-after-truncation for the
+    # model — the embedding casts to int, SN fractions truncate
+    # identically — and int16 halves the pickle + H2D volume).
     model_rows: List[np.ndarray] = []
     window_pos: List[int] = []
     skipped: List[stitch_utils.DCModelOutput] = []
-    for f in dc_whole.iter_feature_dicts():
+    for f in dc_whole.iter_feature_dicts(
+        pw_max=255, ip_max=255, sn_max=500,  # config.py:89-92 defaults
+        out_dtype=np.int16,
+    ):
         skip = bool(f["overflow"])
         if not skip and options.skip_windows_above:
             avg_q = phred.avg_phred(f["ccs_base_quality_scores"])
@@ -174,8 +169,7 @@ def preprocess_one_zmw(one_zmw) -> ZmwWindows:
         if skip:
             skipped.append(process_skipped_window(f, options))
             continue
-        rows = data_lib.format_rows(np.asarray(f["subreads"]), fmt_params)
-        model_rows.append(rows[:, :, 0].astype(np.int16))
+        model_rows.append(f["subreads"][:, :, 0])
         window_pos.append(int(f["window_pos"]))
     counter = dc_whole.counter
     if expand_counter:

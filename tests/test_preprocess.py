@@ -354,6 +354,37 @@ def test_iter_feature_dicts_matches_slow(case):
                 assert f[key] == s[key], key
 
 
+@pytest.mark.parametrize("case", ["plain", "smart_overflow"])
+def test_iter_feature_dicts_formatted_matches_format_rows(case):
+    """Fast path with clipping/int16 == slow dicts + format_rows +
+    astype(int16) (the worker's previous finishing steps)."""
+    from deepconsensus_amd.models import data as data_lib
+    from deepconsensus_amd.models.config import Params
+
+    kwargs = {}
+    if case == "smart_overflow":
+        kwargs["window_widths"] = np.array([120, 80, 50])
+    ex_fast = _make_zmw_with_insertions(**kwargs)
+    ex_slow = _make_zmw_with_insertions(**kwargs)
+    fast = list(ex_fast.iter_feature_dicts(
+        pw_max=255, ip_max=255, sn_max=500, out_dtype=np.int16
+    ))
+    fmt_params = Params(
+        max_passes=20, use_ccs_bq=False,
+        total_rows=ex_slow.config.tensor_height,
+        PW_MAX=255, IP_MAX=255, SN_MAX=500,
+    )
+    slow = [x.to_features_dict() for x in ex_slow.iter_examples()]
+    assert len(fast) == len(slow) and fast
+    for f, s in zip(fast, slow):
+        assert f["fmt"] is True
+        want = data_lib.format_rows(s["subreads"], fmt_params).astype(
+            np.int16
+        )
+        assert f["subreads"].dtype == np.int16
+        np.testing.assert_array_equal(f["subreads"], want)
+
+
 def test_dc_example_windows_and_features():
     ex = _make_zmw(length=250)
     assert ex.n_subreads == 3
