@@ -116,6 +116,21 @@ def test_cli_dispatch_help(capsys):
     assert "run" in out and "preprocess" in out
 
 
+def test_cli_index_subcommand(tmp_path, capsys):
+    """`deepconsensus index <bam>` writes a loadable sidecar."""
+    from deepconsensus_amd import cli
+    from deepconsensus_amd.dcio import bam as bam_lib
+    from tests.test_io_and_pipeline import make_test_bams
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=3)
+    cli.main(["index", sub, ccs])
+    out = capsys.readouterr().out
+    assert "3 ZMW groups" in out
+    idx = bam_lib.load_zmw_index(sub)
+    assert idx is not None and len(idx["zmw"]) == 3
+    assert bam_lib.load_zmw_index(ccs) is not None
+
+
 def test_parse_calibration_string():
     from deepconsensus_amd.calibration import calibration as cal
 
