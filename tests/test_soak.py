@@ -112,3 +112,53 @@ def test_soak_pipeline_random_configs():
                 assert len(r.sequence) == len(r.quality) > 0
             stats = json.load(open(os.path.join(td, "o.inference.json")))
             assert stats["n_zmw_processed"] == n_zmws
+
+
+@pytest.mark.soak
+def test_soak_fast_featurizer_parity():
+    """Randomized ZMWs: iter_feature_dicts (both plain and fmt/int16
+    modes) vs the per-window path, including smart windows."""
+    from test_preprocess import _make_zmw_with_insertions
+
+    from deepconsensus_amd.models import data as data_lib
+    from deepconsensus_amd.models.config import Params
+
+    rng = np.random.default_rng(4242)
+    for trial in range(150):
+        length = int(rng.integers(40, 400))
+        use_bq = bool(rng.integers(0, 2))
+        window_widths = None
+        if rng.integers(0, 2):
+            # Random smart widths summing to length, some > max_length.
+            widths = []
+            left = length
+            while left > 0:
+                w = int(min(left, rng.integers(10, 160)))
+                widths.append(w)
+                left -= w
+            window_widths = np.array(widths)
+        kwargs = dict(length=length, seed=trial, use_ccs_bq=use_bq,
+                      window_widths=window_widths)
+        fast = list(_make_zmw_with_insertions(**kwargs).iter_feature_dicts())
+        ex_slow = _make_zmw_with_insertions(**kwargs)
+        slow = [x.to_features_dict() for x in ex_slow.iter_examples()]
+        assert len(fast) == len(slow)
+        for f, s in zip(fast, slow):
+            for key in s:
+                if isinstance(s[key], np.ndarray):
+                    np.testing.assert_array_equal(f[key], s[key],
+                                                  err_msg=key)
+                else:
+                    assert f[key] == s[key], key
+        # fmt/int16 mode vs format_rows over the slow dicts.
+        fast_fmt = list(_make_zmw_with_insertions(**kwargs).iter_feature_dicts(
+            pw_max=255, ip_max=255, sn_max=500, out_dtype=np.int16
+        ))
+        fmt_params = Params(max_passes=20, use_ccs_bq=use_bq,
+                            total_rows=ex_slow.config.tensor_height,
+                            PW_MAX=255, IP_MAX=255, SN_MAX=500)
+        for f, s in zip(fast_fmt, slow):
+            want = data_lib.format_rows(s["subreads"], fmt_params)
+            np.testing.assert_array_equal(
+                f["subreads"], want.astype(np.int16)
+            )
