@@ -811,19 +811,36 @@ def build_zmw_index(path: str, out_path: Optional[str] = None,
             coffset=np.asarray(coffs, np.int64),
             uoffset=np.asarray(uoffs, np.uint16),
             sorted_flag=np.array([1 if is_sorted else 0], np.int8),
+            bam_size=np.array([os.path.getsize(path)], np.int64),
         )
     os.replace(tmp, out_path)
     return out_path
 
 
 def load_zmw_index(path: str) -> Optional[Dict[str, np.ndarray]]:
-    """Loads a sidecar written by build_zmw_index; None if absent."""
+    """Loads a sidecar written by build_zmw_index.
+
+    Returns None if absent, unreadable, or stale (the BAM's size no
+    longer matches the one recorded at build time — a stale index
+    would silently mis-shard)."""
     idx_path = path + ZMW_INDEX_SUFFIX
     if not os.path.exists(idx_path):
         return None
-    with np.load(idx_path) as z:
-        return {k: z[k] for k in ("zmw", "coffset", "uoffset",
-                                  "sorted_flag")}
+    try:
+        with np.load(idx_path) as z:
+            idx = {k: z[k] for k in ("zmw", "coffset", "uoffset",
+                                     "sorted_flag", "bam_size")}
+    except (OSError, KeyError, ValueError):
+        return None
+    if idx["bam_size"][0] != os.path.getsize(path):
+        import logging
+
+        logging.getLogger(__name__).warning(
+            "%s is stale (BAM size changed); ignoring — rerun "
+            "`deepconsensus index`", idx_path,
+        )
+        return None
+    return idx
 
 
 def _read_bgzf_blocks_parallel_offsets(

@@ -515,14 +515,19 @@ def run(
         # preprocesses batch 1 while InferenceRunner loads weights onto
         # the device and the HIP extension initializes — for short runs
         # this hides most of the pool spin-up + first-batch latency.
+        before = time.time()
         first_inputs = next(batch_iter, None)
+        timelog.add("feeder", "batch 0", before,
+                    num_zmws=len(first_inputs) if first_inputs else 0)
         if first_inputs is not None:
             pending = (first_inputs,
                        prefetcher.submit(preprocess_batch, first_inputs))
+        before = time.time()
         runner = InferenceRunner(
             params, model, device=device, calibration=calib_str,
             max_qual=options.max_base_quality,
         )
+        timelog.add("startup_runner", "startup", before)
         log.info("model on %s (native kernels: %s)", runner.device,
                  runner.native)
 
@@ -535,7 +540,15 @@ def run(
                         len(prev_inputs))
             infer_batch(prev_inputs, outputs, name)
 
-        for inputs in batch_iter:
+        while True:
+            # Serial ZMW streaming is a real pipeline stage: account it
+            # (the reference folded it into its preprocess rows).
+            before = time.time()
+            inputs = next(batch_iter, None)
+            timelog.add("feeder", f"batch {n_batches + 1}", before,
+                        num_zmws=len(inputs) if inputs else 0)
+            if inputs is None:
+                break
             fut = prefetcher.submit(preprocess_batch, inputs)
             if pending is not None:
                 prev_inputs, prev_fut = pending
