@@ -24,7 +24,7 @@ import json
 import logging
 import os
 import time
-from typing import Any, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 import numpy as np
 import torch
@@ -82,6 +82,11 @@ class InferenceOptions:
         )
     )
     batch_zmws: int = 100
+    # ZMW batches preprocessed ahead of the model loop. >1 hides device/
+    # runner startup (~20 s HIP context + extension init on a cold box)
+    # and model-time jitter behind the worker pool; memory cost is
+    # ~batch_zmws x windows x height x width x 2 bytes per slot.
+    prefetch_batches: int = 3
     ins_trim: int = 0
     use_ccs_smart_windows: bool = False
     # ZMW sharding for multi-process / multi-GPU runs ("i/N", like the
@@ -506,22 +511,37 @@ def run(
         timelog.add("stitch_and_write_fastq", batch_name, before,
                     n_examples, n_subreads, len(inputs))
 
-    # Pipelined loop: preprocess batch N+1 while batch N runs the model.
+    # Pipelined loop: up to `prefetch_batches` ZMW batches preprocess in
+    # the worker pool while the current batch runs the model.
     batch_iter = zmw_batches()
-    pending: Optional[Tuple[Sequence, Any]] = None
+    lookahead = max(1, options.prefetch_batches)
+    pending: "collections.deque" = collections.deque()
+    n_fed = 0
     n_batches = 0
-    with concurrent.futures.ThreadPoolExecutor(1) as prefetcher:
+    with concurrent.futures.ThreadPoolExecutor(lookahead) as prefetcher:
+
+        def feed_one() -> bool:
+            # Serial ZMW streaming is a real pipeline stage: account it
+            # (the reference folded it into its preprocess rows).
+            nonlocal n_fed
+            before = time.time()
+            inputs = next(batch_iter, None)
+            timelog.add("feeder", f"batch {n_fed}", before,
+                        num_zmws=len(inputs) if inputs else 0)
+            if inputs is None:
+                return False
+            pending.append(
+                (inputs, prefetcher.submit(preprocess_batch, inputs))
+            )
+            n_fed += 1
+            return True
+
         # Prime the pipeline BEFORE device/model setup: the worker pool
-        # preprocesses batch 1 while InferenceRunner loads weights onto
-        # the device and the HIP extension initializes — for short runs
-        # this hides most of the pool spin-up + first-batch latency.
-        before = time.time()
-        first_inputs = next(batch_iter, None)
-        timelog.add("feeder", "batch 0", before,
-                    num_zmws=len(first_inputs) if first_inputs else 0)
-        if first_inputs is not None:
-            pending = (first_inputs,
-                       prefetcher.submit(preprocess_batch, first_inputs))
+        # preprocesses the first batches while InferenceRunner brings up
+        # the device, loads the HIP extension and moves weights — on a
+        # cold box that startup is ~20 s and fully overlapped.
+        while len(pending) < lookahead and feed_one():
+            pass
         before = time.time()
         runner = InferenceRunner(
             params, model, device=device, calibration=calib_str,
@@ -531,34 +551,19 @@ def run(
         log.info("model on %s (native kernels: %s)", runner.device,
                  runner.native)
 
-        def drain(prev_inputs, prev_fut, name):
-            # Time spent blocked on the overlapped preprocessing (not a
-            # reference stage; extends the CSV so wall time is accounted).
+        while pending:
+            prev_inputs, prev_fut = pending.popleft()
+            name = f"batch {n_batches}"
+            # Time blocked on the overlapped preprocessing (not a
+            # reference stage; extends the CSV so wall is accounted).
             before = time.time()
             outputs = prev_fut.result()
             timelog.add("wait_preprocess", name, before, None, None,
                         len(prev_inputs))
             infer_batch(prev_inputs, outputs, name)
-
-        while True:
-            # Serial ZMW streaming is a real pipeline stage: account it
-            # (the reference folded it into its preprocess rows).
-            before = time.time()
-            inputs = next(batch_iter, None)
-            timelog.add("feeder", f"batch {n_batches + 1}", before,
-                        num_zmws=len(inputs) if inputs else 0)
-            if inputs is None:
-                break
-            fut = prefetcher.submit(preprocess_batch, inputs)
-            if pending is not None:
-                prev_inputs, prev_fut = pending
-                drain(prev_inputs, prev_fut, f"batch {n_batches}")
-                n_batches += 1
-            pending = (inputs, fut)
-        if pending is not None:
-            prev_inputs, prev_fut = pending
-            drain(prev_inputs, prev_fut, f"batch {n_batches}")
             n_batches += 1
+            while len(pending) < lookahead and feed_one():
+                pass
 
     if pool is not None:
         pool.shutdown()
