@@ -185,6 +185,27 @@ def main():
         one_step(i)
     if have_cuda:
         torch.cuda.synchronize()
+
+    # Output-validity audit: the timed number only counts if the native
+    # path computes the same calls as the torch reference AT THIS BATCH
+    # (a large-batch corruption was observed on the native path —
+    # tests/test_gpu_large_batch.py). Reported in the JSON, not timed.
+    native_agree = None
+    if have_cuda and runner.native:
+        with torch.no_grad():
+            # Full-batch native forward (the corruption is batch-size
+            # dependent), torch reference on head+tail slices.
+            full = pool[0].to(device)
+            bases_n, _ = runner.forward_windows(full)
+            model_f = runner.model.float()
+            agree = []
+            for s in (slice(0, 128), slice(batch - 128, batch)):
+                probs = model_f(full[s].float(), training=False)
+                agree.append(
+                    (bases_n[s].long() == probs.argmax(-1)).float().mean()
+                )
+            native_agree = round(torch.stack(agree).mean().item(), 4)
+        torch.cuda.synchronize()
     if distributed:
         import torch.distributed as dist
 
@@ -231,6 +252,7 @@ def main():
                 "windows_per_zmw": windows_per_zmw,
                 "parallelism": f"dp{world}",
                 "native_kernels": bool(runner.native),
+                "native_vs_torch_agree": native_agree,
                 "hipgraph": used_graphs,
             },
         }
