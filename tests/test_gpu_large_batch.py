@@ -65,6 +65,47 @@ def test_embed_gather_large_batch_matches_small():
         )
 
 
+def test_condenser_matmul_large_m():
+    """The hipBLASLt condenser GEMM at M=409600 vs fp32 reference and
+    vs the same rows at small M (algorithm-selection differences)."""
+    r = _runner()
+    M = B_LARGE * 100
+    emb = (torch.randn(M, r.cond_wt.shape[0], device="cuda") * 0.2).to(
+        torch.bfloat16
+    )
+    big = emb @ r.cond_wt
+    small = emb[:4096] @ r.cond_wt
+    assert torch.equal(big[:4096], small), (
+        "condenser GEMM rows differ between M=409600 and M=4096"
+    )
+    ref = (emb[-4096:].float() @ r.cond_wt.float()).to(torch.bfloat16)
+    err = (big[-4096:].float() - ref.float()).abs().max().item()
+    assert err < 0.5, f"condenser GEMM tail rows wrong (max err {err})"
+
+
+def test_encode_native_equals_condensed_input_at_alpha0():
+    """With random-init ReZero alphas (all 0) the whole encoder stack is
+    an identity over cond+pos: encode_native at B=4096 must equal the
+    directly computed embed->condense->+pos tensor."""
+    r = _runner()
+    rows = torch.from_numpy(
+        _rows(r.params, B_LARGE).astype(np.int16)
+    ).cuda()
+    x = r.encode_native(rows)
+    emb = r.ext.embed_gather(rows.contiguous(), r.table_flat, r.row_shift,
+                             r.row_vocab, r.chunk_cnt, r.chunk_entries)
+    b, l, _ = emb.shape
+    want = (emb.reshape(b * l, -1) @ r.cond_wt).view(b, l, -1)
+    if r.pos is not None:
+        want = want + r.pos[:l]
+    bad = (x != want).any(dim=-1)
+    assert not bad.any(), (
+        f"alpha=0 encoder is not identity for {int(bad.sum())} of {b * l} "
+        f"positions; first bad (window, pos) = "
+        f"{tuple(int(v) for v in bad.nonzero()[0])}"
+    )
+
+
 def test_fused_linear_passthrough_large_m():
     """alpha=0 + residual: out must be bitwise == resid at M=409600."""
     r = _runner()
