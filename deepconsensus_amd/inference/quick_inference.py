@@ -273,7 +273,21 @@ def run_model_on_zmws(
             buf.copy_(rows_t)
             rows_t = buf
         bases_t, quals_t = runner.forward_windows(rows_t)
-        seq_mat = _SEQ_LUT[bases_t.cpu().numpy().astype(np.int64)]
+        bases_np = bases_t.cpu().numpy()
+        # Corruption tripwire: a healthy model rarely predicts majority
+        # gap. A batch-size-dependent native-path regression produced
+        # silent all-gap output (profiles/r01_perf_journal.md "GAP
+        # REGRESSION") — surface it instead of writing garbage FASTQ.
+        gap_frac = float((bases_np == 0).mean())
+        if gap_frac > 0.6 and len(chunk) > 1:
+            log.warning(
+                "model predicted %.0f%% gaps over %d windows — possible "
+                "native-path corruption (see ROADMAP item 0); consider "
+                "re-running with a smaller --batch_size or comparing "
+                "against the torch path",
+                100.0 * gap_frac, len(chunk),
+            )
+        seq_mat = _SEQ_LUT[bases_np.astype(np.int64)]
         qual_mat = (quals_t.cpu().numpy() + 33).astype(np.uint8)
         for j in range(len(chunk)):
             z = owners[i + j]
