@@ -50,6 +50,7 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
+template <bool PREFETCH>
 __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
     const bf16* __restrict__ qkv, bf16* __restrict__ out,
     int B, int L, int H, int win, float scale) {
@@ -161,11 +162,12 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
   // one-shot version (grid = B*H tiny blocks) was launch/drain-bound: 74%
   // of wave-slots parked, ~20 us wall per ~1 us of math.
   const int BH = B * H;
-  if (blockIdx.x < BH) issue_kv(blockIdx.x);
+  if (PREFETCH && blockIdx.x < BH) issue_kv(blockIdx.x);
   for (int item = blockIdx.x; item < BH; item += gridDim.x) {
     const bf16* base =
         qkv + (size_t)(item / H) * L * RS + (size_t)(item % H) * D;
     __syncthreads();  // prior item's LDS reads done before re-staging
+    if (!PREFETCH) issue_kv(item);  // DC_ATTN_PREFETCH=0 diagnostic mode
 
     // ---- Issue Q B-fragment loads (this lane's query row, 9 k-steps of
     // 16) before the LDS drain so their latency hides behind it. ----
@@ -214,8 +216,10 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
 
     // Prefetch the next item's K/V now: the loads fly during softmax + PV
     // and are drained into LDS only after the top-of-loop barrier.
-    const int next_item = item + gridDim.x;
-    if (next_item < BH) issue_kv(next_item);
+    if (PREFETCH) {
+      const int next_item = item + gridDim.x;
+      if (next_item < BH) issue_kv(next_item);
+    }
 
     // ---- Masked softmax, in-lane + one half-merge. ----
     // Key-local index of st[i]: kl(i) = 32*(i/16) + (i&3) + 8*((i&15)>>2)
@@ -307,9 +311,22 @@ at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
   dim3 grid(std::min(B * (int)H, 512));
   dim3 block(256);
   hipStream_t stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(banded_attn_mfma_kernel, grid, block, 0, stream,
-                     reinterpret_cast<bf16*>(q.data_ptr()),
-                     reinterpret_cast<bf16*>(out.data_ptr()),
-                     B, L, (int)H, (int)win, (float)scale);
+  // DC_ATTN_PREFETCH=0 disables the cross-item K/V prefetch pipeline
+  // (diagnostic knob for the large-batch corruption investigation).
+  static const bool prefetch = [] {
+    const char* e = std::getenv("DC_ATTN_PREFETCH");
+    return !(e && e[0] == '0');
+  }();
+  if (prefetch) {
+    hipLaunchKernelGGL(banded_attn_mfma_kernel<true>, grid, block, 0,
+                       stream, reinterpret_cast<bf16*>(q.data_ptr()),
+                       reinterpret_cast<bf16*>(out.data_ptr()),
+                       B, L, (int)H, (int)win, (float)scale);
+  } else {
+    hipLaunchKernelGGL(banded_attn_mfma_kernel<false>, grid, block, 0,
+                       stream, reinterpret_cast<bf16*>(q.data_ptr()),
+                       reinterpret_cast<bf16*>(out.data_ptr()),
+                       B, L, (int)H, (int)win, (float)scale);
+  }
   return out;
 }
