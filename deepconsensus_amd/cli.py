@@ -51,6 +51,9 @@ def _run_main(argv: List[str]) -> None:
                     help=".fastq or .bam output path")
     ap.add_argument("--batch_size", type=int, default=1024)
     ap.add_argument("--batch_zmws", type=int, default=100)
+    ap.add_argument("--prefetch_batches", type=int, default=3,
+                    help="ZMW batches preprocessed ahead of the model "
+                    "loop (hides device startup and model jitter)")
     ap.add_argument("--max_length", type=int, default=100)
     ap.add_argument("--min_quality", type=int, default=20)
     ap.add_argument("--min_length", type=int, default=0)
@@ -91,6 +94,7 @@ def _run_main(argv: List[str]) -> None:
         min_length=args.min_length,
         batch_size=args.batch_size,
         batch_zmws=args.batch_zmws,
+        prefetch_batches=args.prefetch_batches,
         cpus=args.cpus,
         skip_windows_above=args.skip_windows_above,
         ins_trim=args.ins_trim,

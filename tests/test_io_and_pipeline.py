@@ -330,6 +330,9 @@ def test_quick_inference_e2e_fastq(tmp_path):
     # Runtime CSV + stats JSON written (incl. the prefetch-wait stage).
     runtime = open(str(tmp_path / "out.runtime.csv")).read()
     assert "wait_preprocess" in runtime and "run_model" in runtime
+    # Full wall accounting: serial feeder pulls and device/runner
+    # startup are stages too (gpurun_out/pipe3.log attribution).
+    assert "feeder" in runtime and "startup_runner" in runtime
     stats = json.load(open(tmp_path / "out.inference.json"))
     assert stats["n_zmw_processed"] == 3
     # Every ZMW produced output (min_quality=0 disables filtering).
