@@ -1,5 +1,10 @@
 """Large-batch bisection of the native inference chain (GPU).
 
+Named test_zz_* so it sorts LAST under the round runner's `-x`: these
+tests are EXPECTED to fail while ROADMAP item 0 is open, and the rest
+of the GPU suite should run before the first diagnostic failure stops
+the session.
+
 A fresh-box repro (profiles/r01_perf_journal.md, gap-regression section)
 showed forward_windows producing ~87% gap calls at B=1908/4096 vs 6% at
 B=64 with identical input distribution — a batch-size-dependent
