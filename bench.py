@@ -189,7 +189,7 @@ def main():
     # Output-validity audit: the timed number only counts if the native
     # path computes the same calls as the torch reference AT THIS BATCH
     # (a large-batch corruption was observed on the native path —
-    # tests/test_zz_gpu_large_batch.py). Reported in the JSON, not timed.
+    # tests/test_gpu_large_batch.py). Reported in the JSON, not timed.
     native_agree = None
     if have_cuda and runner.native:
         with torch.no_grad():

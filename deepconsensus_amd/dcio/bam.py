@@ -812,6 +812,9 @@ def build_zmw_index(path: str, out_path: Optional[str] = None,
             uoffset=np.asarray(uoffs, np.uint16),
             sorted_flag=np.array([1 if is_sorted else 0], np.int8),
             bam_size=np.array([os.path.getsize(path)], np.int64),
+            # mtime (ns) guards against a rewritten BAM of identical byte
+            # size passing the staleness check.
+            bam_mtime_ns=np.array([os.stat(path).st_mtime_ns], np.int64),
         )
     os.replace(tmp, out_path)
     return out_path
@@ -830,13 +833,19 @@ def load_zmw_index(path: str) -> Optional[Dict[str, np.ndarray]]:
         with np.load(idx_path) as z:
             idx = {k: z[k] for k in ("zmw", "coffset", "uoffset",
                                      "sorted_flag", "bam_size")}
+            # Older sidecars lack the mtime field; treat as unknown.
+            mtime = (int(z["bam_mtime_ns"][0])
+                     if "bam_mtime_ns" in z.files else None)
     except (OSError, KeyError, ValueError):
         return None
-    if idx["bam_size"][0] != os.path.getsize(path):
+    stale = idx["bam_size"][0] != os.path.getsize(path) or (
+        mtime is not None and mtime != os.stat(path).st_mtime_ns
+    )
+    if stale:
         import logging
 
         logging.getLogger(__name__).warning(
-            "%s is stale (BAM size changed); ignoring — rerun "
+            "%s is stale (BAM size or mtime changed); ignoring — rerun "
             "`deepconsensus index`", idx_path,
         )
         return None

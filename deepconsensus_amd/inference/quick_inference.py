@@ -279,13 +279,35 @@ def run_model_on_zmws(
         # silent all-gap output (profiles/r01_perf_journal.md "GAP
         # REGRESSION") — surface it instead of writing garbage FASTQ.
         gap_frac = float((bases_np == 0).mean())
-        if gap_frac > 0.6 and len(chunk) > 1:
-            log.warning(
-                "model predicted %.0f%% gaps over %d windows — possible "
-                "native-path corruption (see ROADMAP item 0); consider "
-                "re-running with a smaller --batch_size or comparing "
-                "against the torch path",
-                100.0 * gap_frac, len(chunk),
+        if gap_frac > 0.6 and len(chunk) > 1 and getattr(
+            runner, "native", False
+        ):
+            # Corruption tripwire turned hard gate: audit a
+            # sample of the suspect batch against the fp32 torch path
+            # and refuse to write FASTQ from a disagreeing native pass.
+            sample = min(32, len(chunk))
+            with torch.no_grad():
+                probs = runner.model(
+                    rows_t[:sample].to(runner.device).float(),
+                    training=False,
+                )
+            torch_bases = probs.argmax(-1).cpu().numpy()
+            agree = float((bases_np[:sample] == torch_bases).mean())
+            if agree < 0.99:
+                raise RuntimeError(
+                    f"native inference path disagrees with the torch "
+                    f"reference on {100 * (1 - agree):.1f}% of base "
+                    f"calls in a {100 * gap_frac:.0f}%-gap batch of "
+                    f"{len(chunk)} windows — refusing to write "
+                    f"corrupted FASTQ. This indicates a native-kernel "
+                    f"regression; re-run with DC_FUSED_FFN=0/"
+                    f"smaller --batch_size and file a bug."
+                )
+            log.info(
+                "high gap fraction %.0f%% over %d windows, but the "
+                "native path agrees with the torch reference "
+                "(%.4f) — input is legitimately gappy",
+                100.0 * gap_frac, len(chunk), agree,
             )
         seq_mat = _SEQ_LUT[bases_np.astype(np.int64)]
         qual_mat = (quals_t.cpu().numpy() + 33).astype(np.uint8)

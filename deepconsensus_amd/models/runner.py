@@ -109,6 +109,21 @@ def build_gather_tables(model: EncoderOnlyLearnedValuesTransformer):
             entries.append((r, base, w))
             j += w
         assert len(entries) <= 4, "chunk spans >4 table rows"
+        # Kernel contract: a chunk is either ONE width-8 entry (16-B
+        # gather) or up to four width-2 entries (one u32 gather per
+        # output dword). Anything else means the row layout misaligns
+        # a wide table against the 8-column grid — fail loudly here
+        # rather than gather garbage on device.
+        if len(entries) == 1:
+            assert entries[0][2] == 8, (
+                f"chunk {c}: single entry must be width 8, got "
+                f"{entries[0][2]} (row layout misaligned)"
+            )
+        else:
+            assert all(w == 2 for _, _, w in entries), (
+                f"chunk {c}: multi-entry chunk must be all width-2 "
+                f"entries, got {[w for _, _, w in entries]}"
+            )
         chunk_cnt[c] = len(entries)
         for k, (r, base, w) in enumerate(entries):
             chunk_entries[c * 4 + k] = torch.tensor([r, base, w, 0])

@@ -71,26 +71,28 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
       raw = *reinterpret_cast<const uint4*>(
           table_flat + e.y + (size_t)id * 8);
     } else {
-      // Narrow rows (e.g. 4 x width-2 strand entries): accumulate into two
-      // 64-bit registers — a local ushort[8] would land in scratch
-      // (runtime-indexed array, cdna_hip_programming.md rule 20).
-      unsigned long long lo = 0, hi = 0;
-      int pos = 0;
-      for (int k = 0; k < cnt; ++k) {
-        const int4 e = chunk_entries[c * 4 + k];
-        const int id = ids[e.x][p];
-        const ushort* src = reinterpret_cast<const ushort*>(
-            table_flat + e.y + (size_t)id * e.z);
-        for (int j = 0; j < e.z; ++j, ++pos) {
-          const unsigned long long v = src[j];
-          if (pos < 4) lo |= v << (16 * pos);
-          else hi |= v << (16 * (pos - 4));
+      // Narrow rows: the host builder guarantees a multi-entry chunk is
+      // exactly `cnt` width-2 entries (strand), so entry k fills output
+      // dword k as one aligned u32 load (element offset e.y + id*2 is
+      // even -> 4-B aligned). Fully unrolled, static indexing (rule 20).
+      // NOTE: the previous shape here (runtime-depth loop + 64-bit shift
+      // accumulation) produced NONDETERMINISTIC garbage in the upper
+      // dwords whenever >1 block was resident per CU (profiles/
+      // r02_embed_gather_bug.md) — keep this path branch-free and flat.
+      unsigned d[4] = {0u, 0u, 0u, 0u};
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        if (k < cnt) {
+          const int4 e = chunk_entries[c * 4 + k];
+          const int id = ids[e.x][p];
+          d[k] = *reinterpret_cast<const unsigned*>(
+              table_flat + e.y + (size_t)id * 2);
         }
       }
-      raw.x = (unsigned)lo;
-      raw.y = (unsigned)(lo >> 32);
-      raw.z = (unsigned)hi;
-      raw.w = (unsigned)(hi >> 32);
+      raw.x = d[0];
+      raw.y = d[1];
+      raw.z = d[2];
+      raw.w = d[3];
     }
     *reinterpret_cast<uint4*>(
         out + ((size_t)b * L + l0 + p) * H + c * 8) = raw;

@@ -229,36 +229,51 @@ class RawSubreadGrouper:
     def __iter__(self):
         return self
 
-    def _emit(self, group: List[bytes]) -> List[bytes]:
+    def _count_run(self) -> None:
+        """Counts one consumed zm-run against max_groups.
+
+        Counts EVERY run (emitted or all-unmapped/empty) so the
+        accounting matches bam.build_zmw_index, which records every
+        zm-run boundary: a shard limited to max_groups=k must stop at
+        the k-th index entry even when some runs emit nothing,
+        otherwise it reads past its byte range and duplicates the next
+        shard's first ZMWs."""
         self.groups_out += 1
         if self.max_groups is not None and self.groups_out >= self.max_groups:
             self.keep_iter = False
-        return group
 
     def __next__(self) -> List[bytes]:
         if not self.keep_iter:
             raise StopIteration
+        exhausted = False
         while self.keep_iter:
             try:
                 buf = next(self._iter)
-                if bam.raw_flag(buf) & bam.FUNMAP:
-                    continue
             except StopIteration:
                 self.keep_iter = False
+                exhausted = True
                 break
             read_zmw = bam.raw_tag(buf, "zm")
             if read_zmw == self.zmw:
-                self.subread_group.append(buf)
-            else:
-                subreads_set = self.subread_group
-                self.subread_group = [buf]
-                self.zmw = read_zmw
-                if subreads_set:
-                    return self._emit(subreads_set)
-        if self.subread_group:
+                if not bam.raw_flag(buf) & bam.FUNMAP:
+                    self.subread_group.append(buf)
+                continue
+            # zm-run boundary: count the finished run, stage the new one.
+            subreads_set = self.subread_group
+            self.subread_group = (
+                [] if bam.raw_flag(buf) & bam.FUNMAP else [buf]
+            )
+            self.zmw = read_zmw
+            self._count_run()
+            if subreads_set:
+                return subreads_set
+            # All-unmapped run: counted but nothing to emit; keep
+            # scanning (the while condition honors a max_groups stop).
+        if exhausted and self.subread_group:
             out = self.subread_group
             self.subread_group = []
-            return self._emit(out)
+            self._count_run()
+            return out
         raise StopIteration
 
 

@@ -1,9 +1,9 @@
 """Large-batch bisection of the native inference chain (GPU).
 
-Named test_zz_* so it sorts LAST under the round runner's `-x`: these
-tests are EXPECTED to fail while ROADMAP item 0 is open, and the rest
-of the GPU suite should run before the first diagnostic failure stops
-the session.
+Regression suite for the (fixed) round-1 large-batch corruption: each
+test isolates one stage of the alpha=0 chain at B=4096 against a torch
+reference, so any future failure names the broken kernel directly.
+Root cause + forensics: profiles/r02_embed_gather_bug.md.
 
 A fresh-box repro (profiles/r01_perf_journal.md, gap-regression section)
 showed forward_windows producing ~87% gap calls at B=1908/4096 vs 6% at
