@@ -28,6 +28,7 @@ from deepconsensus_amd.models.model import (
     get_model,
 )
 from deepconsensus_amd.utils import constants
+from deepconsensus_amd.utils import trace
 
 
 def build_gather_tables(model: EncoderOnlyLearnedValuesTransformer):
@@ -313,12 +314,14 @@ class InferenceRunner:
     @torch.no_grad()
     def encode_native(self, rows: torch.Tensor) -> torch.Tensor:
         """Native path up to (but excluding) the final LayerNorm: [B,L,H] bf16."""
-        emb = self.ext.embed_gather(
-            rows.contiguous(), self.table_flat, self.row_shift,
-            self.row_vocab, self.chunk_cnt, self.chunk_entries,
-        )  # [B, L, concat] bf16
+        with trace.range("K2_embed_gather"):
+            emb = self.ext.embed_gather(
+                rows.contiguous(), self.table_flat, self.row_shift,
+                self.row_vocab, self.chunk_cnt, self.chunk_entries,
+            )  # [B, L, concat] bf16
         b, l, _ = emb.shape
-        x = emb.reshape(b * l, -1) @ self.cond_wt  # [B*L, H]
+        with trace.range("K3_condenser"):
+            x = emb.reshape(b * l, -1) @ self.cond_wt  # [B*L, H]
         x = x.view(b, l, -1)
         if self.pos is not None:
             x = x + self.pos[:l]
@@ -327,33 +330,39 @@ class InferenceRunner:
             flat = x.reshape(b * l, h)
             empty = flat.new_empty(0)
             for lw in self.layer_w:
-                if self.ffn_fused_ok:
-                    qkv = self.ext.fused_linear(
-                        flat, lw["wqkv_pad"], empty, empty, 840, False, 0.0
-                    ).view(b, l, -1)
-                else:
-                    qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
-                a = self._attn(qkv)
-                if self.ffn_fused_ok:
-                    flat = self.ext.fused_linear(
-                        a.view(b * l, h), lw["wout_pad"], empty, flat,
-                        280, False, lw["alpha_attn"],
-                    )
-                    if self.ffn_v3:
-                        flat = self.ext.fused_ffn_v3(
-                            flat, lw["w1_v2"], lw["w2_pad"],
-                            lw["b2_f32"], lw["alpha_ffn"],
-                        )
-                    elif self.ffn_v2:
-                        flat = self.ext.fused_ffn_v2(
-                            flat, lw["w1_v2"], lw["w2_pad"],
-                            lw["b2_f32"], lw["alpha_ffn"],
-                        )
+                with trace.range("K5_qkv_proj"):
+                    if self.ffn_fused_ok:
+                        qkv = self.ext.fused_linear(
+                            flat, lw["wqkv_pad"], empty, empty, 840,
+                            False, 0.0
+                        ).view(b, l, -1)
                     else:
-                        flat = self.ext.fused_ffn(
-                            flat, lw["w1_pad"], lw["b1_f32"], lw["w2_pad"],
-                            lw["b2_f32"], lw["alpha_ffn"],
+                        qkv = (flat @ lw["wqkv_t"]).view(b, l, -1)
+                with trace.range("K6_banded_attn"):
+                    a = self._attn(qkv)
+                if self.ffn_fused_ok:
+                    with trace.range("K7_out_proj"):
+                        flat = self.ext.fused_linear(
+                            a.view(b * l, h), lw["wout_pad"], empty, flat,
+                            280, False, lw["alpha_attn"],
                         )
+                    with trace.range("K9_fused_ffn"):
+                        if self.ffn_v3:
+                            flat = self.ext.fused_ffn_v3(
+                                flat, lw["w1_v2"], lw["w2_pad"],
+                                lw["b2_f32"], lw["alpha_ffn"],
+                            )
+                        elif self.ffn_v2:
+                            flat = self.ext.fused_ffn_v2(
+                                flat, lw["w1_v2"], lw["w2_pad"],
+                                lw["b2_f32"], lw["alpha_ffn"],
+                            )
+                        else:
+                            flat = self.ext.fused_ffn(
+                                flat, lw["w1_pad"], lw["b1_f32"],
+                                lw["w2_pad"], lw["b2_f32"],
+                                lw["alpha_ffn"],
+                            )
                     continue
                 # Residual fused into the GEMM epilogue (alpha pre-folded).
                 flat = torch.addmm(flat, a.view(b * l, h), lw["wout_t_a"])
@@ -377,7 +386,8 @@ class InferenceRunner:
         rows = rows.to(self.device, non_blocking=True)
         if self.native:
             x = self.encode_native(rows)
-            out = self.ext.fused_ln_head_qv(
+            with trace.range("K10_K12_ln_head_qv"):
+                out = self.ext.fused_ln_head_qv(
                 x.reshape(-1, x.shape[-1]),
                 self.ln_gamma,
                 self.ln_beta,

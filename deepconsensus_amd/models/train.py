@@ -28,6 +28,8 @@ from deepconsensus_amd.models import lamb as lamb_lib
 from deepconsensus_amd.models import losses as losses_lib
 from deepconsensus_amd.models.model import get_model
 from deepconsensus_amd.parallel import comm
+from deepconsensus_amd.utils import trace
+from deepconsensus_amd.utils.events import EventWriter
 
 log = logging.getLogger(__name__)
 
@@ -133,6 +135,14 @@ def train_model(
     summary = {}
     t0 = time.time()
     steps_this_session = 0
+    # TensorBoard-equivalent event files (model_utils.py:549-583):
+    # train/ and eval/ writers under <out_dir>/summaries.
+    train_writer = eval_writer = None
+    if main:
+        train_writer = EventWriter(os.path.join(out_dir, "summaries",
+                                                "train"))
+        eval_writer = EventWriter(os.path.join(out_dir, "summaries",
+                                               "eval"))
     for epoch in range(initial_epoch, params.num_epochs):
         for batch in train_ds.iterate(epoch):
             lr = schedule.apply(optimizer, step)
@@ -141,25 +151,39 @@ def train_model(
             # bf16 autocast (BASELINE config #4): GEMMs/attention run bf16,
             # LN/softmax and the alignment loss stay fp32; fp32 master
             # weights and fp32 gradient all-reduce.
-            if use_bf16 and rows.is_cuda:
-                with torch.autocast("cuda", dtype=torch.bfloat16):
-                    probs = model(rows, training=True)
-            else:
-                probs = model(rows, training=True)
-            # compute_average_loss: sum / global batch
-            # (model_train_custom_loop.py:148-154).
-            loss = loss_fn(label, probs.float()) / global_batch
-            loss.backward()
-            reducer.reduce()
-            optimizer.step()
+            with trace.range("train_step"):
+                if use_bf16 and rows.is_cuda:
+                    with torch.autocast("cuda", dtype=torch.bfloat16):
+                        with trace.range("forward"):
+                            probs = model(rows, training=True)
+                else:
+                    with trace.range("forward"):
+                        probs = model(rows, training=True)
+                # compute_average_loss: sum / global batch
+                # (model_train_custom_loop.py:148-154).
+                with trace.range("alignment_loss"):
+                    loss = loss_fn(label, probs.float()) / global_batch
+                with trace.range("backward"):
+                    loss.backward()
+                with trace.range("allreduce"):
+                    reducer.reduce()
+                with trace.range("lamb_step"):
+                    optimizer.step()
             step += 1
             steps_this_session += 1
             if main and step % 10 == 0:
+                steps_per_sec = steps_this_session / (time.time() - t0)
                 log.info(
                     "epoch %d step %d loss %.4f lr %.2e (%.2f steps/s)",
-                    epoch, step, float(loss) * world, lr,
-                    steps_this_session / (time.time() - t0),
+                    epoch, step, float(loss) * world, lr, steps_per_sec,
                 )
+                train_writer.add_scalars(step, {
+                    "train/loss": float(loss) * world,
+                    "train/learning_rate": lr,
+                    "train/steps_per_second": steps_per_sec,
+                    "train/epoch": epoch + (step - epoch * steps_per_epoch)
+                    / max(steps_per_epoch, 1),
+                })
             if step % eval_every == 0 or (
                 limit_steps and steps_this_session >= limit_steps
             ):
@@ -178,6 +202,9 @@ def train_model(
                         metrics["eval/per_example_accuracy"],
                     )
                     log.info("eval @%d: %s", step, metrics)
+                    eval_writer.add_scalars(step, metrics)
+                    eval_writer.flush()
+                    train_writer.flush()
                 summary = metrics
             if limit_steps and steps_this_session >= limit_steps:
                 break
@@ -198,6 +225,9 @@ def train_model(
     summary = metrics
     summary["steps"] = step
     if main:
+        eval_writer.add_scalars(step, metrics)
+        train_writer.close()
+        eval_writer.close()
         with open(os.path.join(out_dir, "training_summary.json"), "w") as f:
             json.dump(summary, f, indent=2)
     return summary

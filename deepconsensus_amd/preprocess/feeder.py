@@ -368,7 +368,17 @@ def create_proc_feeder(
                     bam.raw_ref_id(read_set[0])
                 ][0]
                 while True:
-                    ccs_bam_read = next(ccs_bam_h)
+                    try:
+                        ccs_bam_read = next(ccs_bam_h)
+                    except StopIteration:
+                        # PEP 479 would surface this as an opaque
+                        # RuntimeError; name the missing ZMW instead
+                        # (ccs filtering or a stale/mismatched index).
+                        raise ValueError(
+                            f"ccs bam does not contain {ccs_seqname} "
+                            "(exhausted while matching qnames — ccs "
+                            "filtering or stale ZMW index?)"
+                        ) from None
                     if bam.raw_qname(ccs_bam_read) == ccs_seqname:
                         break
                 window_widths = None
@@ -385,11 +395,15 @@ def create_proc_feeder(
                 continue
             ccs_seqname = read_set[0].reference_name
             while True:
-                ccs_bam_read = next(ccs_bam_h)
+                try:
+                    ccs_bam_read = next(ccs_bam_h)
+                except StopIteration:
+                    raise ValueError(
+                        f"ccs bam does not contain {ccs_seqname} "
+                        "(exhausted while matching qnames)"
+                    ) from None
                 if ccs_bam_read.qname == ccs_seqname:
                     break
-            if ccs_bam_read.qname != ccs_seqname:
-                raise ValueError(f"ccs bam does not contain {ccs_seqname}")
             window_widths = None
             if use_ccs_smart_windows:
                 window_widths = np.array(ccs_bam_read.get_tag("wl"))

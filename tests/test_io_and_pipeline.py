@@ -17,8 +17,14 @@ def _random_seq(rng, n):
     return "".join(rng.choice(list("ATCG"), size=n))
 
 
-def make_test_bams(tmp_path, n_zmws=3, length=220, n_subreads=4, seed=7):
-    """Writes synthetic subreads_to_ccs.bam + ccs.bam; returns paths."""
+def make_test_bams(tmp_path, n_zmws=3, length=220, n_subreads=4, seed=7,
+                   unmapped_zmws=()):
+    """Writes synthetic subreads_to_ccs.bam + ccs.bam; returns paths.
+
+    ``unmapped_zmws``: ZMW ordinals (0-based) whose subreads are all
+    written FUNMAP — such a zm-run is indexed by build_zmw_index but
+    never emitted by the grouper (the sharding accounting regression
+    from ADVICE r1)."""
     rng = np.random.default_rng(seed)
     refs = []
     zmw_seqs = {}
@@ -34,6 +40,19 @@ def make_test_bams(tmp_path, n_zmws=3, length=220, n_subreads=4, seed=7):
         for rid, (name, ln) in enumerate(refs):
             zm = int(name.split("/")[1])
             seq = zmw_seqs[name]
+            if rid in unmapped_zmws:
+                read = bam_lib.BamRead(
+                    qname=f"m000/{zm}/0_{ln}",
+                    flag=4,
+                    ref_id=-1,
+                    pos=-1,
+                    cigartuples=[],
+                    seq=seq,
+                    query_qualities=[30] * ln,
+                    tags={"zm": zm},
+                )
+                w.write(read)
+                continue
             for i in range(n_subreads):
                 # Introduce a small mutation region per subread.
                 s = list(seq)
@@ -213,6 +232,48 @@ def test_feeder_byte_range_sharding(tmp_path):
         assert not (set(s) & set(seen))
         seen.update(s)
     assert seen == full
+
+
+def test_feeder_byte_range_sharding_with_unmapped_zmw(tmp_path):
+    """An all-unmapped zm-run is indexed but never emitted; shards must
+    still be disjoint and complete (the run counts toward max_groups —
+    ADVICE r1: a shard otherwise read past its boundary and duplicated
+    the next shard's first ZMW)."""
+    import collections
+
+    from deepconsensus_amd.preprocess import feeder as pre_feeder
+    from deepconsensus_amd.preprocess.windows import DcConfig
+
+    # ZMW ordinal 1 sits inside shard 0's range of the 6-group index.
+    sub, ccs = make_test_bams(tmp_path, n_zmws=6, unmapped_zmws=(1, 4))
+    bam_lib.build_zmw_index(sub)
+    bam_lib.build_zmw_index(ccs)
+    idx = bam_lib.load_zmw_index(sub)
+    assert len(idx["zmw"]) == 6  # all-unmapped runs ARE indexed
+    dc_config = DcConfig(20, 100, False)
+
+    def collect(shard_index, shard_count):
+        pf, _ = pre_feeder.create_proc_feeder(
+            subreads_to_ccs=sub, ccs_bam=ccs, dc_config=dc_config,
+            defer_expansion=True, shard_index=shard_index,
+            shard_count=shard_count,
+        )
+        out = {}
+        for job, zmw, *_ in pf():
+            reads = job.materialize(collections.Counter())
+            out[zmw] = [r.name for r in reads]
+        return out
+
+    full = collect(0, 1)
+    assert len(full) == 4  # unmapped ZMWs never emit
+    for n_shards in (2, 3):
+        seen = {}
+        for i in range(n_shards):
+            s = collect(i, n_shards)
+            dup = set(s) & set(seen)
+            assert not dup, f"duplicate ZMWs across shards: {dup}"
+            seen.update(s)
+        assert seen == full, (n_shards, sorted(seen), sorted(full))
 
 
 def test_quick_inference_sharding_with_index(tmp_path):
