@@ -260,8 +260,53 @@ class AlignmentMetric:
         self._pid_sum = 0.0
         self._pid_n = 0
 
+    def _device_alignment(self, y_true, y_pred):
+        """K14 on device: affine-NW counts via the HIP wavefront kernel
+        (ops/hip/alignment_metric.hip). Same mv contract as the numpy
+        path; paths are not materialized (counts only)."""
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        yt = left_shift_sequence(y_true.long()).int()
+        yp_tok = left_shift_sequence(y_pred.argmax(-1).long()).int()
+        yt_len = (yt != GAP).sum(-1).int()
+        yp_len = (yp_tok != GAP).sum(-1).int()
+        v_opt, counts = ext.alignment_metric_counts(
+            yt, yp_tok, yt_len, yp_len,
+            self.matching_score, self.mismatch_penalty,
+            self.gap_open_penalty, self.gap_extend_penalty,
+        )
+        c = counts.cpu().numpy().astype(np.int64)
+        mv = {
+            "num_matches": c[:, 0],
+            "num_insertions": c[:, 1],
+            "num_deletions": c[:, 2],
+            "num_correct_matches": c[:, 3],
+        }
+        mv["alignment_length"] = (
+            mv["num_matches"] + mv["num_insertions"] + mv["num_deletions"]
+        )
+        with np.errstate(divide="ignore", invalid="ignore"):
+            unsafe = mv["num_correct_matches"] / mv["alignment_length"]
+        mv["pid"] = np.where(mv["alignment_length"] > 0, unsafe, 1.0)
+        return v_opt.cpu().numpy(), None, mv
+
     def alignment(self, y_true, y_pred):
-        """Returns (v_opt, paths, metric_values)."""
+        """Returns (v_opt, paths, metric_values).
+
+        On CUDA inputs with the HIP extension available, runs the K14
+        device kernel (paths returned as None); otherwise the numpy
+        reference path."""
+        if (
+            isinstance(y_true, torch.Tensor)
+            and y_true.is_cuda
+            and isinstance(y_pred, torch.Tensor)
+            and y_pred.is_cuda
+        ):
+            from deepconsensus_amd import ops as dc_ops
+
+            if dc_ops.have_ext():
+                return self._device_alignment(y_true, y_pred)
         y_true = np.asarray(
             y_true.detach().cpu().numpy()
             if isinstance(y_true, torch.Tensor)
@@ -431,10 +476,16 @@ def get_batch_identity_ccs_pred(
     """(identity_ccs, identity_pred) (losses_and_metrics.py:1061-1098)."""
     _, _, mv_pred = alignment_metric.alignment(y_true, y_pred)
     identity_pred = per_batch_identity(mv_pred)
-    ccs_np = (
-        ccs.detach().cpu().numpy() if isinstance(ccs, torch.Tensor) else ccs
-    ).astype(np.int64)
-    ccs_oh = np.eye(constants.SEQ_VOCAB_SIZE, dtype=np.float32)[ccs_np]
+    if isinstance(ccs, torch.Tensor) and ccs.is_cuda:
+        ccs_oh = torch.nn.functional.one_hot(
+            ccs.long(), constants.SEQ_VOCAB_SIZE
+        ).float()
+    else:
+        ccs_np = (
+            ccs.detach().cpu().numpy()
+            if isinstance(ccs, torch.Tensor) else ccs
+        ).astype(np.int64)
+        ccs_oh = np.eye(constants.SEQ_VOCAB_SIZE, dtype=np.float32)[ccs_np]
     _, _, mv_ccs = alignment_metric.alignment(y_true, ccs_oh)
     identity_ccs = per_batch_identity(mv_ccs)
     return identity_ccs, identity_pred
@@ -449,8 +500,11 @@ class PerExampleAccuracy:
         self.total = 0
 
     def update_state(self, y_true, y_pred_scores):
-        y_true = torch.as_tensor(np.asarray(y_true)).long()
-        scores = torch.as_tensor(np.asarray(y_pred_scores))
+        y_true = (y_true if isinstance(y_true, torch.Tensor)
+                  else torch.as_tensor(np.asarray(y_true))).long()
+        scores = (y_pred_scores
+                  if isinstance(y_pred_scores, torch.Tensor)
+                  else torch.as_tensor(np.asarray(y_pred_scores)))
         y_true = left_shift_sequence(y_true)
         y_pred = left_shift_sequence(scores.argmax(-1).long())
         matches = (y_true == y_pred).sum(-1)
@@ -474,8 +528,12 @@ class PerClassAccuracy:
         self.total = 0
 
     def update_state(self, y_true, y_pred_scores):
-        y_true = torch.as_tensor(np.asarray(y_true)).long()
-        y_pred = torch.as_tensor(np.asarray(y_pred_scores)).argmax(-1)
+        y_true = (y_true if isinstance(y_true, torch.Tensor)
+                  else torch.as_tensor(np.asarray(y_true))).long()
+        y_pred = (y_pred_scores
+                  if isinstance(y_pred_scores, torch.Tensor)
+                  else torch.as_tensor(np.asarray(y_pred_scores))
+                  ).argmax(-1)
         mask = y_true == self.class_value
         self.correct += int(((y_true == y_pred) & mask).sum())
         self.total += int(mask.sum())

@@ -117,10 +117,12 @@ def train_model(
                 )
                 total_loss += float(loss)
                 n_batches += 1
-                acc.update_state(label.cpu(), probs.cpu())
+                # Metrics run on-device when the K14 HIP kernel is
+                # available (AlignmentMetric dispatches); CPU otherwise.
+                acc.update_state(label, probs)
                 ccs_rows = rows[:, 4 * params.max_passes, :]
                 ic, ip = losses_lib.get_batch_identity_ccs_pred(
-                    ccs_rows.cpu(), probs.cpu(), label.cpu(), align_metric
+                    ccs_rows, probs, label, align_metric
                 )
                 yield_metric.update_state(ic, ip)
         model.train()

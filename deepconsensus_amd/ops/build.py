@@ -22,6 +22,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "fused_ffn_glds.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_ffn_v3.hip"),
     os.path.join(_OPS_DIR, "hip", "embed_grad.hip"),
+    os.path.join(_OPS_DIR, "hip", "alignment_metric.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

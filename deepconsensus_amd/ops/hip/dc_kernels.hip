@@ -197,6 +197,9 @@ std::vector<at::Tensor> alignment_dp_bwd(at::Tensor grad_out,
                                          at::Tensor weights,
                                          at::Tensor seq_lens, int64_t m,
                                          int64_t n, int64_t width);
+std::vector<at::Tensor> alignment_metric_counts(
+    at::Tensor yt, at::Tensor yp, at::Tensor yt_len, at::Tensor yp_len,
+    double ms, double mp, double go, double ge);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
@@ -221,4 +224,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Alignment-loss wavefront DP forward (K13)");
   m.def("alignment_dp_bwd", &alignment_dp_bwd,
         "Alignment-loss wavefront DP backward (K13 VJP)");
+  m.def("alignment_metric_counts", &alignment_metric_counts,
+        "Device AlignmentMetric (K14): affine NW + backtrace counts");
 }

@@ -583,3 +583,66 @@ def test_ffn_knobs_agree(monkeypatch, env):
     agree = (b0.cpu() == b1.cpu()).float().mean()
     assert agree > 0.995, float(agree)  # bf16 path differences only
     assert (q0.cpu().float() - q1.cpu().float()).abs().mean() < 0.5
+
+
+def test_alignment_metric_device_matches_numpy():
+    """K14 device kernel (affine NW + backtrace counts) vs the numpy
+    oracle on random gapped sequences."""
+    from deepconsensus_amd.models import losses as losses_lib
+
+    rng = np.random.default_rng(17)
+    B, L = 48, 100
+    y_true = rng.integers(0, 5, size=(B, L)).astype(np.float32)
+    # Sprinkle extra gaps + a few fully/nearly empty rows.
+    y_true[rng.random((B, L)) < 0.2] = 0
+    y_true[0] = 0
+    y_true[1, 2:] = 0
+    logits = rng.normal(size=(B, L, 5)).astype(np.float32)
+    probs = torch.from_numpy(logits).softmax(-1)
+    # Make some rows agree with the label closely (high pid cases).
+    oh = torch.nn.functional.one_hot(
+        torch.from_numpy(y_true[:16].astype(np.int64)), 5
+    ).float()
+    probs[:16] = 0.9 * oh + 0.1 * probs[:16]
+
+    metric_cpu = losses_lib.AlignmentMetric()
+    v_cpu, _, mv_cpu = metric_cpu.alignment(
+        torch.from_numpy(y_true), probs
+    )
+    metric_gpu = losses_lib.AlignmentMetric()
+    v_gpu, paths, mv_gpu = metric_gpu.alignment(
+        torch.from_numpy(y_true).cuda(), probs.cuda()
+    )
+    assert paths is None  # device path returns counts only
+    np.testing.assert_allclose(v_gpu, v_cpu, rtol=1e-5, atol=1e-4)
+    for k in ("num_matches", "num_insertions", "num_deletions",
+              "num_correct_matches", "alignment_length"):
+        np.testing.assert_array_equal(mv_gpu[k], mv_cpu[k], err_msg=k)
+    np.testing.assert_allclose(mv_gpu["pid"], mv_cpu["pid"], rtol=1e-6)
+
+
+def test_alignment_metric_device_in_eval_path():
+    """get_batch_identity_ccs_pred on device tensors equals the CPU
+    path (the train-loop eval contract)."""
+    from deepconsensus_amd.models import losses as losses_lib
+
+    rng = np.random.default_rng(3)
+    B, L = 16, 100
+    label = torch.from_numpy(
+        rng.integers(0, 5, size=(B, L)).astype(np.float32)
+    )
+    ccs = torch.from_numpy(
+        rng.integers(0, 5, size=(B, L)).astype(np.float32)
+    )
+    probs = torch.from_numpy(
+        rng.normal(size=(B, L, 5)).astype(np.float32)
+    ).softmax(-1)
+    ic_cpu, ip_cpu = losses_lib.get_batch_identity_ccs_pred(
+        ccs, probs, label, losses_lib.AlignmentMetric()
+    )
+    ic_gpu, ip_gpu = losses_lib.get_batch_identity_ccs_pred(
+        ccs.cuda(), probs.cuda(), label.cuda(),
+        losses_lib.AlignmentMetric()
+    )
+    assert abs(ic_gpu - ic_cpu) < 1e-6
+    assert abs(ip_gpu - ip_cpu) < 1e-6
