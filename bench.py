@@ -81,7 +81,7 @@ def pipeline_main(args):
     zmws = args.pipeline_zmws
     with tempfile.TemporaryDirectory() as td:
         t0 = time.perf_counter()
-        sub, ccs = make_synth_bams(
+        sub, ccs, _ = make_synth_bams(
             td, zmws, args.pipeline_length, args.pipeline_subreads, 3
         )
         bam_lib.build_zmw_index(sub)

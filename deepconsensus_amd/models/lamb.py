@@ -31,11 +31,15 @@ class LAMB(torch.optim.Optimizer):
         eps: float = 1e-6,
         weight_decay: float = 0.0,
         capturable: bool = False,
+        foreach: bool = True,
     ):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
         self.capturable = capturable
+        # foreach=False: per-tensor loop (diagnostic — the hipGraph
+        # bisect isolates multi-tensor-apply as a capture suspect).
+        self.foreach = foreach
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -74,6 +78,33 @@ class LAMB(torch.optim.Optimizer):
                 state["step"] += 1
                 ms.append(state["exp_avg"])
                 vs.append(state["exp_avg_sq"])
+            if not self.foreach:
+                if self.capturable:
+                    group["step_t"] += 1
+                    t = group["step_t"]
+                    neg_lr = -group["lr_t"]
+                for idx, p in enumerate(ps):
+                    g_, m_, v_ = grads[idx], ms[idx], vs[idx]
+                    m_.mul_(beta1).add_(g_, alpha=1 - beta1)
+                    v_.mul_(beta2).addcmul_(g_, g_, value=1 - beta2)
+                    if self.capturable:
+                        bc1_ = 1.0 - beta1**t
+                        bc2_ = 1.0 - beta2**t
+                    else:
+                        bc1_ = 1.0 - beta1 ** self.state[p]["step"]
+                        bc2_ = 1.0 - beta2 ** self.state[p]["step"]
+                        neg_lr = -lr
+                    upd = (m_ / bc1_) / ((v_ / bc2_).sqrt() + eps)
+                    if wd != 0:
+                        upd = upd.add(p, alpha=wd)
+                    w_n = p.norm()
+                    u_n = upd.norm()
+                    ratio = torch.where(
+                        (w_n > 0) & (u_n > 0), w_n / u_n,
+                        torch.ones_like(w_n),
+                    ) * neg_lr
+                    p.add_(upd * ratio)
+                continue
             torch._foreach_mul_(ms, beta1)
             torch._foreach_add_(ms, grads, alpha=1 - beta1)
             torch._foreach_mul_(vs, beta2)

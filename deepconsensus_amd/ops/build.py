@@ -63,6 +63,35 @@ def build(verbose: bool = False):
     return module
 
 
+def build_asan(verbose: bool = False):
+    """Device-AddressSanitizer build (SURVEY section 5.2 sanitizer pass).
+
+    Compiles the same kernels for gfx950:xnack+ with -fsanitize=address
+    into _build_asan/dc_hip_kernels_asan.so. Run on a GPU box with
+    HSA_XNACK=1 and the host ASAN runtime preloaded:
+      LD_PRELOAD=$(hipcc -print-file-name=libclang_rt.asan-x86_64.so)
+      ASAN_OPTIONS=detect_leaks=0 HSA_XNACK=1 python ...
+    """
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950:xnack+")
+    build_dir = os.path.join(_OPS_DIR, "_build_asan")
+    os.makedirs(build_dir, exist_ok=True)
+    from torch.utils.cpp_extension import load
+
+    flags = ["-O2", "--offload-arch=gfx950:xnack+",
+             "-fsanitize=address", "-shared-libsan"]
+    module = load(
+        name=EXT_NAME + "_asan",
+        sources=_SRCS,
+        build_directory=build_dir,
+        extra_cuda_cflags=flags,
+        extra_ldflags=["-fsanitize=address", "-shared-libsan",
+                       "-L/opt/rocm/lib/llvm/lib/clang/22/lib/"
+                       "x86_64-unknown-linux-gnu"],
+        verbose=verbose,
+    )
+    return module
+
+
 def clean():
     if os.path.isdir(_BUILD_DIR):
         shutil.rmtree(_BUILD_DIR)

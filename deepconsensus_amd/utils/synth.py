@@ -10,7 +10,12 @@ import numpy as np
 from deepconsensus_amd.dcio import bam as bam_lib
 
 
-def make_synth_bams(out_dir, n_zmws, length, n_subreads, seed):
+def make_synth_bams(out_dir, n_zmws, length, n_subreads, seed,
+                    err_rate=0.005, n_ins=1, ccs_err_rate=0.0):
+    """err_rate: per-base substitution rate per subread; n_ins: short
+    insertions per subread; ccs_err_rate: substitution rate of the
+    draft CCS vs truth (a nonzero value makes the consensus task
+    require more than copying the CCS input row)."""
     rng = np.random.default_rng(seed)
     refs, zmw_seqs = [], {}
     for z in range(n_zmws):
@@ -26,16 +31,29 @@ def make_synth_bams(out_dir, n_zmws, length, n_subreads, seed):
             zm = int(name.split("/")[1])
             seq = zmw_seqs[name]
             for i in range(n_subreads):
-                # Mutate ~0.5% of bases and add one small insertion so the
-                # multi-read spacing has actual gap columns to create.
+                # Mutate err_rate of bases and add n_ins short
+                # insertions so the multi-read spacing has actual gap
+                # columns to create.
                 s = list(seq)
-                for p in rng.integers(0, ln, max(ln // 200, 1)):
+                n_mut = max(int(ln * err_rate), 1)
+                for p in rng.integers(0, ln, n_mut):
                     s[p] = rng.choice(list("ATCG"))
-                ins_pos = int(rng.integers(1, ln - 1))
-                ins_len = int(rng.integers(1, 4))
-                ins = "".join(rng.choice(list("ATCG"), size=ins_len))
-                full = "".join(s[:ins_pos]) + ins + "".join(s[ins_pos:])
-                cig = [(0, ins_pos), (1, ins_len), (0, ln - ins_pos)]
+                cuts = np.sort(rng.integers(1, ln - 1, size=n_ins))
+                full_parts, cig, prev = [], [], 0
+                for cpos in cuts:
+                    cpos = int(cpos)
+                    if cpos <= prev:
+                        continue
+                    ins_len = int(rng.integers(1, 4))
+                    ins = "".join(rng.choice(list("ATCG"), size=ins_len))
+                    full_parts.append("".join(s[prev:cpos]))
+                    cig.append((0, cpos - prev))
+                    full_parts.append(ins)
+                    cig.append((1, ins_len))
+                    prev = cpos
+                full_parts.append("".join(s[prev:]))
+                cig.append((0, ln - prev))
+                full = "".join(full_parts)
                 n = len(full)
                 w.write(bam_lib.BamRead(
                     qname=f"m000/{zm}/{i * (ln + 50)}_{i * (ln + 50) + n}",
@@ -54,13 +72,20 @@ def make_synth_bams(out_dir, n_zmws, length, n_subreads, seed):
     with bam_lib.BamWriter(ccs_path, header) as w:
         for rid, (name, ln) in enumerate(refs):
             zm = int(name.split("/")[1])
+            ccs_seq = zmw_seqs[name]
+            if ccs_err_rate > 0:
+                cs = list(ccs_seq)
+                for p in rng.integers(0, ln, max(int(ln * ccs_err_rate),
+                                                 1)):
+                    cs[p] = rng.choice(list("ATCG"))
+                ccs_seq = "".join(cs)
             w.write(bam_lib.BamRead(
                 qname=name, flag=4, ref_id=-1, pos=-1, cigartuples=[],
-                seq=zmw_seqs[name],
+                seq=ccs_seq,
                 query_qualities=rng.integers(20, 40, ln),
                 tags={"zm": zm, "ec": 11.5, "np": n_subreads, "rq": 0.998,
                       "RG": "rg0"},
             ))
-    return sub_path, ccs_path
+    return sub_path, ccs_path, zmw_seqs
 
 

@@ -45,7 +45,7 @@ def main():
 
     with tempfile.TemporaryDirectory() as td:
         t0 = time.perf_counter()
-        sub, ccs = make_synth_bams(td, args.zmws, args.length, args.subreads, 3)
+        sub, ccs, _ = make_synth_bams(td, args.zmws, args.length, args.subreads, 3)
         gen_s = time.perf_counter() - t0
         out = os.path.join(td, "out.fastq")
         options = qi.InferenceOptions(
