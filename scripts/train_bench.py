@@ -30,6 +30,11 @@ def main():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--bf16", action="store_true")
+    ap.add_argument("--graph", action="store_true",
+                    help="hipGraph-capture the whole train step (fwd + "
+                    "AlignmentLoss + bwd + capturable LAMB) and replay "
+                    "it; see scripts/graph_bisect.py for the capture "
+                    "validation ladder")
     args = ap.parse_args()
 
     rank, world = comm.init_distributed()
@@ -44,7 +49,9 @@ def main():
     torch.manual_seed(7)
     model = get_model(params).to(device)
     comm.broadcast_parameters(model)
-    optimizer, schedule = lamb_lib.create_optimizer(params, 10000, model)
+    optimizer, schedule = lamb_lib.create_optimizer(
+        params, 10000, model, capturable=args.graph
+    )
     reducer = comm.FlatGradAllreducer(model)
     loss_fn = losses_lib.AlignmentLoss(
         del_cost=params.del_cost, loss_reg=params.loss_reg, reduction="sum"
@@ -83,6 +90,31 @@ def main():
         schedule.apply(optimizer, i)
         return inner_step()
 
+    graph = None
+    if args.graph:
+        assert have_cuda and world == 1, (
+            "--graph currently captures the single-rank step"
+        )
+        # Warmup on a side stream, then capture the full step
+        # (grad-zeroing included) and replay it thereafter.
+        schedule.apply(optimizer, 0)
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                inner_step()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            inner_step()
+        torch.cuda.synchronize()
+
+        def step(i):  # noqa: F811
+            schedule.apply(optimizer, i)
+            graph.replay()
+            return None
+
     for i in range(args.warmup):
         step(i)
     if have_cuda:
@@ -100,7 +132,8 @@ def main():
         print(
             f"train step: {elapsed / args.steps * 1000:.1f} ms "
             f"({sps:.2f} steps/s, {sps * global_batch:.0f} examples/s, "
-            f"global batch {global_batch}, world {world})"
+            f"global batch {global_batch}, world {world}, "
+            f"graph={bool(graph)})"
         )
 
 
