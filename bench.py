@@ -210,8 +210,14 @@ def main():
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29571")
+        # DC_BENCH_BACKEND=gloo: rendezvous/reduction over gloo so a
+        # 1-GPU box can smoke-test the N>1 rank logic (RCCL refuses two
+        # ranks on one device: "Duplicate GPU detected", gpurun_out/
+        # rank0.log r2). The driver's 8-GPU run uses the default nccl.
         dist.init_process_group(
-            backend="nccl" if have_cuda else "gloo",
+            backend=os.environ.get(
+                "DC_BENCH_BACKEND", "nccl" if have_cuda else "gloo"
+            ),
             rank=rank,
             world_size=world,
             device_id=torch.device(device) if have_cuda else None,

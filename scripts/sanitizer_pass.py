@@ -38,35 +38,71 @@ def load_asan_ext():
 
 
 def main():
-    from deepconsensus_amd import ops as dc_ops
+    # Raw-kernel invocations only: no InferenceRunner / hipBLASLt in
+    # the process (library init under the interposed host-ASAN runtime
+    # SEGVs; the target here is OUR kernels, not the libraries).
+    ext = load_asan_ext()
+    print("ASAN extension loaded", flush=True)
+    torch.manual_seed(5)
+    B, L = 512, 100
+
+    # K2 embed_gather with a production-shaped chunk map.
     from deepconsensus_amd.models import config as cfg
     from deepconsensus_amd.models.model import get_model
-    from deepconsensus_amd.models import runner as runner_lib
-
-    ext = load_asan_ext()
-    # Patch the extension into the ops loader so InferenceRunner uses it.
-    dc_ops._ext = ext
-    print("ASAN extension loaded", flush=True)
+    from deepconsensus_amd.models.runner import build_gather_tables
 
     params = cfg.get_config("transformer_learn_values+custom")
     cfg.modify_params(params, is_training=False)
-    torch.manual_seed(5)
-    r = runner_lib.InferenceRunner(params, get_model(params),
-                                   device="cuda")
-    assert r.ext is ext
+    model = get_model(params)
+    tf, rs, rv, cc, ce = build_gather_tables(model)
     rng = np.random.default_rng(11)
-    B, L, mp = 512, params.max_length, params.max_passes
+    mp = params.max_passes
     rows = np.zeros((B, params.total_rows, L), np.float32)
     rows[:, 0:mp] = rng.integers(0, 5, size=(B, mp, L))
     rows[:, mp:3 * mp] = rng.integers(0, 60, size=(B, 2 * mp, L))
     rows[:, 3 * mp:4 * mp] = rng.integers(1, 3, size=(B, mp, L))
     rows[:, 4 * mp] = rng.integers(0, 5, size=(B, L))
     rows[:, -4:] = rng.uniform(3, 10, size=(B, 4, 1))
-    x = torch.from_numpy(rows.astype(np.int16))
-    bases, quals = r.forward_windows(x)  # embed/linear/attn/ffn/ln_head
+    emb = ext.embed_gather(
+        torch.from_numpy(rows).cuda().contiguous(), tf.cuda(), rs.cuda(),
+        rv.cuda(), cc.cuda(), ce.cuda(),
+    )
     torch.cuda.synchronize()
-    print("serving chain (K2,K3,K5-K12) OK:",
-          bases.shape, int(bases.sum()), flush=True)
+    print("embed_gather OK:", emb.shape, flush=True)
+
+    M = B * L
+    xb = (torch.randn(M, 280, device="cuda") * 0.3).to(torch.bfloat16)
+    wqkv = (torch.randn(896, 296, device="cuda") * 0.05).to(
+        torch.bfloat16)
+    empty = xb.new_empty(0)
+    qkv = ext.fused_linear(xb, wqkv, empty, empty, 840, False, 0.0)
+    torch.cuda.synchronize()
+    print("fused_linear OK:", qkv.shape, flush=True)
+
+    qkv3 = (torch.randn(B, L, 840, device="cuda") * 0.3).to(
+        torch.bfloat16)
+    attn = ext.banded_attn_mfma(qkv3, 2, 12, 140 ** -0.5)
+    torch.cuda.synchronize()
+    print("banded_attn_mfma OK:", attn.shape, flush=True)
+
+    w1 = (torch.randn(2048, 296, device="cuda") * 0.05).to(
+        torch.bfloat16)
+    w2 = (torch.randn(320, 2048, device="cuda") * 0.05).to(
+        torch.bfloat16)
+    b2 = torch.randn(320, device="cuda")
+    for name in ("fused_ffn_v3", "fused_ffn_v2"):
+        out = getattr(ext, name)(xb, w1, w2, b2, 0.5)
+        torch.cuda.synchronize()
+        print(name, "OK:", out.shape, flush=True)
+
+    g = torch.randn(280, device="cuda")
+    be = torch.randn(280, device="cuda")
+    wh = torch.randn(5, 280, device="cuda") * 0.05
+    bh = torch.randn(5, device="cuda")
+    lh = ext.fused_ln_head_qv(xb, g, be, wh, bh, 0.0, 1.2, -1.0, 93.0,
+                              False)
+    torch.cuda.synchronize()
+    print("fused_ln_head_qv OK:", lh[0].shape, flush=True)
 
     # K13 alignment DP fwd+bwd.
     Bd, m, n = 64, 100, 100
