@@ -96,15 +96,8 @@ __global__ __launch_bounds__(512, 1) void fused_ffn_v3_kernel(
         uint4 v = {};
         const bool rv = (m0 + row) < M;
         if (rv && 8 * q4 + 8 <= K1) {
-          // Non-temporal: x streams once per block; keeping it out of
-          // L2 preserves residency for the 2.5 MB weight image, which
-          // every block re-reads (the ablation probe measured the
-          // load skeleton at 53% of kernel time — weight refetch).
-          typedef unsigned u32x4 __attribute__((ext_vector_type(4)));
-          const u32x4 nv = __builtin_nontemporal_load(
-              reinterpret_cast<const u32x4*>(
-                  x + (size_t)(m0 + row) * K1 + 8 * q4));
-          v.x = nv.x; v.y = nv.y; v.z = nv.z; v.w = nv.w;
+          v = *reinterpret_cast<const uint4*>(
+              x + (size_t)(m0 + row) * K1 + 8 * q4);
         } else if (rv && 8 * q4 + 7 == BIAS_COL) {
           v.w = 0x3f800000u;  // upper half bf16(1.0) at col 287
         }
@@ -243,16 +236,8 @@ __global__ __launch_bounds__(512, 1) void fused_ffn_v3_kernel(
       const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * wave;
       if (m0 + row < M) {
         const size_t off = (size_t)(m0 + row) * K1 + col;
-        const unsigned short rb = __builtin_nontemporal_load(
-            reinterpret_cast<const unsigned short*>(x) + off);
-        bf16 rv16;
-        *reinterpret_cast<unsigned short*>(&rv16) = rb;
-        const float resid = __bfloat162float(rv16);
-        const bf16 o16 =
-            __float2bfloat16(resid + alpha * (oacc[ct][r] + bias));
-        __builtin_nontemporal_store(
-            *reinterpret_cast<const unsigned short*>(&o16),
-            reinterpret_cast<unsigned short*>(out) + off);
+        const float resid = __bfloat162float(x[off]);
+        out[off] = __float2bfloat16(resid + alpha * (oacc[ct][r] + bias));
       }
     }
   }

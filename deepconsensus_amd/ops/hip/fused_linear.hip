@@ -60,14 +60,8 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
       const int row = 64 * half + r;
       uint4 v = {};
       if (m0 + row < M && 8 * q4 + 8 <= K1) {
-        // Non-temporal: x/resid/out stream once per block; keep L2
-        // for the weight image every block re-reads (see the FFN
-        // ablation note in fused_ffn_v3.hip).
-        typedef unsigned u32x4 __attribute__((ext_vector_type(4)));
-        const u32x4 nv = __builtin_nontemporal_load(
-            reinterpret_cast<const u32x4*>(
-                x + (size_t)(m0 + row) * K1 + 8 * q4));
-        v.x = nv.x; v.y = nv.y; v.z = nv.z; v.w = nv.w;
+        v = *reinterpret_cast<const uint4*>(
+            x + (size_t)(m0 + row) * K1 + 8 * q4);
       }
       *reinterpret_cast<uint4*>(&w_lds[r * W_STRIDE + 8 * q4]) = v;
     }
@@ -143,9 +137,8 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
           &o_lds[row * O_STRIDE + 8 * c8]);
       if (RESIDUAL) {
-        const bf16x8 res = __builtin_nontemporal_load(
-            reinterpret_cast<const bf16x8*>(
-                resid + (size_t)(m0 + row) * K1 + col));
+        const bf16x8 res = *reinterpret_cast<const bf16x8*>(
+            resid + (size_t)(m0 + row) * K1 + col);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float rv = (float)res[j];
@@ -154,9 +147,8 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
         }
       }
       if (col + 8 <= N) {
-        __builtin_nontemporal_store(
-            v, reinterpret_cast<bf16x8*>(
-                   out + (size_t)(m0 + row) * N + col));
+        *reinterpret_cast<bf16x8*>(
+            out + (size_t)(m0 + row) * N + col) = v;
       } else {
         const unsigned short* vs =
             reinterpret_cast<const unsigned short*>(&v);
