@@ -63,6 +63,53 @@ def trim_insertions(
     return read
 
 
+def _spacing_ext():
+    global _SPACING
+    if _SPACING is None:
+        try:
+            from deepconsensus_amd.ops.build import build_spacing
+
+            _SPACING = build_spacing()
+        except Exception:  # pragma: no cover - no compiler at runtime
+            _SPACING = False
+    return _SPACING
+
+
+_SPACING = None
+
+
+def _expand_native(read: BamRead, ext) -> Read:
+    """C++ single-pass expand (inference path: no truth_range)."""
+    cig = np.asarray(read.cigartuples, dtype=np.int32).reshape(-1, 2)
+    pw_vals = np.minimum(np.asarray(read.get_tag("pw")), 255).astype(
+        np.uint8
+    )
+    ip_vals = np.minimum(np.asarray(read.get_tag("ip")), 255).astype(
+        np.uint8
+    )
+    bases_u32, cigar_u8, pw, ip, ccs_idx = ext.expand_read(
+        read.seq.encode("ascii"), cig, pw_vals, ip_vals,
+        int(read.pos) if read.pos and read.pos > 0 else 0,
+        bool(read.is_reverse), True,
+    )
+    strand = (
+        constants.Strand.REVERSE
+        if read.is_reverse
+        else constants.Strand.FORWARD
+    )
+    return Read(
+        name=read.qname,
+        bases=bases_u32.view("<U1"),
+        cigar=cigar_u8,
+        pw=pw,
+        ip=ip,
+        sn=np.array(read.get_tag("sn")),
+        strand=strand,
+        ccs_idx=ccs_idx,
+        truth_range=None,
+    )
+
+
 def expand_clip_indent(
     read: BamRead,
     truth_range: Union[Dict[str, Any], None] = None,
@@ -72,6 +119,10 @@ def expand_clip_indent(
     """Expands an alignment into CCS space (pre_lib.py:1128-1239)."""
     if ins_trim > 0:
         read = trim_insertions(read, ins_trim, counter)
+    if truth_range is None:
+        ext = _spacing_ext()
+        if ext:
+            return _expand_native(read, ext)
 
     read_idx, ccs_idx = read.aligned_index_arrays()
     aln_len = len(read_idx)

@@ -430,3 +430,61 @@ def test_dc_config_from_shape_invalid_raises():
     assert ok.max_passes == 20
     with pytest.raises(ValueError, match="Invalid subreads shape"):
         dc_config_from_shape((86, 100, 1), use_ccs_bq=False)
+
+
+def test_expand_native_matches_python(tmp_path):
+    """C++ expand_read must reproduce the python expand_clip_indent
+    field-for-field on reads with soft clips, indels, indent and
+    reverse strand."""
+    import numpy as np
+
+    from deepconsensus_amd.dcio.bam import BamRead
+    from deepconsensus_amd.preprocess import expand as expand_lib
+
+    ext = expand_lib._spacing_ext()
+    assert ext, "spacing extension must build in CI"
+
+    rng = np.random.default_rng(5)
+
+    def mk(cigar, pos=0, flag=0):
+        qlen = sum(n for op, n in cigar if op in (0, 1, 4, 7, 8))
+        seq = "".join(rng.choice(list("ATCG"), size=qlen))
+        return BamRead(
+            qname="q", flag=flag, ref_id=0, pos=pos, mapq=60,
+            cigartuples=cigar, seq=seq,
+            query_qualities=[30] * qlen,
+            tags={
+                "zm": 1,
+                "pw": rng.integers(0, 300, qlen),
+                "ip": rng.integers(0, 300, qlen),
+                "sn": np.array([6.0, 7.0, 5.5, 9.1], np.float32),
+            },
+        )
+
+    cases = [
+        ([(0, 50)], 0, 0),
+        ([(0, 30), (1, 3), (0, 10), (2, 4), (0, 20)], 0, 0),
+        ([(4, 7), (0, 40), (4, 5)], 0, 0),
+        ([(5, 9), (4, 3), (0, 25), (1, 2), (2, 2), (0, 10), (4, 6),
+          (5, 2)], 12, 0),
+        ([(0, 30), (2, 5), (0, 30)], 4, 16),          # reverse strand
+        ([(4, 4), (0, 20), (1, 5), (0, 15), (4, 8)], 9, 16),
+    ]
+    for cigar, pos, flag in cases:
+        read = mk(cigar, pos, flag)
+        nat = expand_lib._expand_native(read, ext)
+        # Force the python path by monkeying the ext lookup.
+        saved = expand_lib._SPACING
+        expand_lib._SPACING = False
+        try:
+            ref = expand_lib.expand_clip_indent(read)
+        finally:
+            expand_lib._SPACING = saved
+        np.testing.assert_array_equal(nat.bases, ref.bases,
+                                      err_msg=str(cigar))
+        np.testing.assert_array_equal(nat.cigar, ref.cigar)
+        np.testing.assert_array_equal(nat.pw, ref.pw)
+        np.testing.assert_array_equal(nat.ip, ref.ip)
+        np.testing.assert_array_equal(nat.ccs_idx, ref.ccs_idx)
+        np.testing.assert_array_equal(nat.sn, ref.sn)
+        assert nat.strand == ref.strand
