@@ -152,7 +152,10 @@ class DcExample:
     @property
     def ccs_width(self) -> int:
         if self._ccs_width is None:
-            self._ccs_width = len(str(self.ccs).rstrip())
+            # == len(str(self.ccs).rstrip()) without building the 10+ kb
+            # string: index of the last non-gap base + 1 (gaps are ' ').
+            nz = np.flatnonzero(self.ccs.bases != constants.GAP)
+            self._ccs_width = int(nz[-1]) + 1 if nz.size else 0
         return self._ccs_width
 
     @property
