@@ -92,6 +92,7 @@ def pipeline_main(args):
               f"{gen_s:.1f}s", file=sys.stderr)
 
         cpus = min(os.cpu_count() or 4, 16)
+        stage_s = {}
         if shards == 1:
             from deepconsensus_amd.inference import quick_inference as qi
 
@@ -109,6 +110,18 @@ def pipeline_main(args):
             )
             elapsed = time.perf_counter() - t0
             success = counter.success
+            stage_s = {}
+            try:
+                with open(os.path.join(td, "out.runtime.csv")) as f:
+                    next(f)
+                    for line in f:
+                        parts = line.strip().split(",")
+                        stage_s[parts[1]] = round(
+                            stage_s.get(parts[1], 0.0) + float(parts[2]),
+                            2,
+                        )
+            except OSError:
+                pass
         else:
             per_shard_cpus = max(cpus // shards, 1)
             procs = []
@@ -158,6 +171,7 @@ def pipeline_main(args):
                 "shards": shards,
                 "cpus": cpus,
                 "reads_written": success,
+                "stage_seconds": stage_s if shards == 1 else None,
             },
         }
         print(json.dumps(result))

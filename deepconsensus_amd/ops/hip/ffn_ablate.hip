@@ -3,6 +3,7 @@
 // skipped, st kept live via asm), 2 B2-only (pa fed from af, B1
 // skipped), 3 loads/loops/barriers only. Guide rule 17: skipped values
 // are kept live with empty asm so upstream work is not DCEd.
+// REGENERATED from the de-spilled v3 (r2).
 //
 // Same math as fused_ffn.hip (ffn_layer.py:69-87 + ReZero residual,
 // encoder_stack.py:88-92), restructured around two measured facts:
@@ -72,6 +73,22 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
+// Lane id recomputed at the call site (volatile: un-hoistable). The
+// allocator was spilling ~8 loop-invariant lane-derived LDS addresses
+// at 256 VGPRs, and every in-loop scratch reload carries a compiler
+// s_waitcnt vmcnt(0) that drains the in-flight weight DMA — the
+// ablation probe measured the result as ZERO transfer/compute overlap
+// (full == loads-only + mfma-only exactly). Two VALU per use beats a
+// scratch round trip + queue drain.
+__device__ __forceinline__ int lane_recompute() {
+  int l;
+  asm volatile(
+      "v_mbcnt_lo_u32_b32 %0, -1, 0\n\t"
+      "v_mbcnt_hi_u32_b32 %0, -1, %0"
+      : "=v"(l));
+  return l;
+}
+
 template <int MODE>
 __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w1,
@@ -81,7 +98,8 @@ __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;
+  // wave id lives in an SGPR (readfirstlane) — free to keep live.
+  const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
   const int c = lane & 31;
   const int hi = lane >> 5;
   const int m0 = blockIdx.x * BM;
@@ -161,6 +179,9 @@ __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
   f32x16 oacc[9] = {};
 
   for (int chunk = 0; chunk < NCHUNK; ++chunk) {
+    const int ln = lane_recompute();
+    const int c = ln & 31;   // shadow the entry values: loop-local,
+    const int hi = ln >> 5;  // dead at the backedge -> nothing to spill
     const int buf = chunk & 1;
     const bool more = chunk + 1 < NCHUNK;
     if (more) {
@@ -190,7 +211,6 @@ __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
         }
         __builtin_amdgcn_s_setprio(0);
       } else {
-        // keep the LDS image + af live without the MFMA chain
         asm volatile("" :: "v"(af[0]), "v"(w1buf[c]));
       }
       float st[16];
