@@ -24,6 +24,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "embed_grad.hip"),
     os.path.join(_OPS_DIR, "hip", "alignment_metric.hip"),
     os.path.join(_OPS_DIR, "hip", "ffn_ablate.hip"),
+    os.path.join(_OPS_DIR, "hip", "fused_ffn_v4.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

@@ -202,6 +202,8 @@ std::vector<at::Tensor> alignment_metric_counts(
     double ms, double mp, double go, double ge);
 at::Tensor ffn_ablate(at::Tensor x, at::Tensor w1, at::Tensor w2,
                       at::Tensor b2, double alpha, int64_t mode);
+at::Tensor fused_ffn_v4(at::Tensor x, at::Tensor w1, at::Tensor w2,
+                        at::Tensor b2, double alpha);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
@@ -228,6 +230,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Alignment-loss wavefront DP backward (K13 VJP)");
   m.def("alignment_metric_counts", &alignment_metric_counts,
         "Device AlignmentMetric (K14): affine NW + backtrace counts");
+  m.def("fused_ffn_v4", &fused_ffn_v4,
+        "Fused FFN v4: B1 on 16x16x32 MFMAs (4 independent chains)");
   m.def("ffn_ablate", &ffn_ablate,
         "fused_ffn_v3 ablation probe (0 full, 1 B1-only, 2 B2-only, "
         "3 loads-only) — perf diagnosis");
