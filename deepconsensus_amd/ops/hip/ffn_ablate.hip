@@ -178,15 +178,37 @@ __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
 
   f32x16 oacc[9] = {};
 
+  // MODE 6: rotate the chunk order per block so the XCD-level weight
+  // working set is the whole 2.5 MB steadily (L2-resident) instead of
+  // a synchronized-then-strangled walk.
+  const int rot = (MODE == 6) ? (int)(blockIdx.x % (unsigned)NCHUNK) : 0;
+  float spin = (float)threadIdx.x;
   for (int chunk = 0; chunk < NCHUNK; ++chunk) {
     const int ln = lane_recompute();
     const int c = ln & 31;   // shadow the entry values: loop-local,
     const int hi = ln >> 5;  // dead at the backedge -> nothing to spill
+    const int cr = (chunk + rot) % NCHUNK;       // this chunk's id
+    const int crn = (chunk + 1 + rot) % NCHUNK;  // next chunk's id
     const int buf = chunk & 1;
     const bool more = chunk + 1 < NCHUNK;
     if (more) {
-      issue_w1(chunk + 1, buf ^ 1);
-      issue_w2(chunk + 1, buf ^ 1);
+      issue_w1(crn, buf ^ 1);
+      issue_w2(crn, buf ^ 1);
+    }
+    if (MODE == 4) {
+      // loads + ~2.5 us of pure-VALU work (no LDS, no MFMA): if the
+      // DMA progresses during compute, MODE 4 ~= MODE 3; if it only
+      // progresses at the wait, MODE 4 ~= MODE 3 + spin time.
+#pragma unroll 4
+      for (int it = 0; it < 6000; ++it) {
+        spin = __builtin_fmaf(spin, 0.999f, 1.0f);
+      }
+      asm volatile("" :: "v"(spin));
+      if (more) {
+        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+      continue;
     }
     const bf16* w1buf = &smem[OFF_W1 + buf * W1_ELEMS];
     const bf16* w2buf = &smem[OFF_W2 + buf * W2_ELEMS];
@@ -200,7 +222,7 @@ __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
       // 18-deep dependent MFMA chain) costs 16 more live VGPRs and sent
       // the allocator from 20 to 316 spills — measured net loss.
       f32x16 acc = {};
-      if (MODE == 0 || MODE == 1) {
+      if (MODE == 0 || MODE == 1 || MODE == 6) {
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int s = 0; s < 18; ++s) {
@@ -231,7 +253,7 @@ __global__ __launch_bounds__(512, 1) void ffn_ablate_kernel(
                          (unsigned)ry[1]};
         pa[s] = *reinterpret_cast<const bf16x8*>(u);
       }
-      if (MODE == 0 || MODE == 2) {
+      if (MODE == 0 || MODE == 2 || MODE == 6) {
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int s = 0; s < 2; ++s) {
@@ -304,6 +326,12 @@ at::Tensor ffn_ablate(at::Tensor x, at::Tensor w1, at::Tensor w2,
   auto w2p = reinterpret_cast<bf16*>(w2.data_ptr());
   auto op = reinterpret_cast<bf16*>(out.data_ptr());
   switch (mode) {
+    case 6: hipLaunchKernelGGL(ffn_ablate_kernel<6>, grid, block, 0,
+                stream, xp, w1p, w2p, b2c.data_ptr<float>(), op, M,
+                (float)alpha); break;
+    case 4: hipLaunchKernelGGL(ffn_ablate_kernel<4>, grid, block, 0,
+                stream, xp, w1p, w2p, b2c.data_ptr<float>(), op, M,
+                (float)alpha); break;
     case 1: hipLaunchKernelGGL(ffn_ablate_kernel<1>, grid, block, 0,
                 stream, xp, w1p, w2p, b2c.data_ptr<float>(), op, M,
                 (float)alpha); break;
