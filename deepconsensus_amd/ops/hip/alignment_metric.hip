@@ -15,8 +15,10 @@
 // direction tensor [B][m+n+1][3][m+1] int8 in global.
 // Backtrace: one lane per example (divergent walk, <= m+n steps).
 
+#ifndef DC_SAN_MAIN
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#endif
 #include <hip/hip_runtime.h>
 
 namespace {
@@ -205,6 +207,8 @@ __global__ void metric_backtrace_kernel(
 
 }  // namespace
 
+#ifndef DC_SAN_MAIN
+
 std::vector<at::Tensor> alignment_metric_counts(
     at::Tensor yt, at::Tensor yp, at::Tensor yt_len, at::Tensor yp_len,
     double ms, double mp, double go, double ge) {
@@ -236,3 +240,5 @@ std::vector<at::Tensor> alignment_metric_counts(
                      counts.data_ptr<int>(), B, m, n);
   return {v_opt, counts};
 }
+
+#endif  // DC_SAN_MAIN

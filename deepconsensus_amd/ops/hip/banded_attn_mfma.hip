@@ -24,8 +24,10 @@
 // corresponding P entries are exactly 0, and the V slots they multiply are
 // zero-filled, so clamped/padded reads never contribute.
 
+#ifndef DC_SAN_MAIN
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#endif
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -296,6 +298,8 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
 
 }  // namespace
 
+#ifndef DC_SAN_MAIN
+
 at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
                             double scale) {
   TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == at::kBFloat16,
@@ -330,3 +334,5 @@ at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
   }
   return out;
 }
+
+#endif  // DC_SAN_MAIN

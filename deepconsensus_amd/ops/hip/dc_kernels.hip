@@ -12,8 +12,10 @@
 // Written CDNA4-first: 64-wide wavefronts, LDS staging for the id tile,
 // fp32 accumulation, no CUDA-compat shims.
 
+#ifndef DC_SAN_MAIN
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#endif
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -124,6 +126,8 @@ __global__ __launch_bounds__(256) void fused_ln_head_qv_kernel(
 }
 
 }  // namespace
+
+#ifndef DC_SAN_MAIN
 
 std::vector<at::Tensor> fused_ln_head_qv(
     at::Tensor x, at::Tensor gamma, at::Tensor beta, at::Tensor w_head,
@@ -236,3 +240,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused_ffn_v3 ablation probe (0 full, 1 B1-only, 2 B2-only, "
         "3 loads-only) — perf diagnosis");
 }
+
+#endif  // DC_SAN_MAIN

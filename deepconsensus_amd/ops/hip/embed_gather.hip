@@ -12,8 +12,10 @@
 // host-built map to 1..4 (row, table, width) entries; the id tile is staged
 // in LDS.
 
+#ifndef DC_SAN_MAIN
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#endif
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -102,6 +104,8 @@ __global__ __launch_bounds__(256) void embed_gather_kernel(
 
 }  // namespace
 
+#ifndef DC_SAN_MAIN
+
 at::Tensor embed_gather(
     at::Tensor rows, at::Tensor table_flat, at::Tensor row_shift,
     at::Tensor row_vocab, at::Tensor chunk_cnt, at::Tensor chunk_entries) {
@@ -140,3 +144,5 @@ at::Tensor embed_gather(
   }
   return out;
 }
+
+#endif  // DC_SAN_MAIN

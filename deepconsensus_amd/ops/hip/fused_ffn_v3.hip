@@ -22,8 +22,10 @@
 // LDS: 2x37,888 (W1) + 2x41,472 (W2) + 2 KiB dma-overflow scratch
 // = 160,768 B of the 163,840 B/CU — one barrier per chunk.
 
+#ifndef DC_SAN_MAIN
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#endif
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -265,6 +267,8 @@ __global__ __launch_bounds__(512, 1) void fused_ffn_v3_kernel(
 
 }  // namespace
 
+#ifndef DC_SAN_MAIN
+
 at::Tensor fused_ffn_v3(at::Tensor x, at::Tensor w1, at::Tensor w2,
                         at::Tensor b2, double alpha) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16,
@@ -293,3 +297,5 @@ at::Tensor fused_ffn_v3(at::Tensor x, at::Tensor w1, at::Tensor w2,
                      (float)alpha);
   return out;
 }
+
+#endif  // DC_SAN_MAIN

@@ -11,8 +11,10 @@
 // W arrives host-padded to [Npad, 296] (= the LDS row stride, so glds
 // streams chunks as raw row-major copies) with Npad a multiple of 64.
 
+#ifndef DC_SAN_MAIN
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#endif
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -170,6 +172,8 @@ __global__ __launch_bounds__(512, 2) void fused_linear_kernel(
 
 }  // namespace
 
+#ifndef DC_SAN_MAIN
+
 at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
                         at::Tensor resid, int64_t n_out, bool relu,
                         double alpha) {
@@ -223,3 +227,5 @@ at::Tensor fused_linear(at::Tensor x, at::Tensor w, at::Tensor bias,
   else launch(F{}, F{});
   return out;
 }
+
+#endif  // DC_SAN_MAIN

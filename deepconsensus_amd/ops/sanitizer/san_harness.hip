@@ -1,0 +1,252 @@
+// Standalone device-ASAN harness: exercises the
+// hand-written kernels with production-shaped inputs in a plain HIP
+// binary — no python, no libtorch — so the host+device
+// AddressSanitizer runtimes interpose cleanly (torch under host-ASAN
+// SEGVs in library init; see profiles/r02_sanitizer.md).
+//
+// Build (gfx950:xnack+, device+host ASAN) and run on a GPU box:
+//   hipcc --offload-arch=gfx950:xnack+ -fsanitize=address -shared-libsan \
+//     -O2 -std=c++17 deepconsensus_amd/ops/sanitizer/san_harness.hip \
+//     -o gpurun_out/san_harness
+//   HSA_XNACK=1 ASAN_OPTIONS=detect_leaks=0 ./gpurun_out/san_harness
+#define DC_SAN_MAIN 1
+
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <cstdint>
+#include <algorithm>
+#include <vector>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+// Each kernel TU is wrapped in its own namespace so per-file constants
+// (BM, K1, MAXW, ...) do not collide; their system #includes are no-ops
+// here (already included above, include guards).
+namespace san_eg {
+#include "../hip/embed_gather.hip"
+}
+namespace san_dp {
+#include "../hip/alignment_dp.hip"
+}
+namespace san_ffn {
+#include "../hip/fused_ffn_v3.hip"
+}
+namespace san_fl {
+#include "../hip/fused_linear.hip"
+}
+namespace san_ln {
+#include "../hip/dc_kernels.hip"
+}
+namespace san_attn {
+#include "../hip/banded_attn_mfma.hip"
+}
+using bf16 = __hip_bfloat16;
+
+#define CHECK(x)                                                        \
+  do {                                                                  \
+    hipError_t e_ = (x);                                                \
+    if (e_ != hipSuccess) {                                             \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(e_),          \
+             __FILE__, __LINE__);                                       \
+      exit(2);                                                          \
+    }                                                                   \
+  } while (0)
+
+static unsigned lcg_state = 12345;
+static unsigned lcg() { return lcg_state = lcg_state * 1664525u + 1013904223u; }
+static float frand() { return (float)(lcg() % 2000) / 1000.0f - 1.0f; }
+static unsigned short f2bf(float f) {
+  unsigned u;
+  memcpy(&u, &f, 4);
+  return (unsigned short)(u >> 16);
+}
+
+template <typename T>
+static T* dalloc(size_t n) {
+  T* p = nullptr;
+  CHECK(hipMalloc(&p, n * sizeof(T)));
+  return p;
+}
+
+template <typename T>
+static T* to_dev(const std::vector<T>& h) {
+  T* p = dalloc<T>(h.size());
+  CHECK(hipMemcpy(p, h.data(), h.size() * sizeof(T),
+                  hipMemcpyHostToDevice));
+  return p;
+}
+
+static void run_embed_gather() {
+  // 80 width-8 rows on a 256-vocab table + 4 width-2 (strand-like,
+  // vocab 3) + 1 width-8 -> 82 chunks incl. one 4-entry narrow chunk.
+  const int B = 256, R = 85, L = 100;
+  const int wide_rows = 80;
+  const int vocab_a = 256, vocab_s = 3;
+  std::vector<unsigned short> table;  // bf16 elems
+  const int base_a = 0;
+  for (int i = 0; i < vocab_a * 8; ++i) table.push_back(f2bf(frand()));
+  const int base_s = (int)table.size();
+  for (int i = 0; i < vocab_s * 2; ++i) table.push_back(f2bf(frand()));
+  while (table.size() % 8) table.push_back(0);
+  const int base_b = (int)table.size();
+  for (int i = 0; i < vocab_a * 8; ++i) table.push_back(f2bf(frand()));
+
+  std::vector<int> shift(R, 0), vocab(R);
+  std::vector<int> cnt;
+  std::vector<int> entries;  // int4 per entry slot
+  for (int r = 0; r < wide_rows; ++r) vocab[r] = vocab_a;
+  for (int r = wide_rows; r < wide_rows + 4; ++r) vocab[r] = vocab_s;
+  vocab[84] = vocab_a;
+  for (int c = 0; c < 80; ++c) {
+    cnt.push_back(1);
+    entries.insert(entries.end(), {c, base_a, 8, 0});
+    entries.insert(entries.end(), 12, 0);
+  }
+  cnt.push_back(4);
+  for (int k = 0; k < 4; ++k)
+    entries.insert(entries.end(), {80 + k, base_s, 2, 0});
+  cnt.push_back(1);
+  entries.insert(entries.end(), {84, base_b, 8, 0});
+  entries.insert(entries.end(), 12, 0);
+  const int nchunk = (int)cnt.size();
+
+  std::vector<short> rows((size_t)B * R * L);
+  for (auto& v : rows) v = (short)(lcg() % 3);
+  auto d_rows = to_dev(rows);
+  auto d_table = to_dev(table);
+  auto d_shift = to_dev(shift);
+  auto d_vocab = to_dev(vocab);
+  auto d_cnt = to_dev(cnt);
+  auto d_entries = to_dev(entries);
+  auto d_out = dalloc<unsigned short>((size_t)B * L * nchunk * 8);
+  dim3 grid(2048 < B * 2 ? 2048 : B * 2), block(256);
+  hipLaunchKernelGGL(san_eg::embed_gather_kernel<short>, grid, block, 0, 0,
+                     d_rows, (const bf16*)d_table, d_shift, d_vocab,
+                     d_cnt, (const int4*)d_entries, (bf16*)d_out, B, R,
+                     L, nchunk);
+  CHECK(hipDeviceSynchronize());
+  printf("embed_gather OK under ASAN\n");
+}
+
+static void run_alignment_dp() {
+  const int B = 64, m = 100, n = 100;
+  std::vector<float> subs((size_t)B * m * n), ins((size_t)B * n);
+  for (auto& v : subs) v = frand() + 1.5f;
+  for (auto& v : ins) v = frand() + 1.5f;
+  std::vector<int> lens(B);
+  for (auto& v : lens) v = 10 + (int)(lcg() % 90);
+  auto d_subs = to_dev(subs);
+  auto d_ins = to_dev(ins);
+  auto d_lens = to_dev(lens);
+  auto d_loss = dalloc<float>(B);
+  auto d_w = dalloc<float>((size_t)B * (m + 1) * (n + 1) * 3);
+  CHECK(hipMemset(d_w, 0, (size_t)B * (m + 1) * (n + 1) * 3 * 4));
+  hipLaunchKernelGGL(san_dp::alignment_fwd_kernel, dim3(B), dim3(128), 0, 0,
+                     d_subs, d_ins, d_lens, d_loss, d_w, B, m, n, 10.f,
+                     0.1f, 0);
+  CHECK(hipDeviceSynchronize());
+  auto d_go = dalloc<float>(B);
+  std::vector<float> ones(B, 1.f);
+  CHECK(hipMemcpy(d_go, ones.data(), B * 4, hipMemcpyHostToDevice));
+  auto d_gs = dalloc<float>((size_t)B * m * n);
+  auto d_gi = dalloc<float>((size_t)B * n);
+  CHECK(hipMemset(d_gs, 0, (size_t)B * m * n * 4));
+  CHECK(hipMemset(d_gi, 0, (size_t)B * n * 4));
+  hipLaunchKernelGGL(san_dp::alignment_bwd_kernel, dim3(B), dim3(128), 0, 0,
+                     d_go, d_w, d_lens, d_gs, d_gi, B, m, n, 0);
+  CHECK(hipDeviceSynchronize());
+  printf("alignment_dp fwd+bwd OK under ASAN\n");
+}
+
+static void run_ffn_v3() {
+  const int M = 4096;
+  std::vector<unsigned short> x((size_t)M * 280), w1(2048 * 296),
+      w2(320 * 2048);
+  for (auto& v : x) v = f2bf(frand() * 0.3f);
+  for (auto& v : w1) v = f2bf(frand() * 0.05f);
+  for (auto& v : w2) v = f2bf(frand() * 0.05f);
+  std::vector<float> b2(320);
+  for (auto& v : b2) v = frand();
+  auto d_x = to_dev(x);
+  auto d_w1 = to_dev(w1);
+  auto d_w2 = to_dev(w2);
+  auto d_b2 = to_dev(b2);
+  auto d_out = dalloc<unsigned short>((size_t)M * 280);
+  hipLaunchKernelGGL(san_ffn::fused_ffn_v3_kernel, dim3((M + 255) / 256),
+                     dim3(512), 0, 0, (const bf16*)d_x,
+                     (const bf16*)d_w1, (const bf16*)d_w2, d_b2,
+                     (bf16*)d_out, M, 0.5f);
+  CHECK(hipDeviceSynchronize());
+  printf("fused_ffn_v3 OK under ASAN\n");
+}
+
+static void run_fused_linear() {
+  const int M = 4096, N = 840, Npad = 896;
+  std::vector<unsigned short> x((size_t)M * 280), w((size_t)Npad * 296);
+  for (auto& v : x) v = f2bf(frand() * 0.3f);
+  for (auto& v : w) v = f2bf(frand() * 0.05f);
+  auto d_x = to_dev(x);
+  auto d_w = to_dev(w);
+  auto d_out = dalloc<unsigned short>((size_t)M * N);
+  hipLaunchKernelGGL((san_fl::fused_linear_kernel<false, false>),
+                     dim3((M + 127) / 128), dim3(512), 0, 0,
+                     (const bf16*)d_x, (const bf16*)d_w, nullptr,
+                     nullptr, (bf16*)d_out, M, N, Npad, 0.f);
+  CHECK(hipDeviceSynchronize());
+  printf("fused_linear OK under ASAN\n");
+}
+
+static void run_attn() {
+  const int B = 1024, L = 100, H = 2, D = 140;
+  std::vector<unsigned short> qkv((size_t)B * L * 3 * H * D);
+  for (auto& v : qkv) v = f2bf(frand() * 0.3f);
+  auto d_qkv = to_dev(qkv);
+  auto d_out = dalloc<unsigned short>((size_t)B * L * H * D);
+  const int BH = B * H;
+  hipLaunchKernelGGL(san_attn::banded_attn_mfma_kernel<true>,
+                     dim3(BH < 512 ? BH : 512), dim3(256), 0, 0,
+                     (const bf16*)d_qkv, (bf16*)d_out, B, L, H, 12,
+                     0.084515f);
+  CHECK(hipDeviceSynchronize());
+  printf("banded_attn_mfma OK under ASAN\n");
+}
+
+static void run_ln_head() {
+  const int N = 40960, H = 280;
+  std::vector<unsigned short> x((size_t)N * H);
+  for (auto& v : x) v = f2bf(frand());
+  std::vector<float> g(H), be(H), wh(5 * H), bh(5);
+  for (auto& v : g) v = frand();
+  for (auto& v : be) v = frand();
+  for (auto& v : wh) v = frand() * 0.05f;
+  for (auto& v : bh) v = frand();
+  auto d_x = to_dev(x);
+  auto d_g = to_dev(g);
+  auto d_be = to_dev(be);
+  auto d_wh = to_dev(wh);
+  auto d_bh = to_dev(bh);
+  auto d_bases = dalloc<unsigned char>(N);
+  auto d_quals = dalloc<unsigned char>(N);
+  hipLaunchKernelGGL(san_ln::fused_ln_head_qv_kernel<bf16>,
+                     dim3(2048 < (N + 3) / 4 ? 2048 : (N + 3) / 4),
+                     dim3(256), 0, 0, (const bf16*)d_x, d_g, d_be, d_wh,
+                     d_bh, d_bases, d_quals, (float*)nullptr, N, H, 0.f,
+                     1.2f, -1.f, 93.f);
+  CHECK(hipDeviceSynchronize());
+  printf("fused_ln_head_qv OK under ASAN\n");
+}
+
+int main() {
+  int n = 0;
+  CHECK(hipGetDeviceCount(&n));
+  printf("devices: %d\n", n);
+  run_embed_gather();
+  run_alignment_dp();
+  run_ffn_v3();
+  run_fused_linear();
+  run_attn();
+  run_ln_head();
+  printf("ALL KERNELS PASSED UNDER DEVICE ASAN\n");
+  return 0;
+}
