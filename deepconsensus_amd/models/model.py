@@ -31,6 +31,54 @@ from deepconsensus_amd.models.config import Params, get_indices
 from deepconsensus_amd.utils import constants
 
 
+_BATTN_AVAILABLE = None
+
+
+def _battn_train_available() -> bool:
+    global _BATTN_AVAILABLE
+    if _BATTN_AVAILABLE is None:
+        try:
+            from deepconsensus_amd import ops as dc_ops
+
+            ext = dc_ops.get_ext()
+            _BATTN_AVAILABLE = bool(
+                ext is not None and hasattr(ext, "banded_attn_train_fwd")
+            )
+        except Exception:  # pragma: no cover
+            _BATTN_AVAILABLE = False
+    return _BATTN_AVAILABLE
+
+
+class _BandedAttnTrain(torch.autograd.Function):
+    """HIP banded attention for the training path (K5-K7 on device):
+    band-only compute + fused softmax/dropout, band-local backward
+    (ops/hip/banded_attn_train.hip). Replaces the full [T,T] torch
+    chain when q/k/v are bf16 CUDA and the band fits the kernel."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, mask, win, p_drop):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        mask_t = mask if mask is not None else q.new_empty(0)
+        out, p = ext.banded_attn_train_fwd(q, k, v, mask_t, win, p_drop)
+        ctx.save_for_backward(q, k, v, p, mask_t)
+        ctx.win = win
+        ctx.p_drop = p_drop
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        q, k, v, p, mask_t = ctx.saved_tensors
+        dq, dk, dv = ext.banded_attn_train_bwd(
+            q, k, v, p, mask_t, dout, ctx.win, ctx.p_drop
+        )
+        return dq, dk, dv, None, None, None
+
+
 def sinusoidal_position_encoding(
     length: int,
     hidden_size: int,
@@ -123,6 +171,35 @@ class BandedSelfAttention(nn.Module):
         q = self.q_proj(x).view(b, t, h, d).transpose(1, 2)  # [B,H,T,D]
         k = self.k_proj(x).view(b, t, h, d).transpose(1, 2)
         v = self.v_proj(x).view(b, t, h, d).transpose(1, 2)
+        if (
+            training
+            and not need_weights
+            and self.attn_win_size is not None
+            and x.is_cuda
+            and q.dtype == torch.bfloat16
+            and d <= 144
+            and t <= 104
+            and 2 * self.attn_win_size + 1 <= 25
+            and _battn_train_available()
+        ):
+            # Fused banded path (band softmax + dropout + PV + band-local
+            # backward); the dropout band mask is drawn here so torch
+            # seeding controls it.
+            drop_mask = None
+            if self.dropout > 0:
+                drop_mask = (
+                    torch.rand(
+                        b, h, t, 2 * self.attn_win_size + 1,
+                        device=x.device,
+                    )
+                    >= self.dropout
+                )
+            ctx = _BandedAttnTrain.apply(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                drop_mask, self.attn_win_size, float(self.dropout),
+            )
+            ctx = ctx.transpose(1, 2).reshape(b, t, self.hidden_size)
+            return self.out_proj(ctx), None
         q = q * (d**-0.5)
         logits = torch.matmul(q, k.transpose(-1, -2))  # [B,H,T,T]
         mask = self.band_mask[:t, :t]

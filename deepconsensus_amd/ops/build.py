@@ -25,6 +25,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "alignment_metric.hip"),
     os.path.join(_OPS_DIR, "hip", "ffn_ablate.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_ffn_v4.hip"),
+    os.path.join(_OPS_DIR, "hip", "banded_attn_train.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

@@ -1,0 +1,293 @@
+// Training banded attention (fwd + bwd) for gfx950.
+//
+// The training path previously ran the torch chain (full [T,T] matmuls
+// + mask + softmax + dropout + PV, and autograd's mirror of all of it
+// backward) — 4x the band's FLOPs plus a long elementwise tail
+// (profiles/r01_train_top_kernels.txt). These kernels compute the
+// banded form directly, matching models/model.py SelfAttention
+// semantics exactly: S = (q * d^-0.5) @ k^T band-masked (-1e9 ==
+// exact-zero weights in fp32 softmax), P = softmax_fp32(S) cast bf16,
+// Pd = dropout(P) via a caller-supplied band mask, ctx = Pd @ v.
+//
+// Backward (band-local, fp32 accumulation):
+//   dPd = dO @ v^T;  dP = dPd * mask / (1-p)
+//   dS  = P * (dP - rowsum(dP * P))          [softmax VJP]
+//   dq  = scale * dS @ k;  dk = scale * dS^T @ q;  dv = Pd^T @ dO
+//
+// v1 is VALU-based (the band GEMMs are [100 x 25 x 140] — tiny per
+// item); one 256-thread block per (b, h), q/k/v(/dO) staged in LDS.
+// Layouts: q,k,v,ctx,dO [B,H,T,D] bf16 contiguous; P [B,H,T,W] bf16;
+// mask [B,H,T,W] uint8 (1 = keep).
+
+#ifndef DC_SAN_MAIN
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#endif
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+namespace {
+
+using bf16_t = __hip_bfloat16;
+
+constexpr int BT_MAXT = 104;   // max window length
+constexpr int BT_MAXW = 25;    // max band width (2*12+1)
+constexpr int BT_MAXD = 144;   // padded head dim (D = 140)
+
+__global__ __launch_bounds__(256) void battn_train_fwd_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const uint8_t* __restrict__ mask,
+    bf16_t* __restrict__ ctx, bf16_t* __restrict__ p_out,
+    int BH, int T, int D, int win, float scale, float keep_inv) {
+  __shared__ bf16_t qs[BT_MAXT][BT_MAXD];
+  __shared__ bf16_t ks[BT_MAXT][BT_MAXD];
+  __shared__ bf16_t vs[BT_MAXT][BT_MAXD];
+  __shared__ float ps[BT_MAXT][BT_MAXW];
+
+  const int item = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int W = 2 * win + 1;
+  const size_t base = (size_t)item * T * D;
+
+  for (int idx = tid; idx < T * D; idx += 256) {
+    const int t = idx / D, d = idx % D;
+    qs[t][d] = q[base + idx];
+    ks[t][d] = k[base + idx];
+    vs[t][d] = v[base + idx];
+  }
+  __syncthreads();
+
+  // S (scaled, fp32) into ps.
+  for (int cell = tid; cell < T * W; cell += 256) {
+    const int l = cell / W, w = cell % W;
+    const int kc = l - win + w;
+    float s = -1e30f;
+    if (kc >= 0 && kc < T) {
+      float acc = 0.f;
+      for (int d = 0; d < D; ++d) {
+        acc += __bfloat162float(qs[l][d]) * __bfloat162float(ks[kc][d]);
+      }
+      s = acc * scale;
+    }
+    ps[l][w] = s;
+  }
+  __syncthreads();
+
+  // Row softmax (fp32) -> bf16 P; one thread per row.
+  for (int l = tid; l < T; l += 256) {
+    float mx = -1e30f;
+    for (int w = 0; w < W; ++w) mx = fmaxf(mx, ps[l][w]);
+    float denom = 0.f;
+    for (int w = 0; w < W; ++w) {
+      const float e = (ps[l][w] <= -1e29f) ? 0.f : __expf(ps[l][w] - mx);
+      ps[l][w] = e;
+      denom += e;
+    }
+    const float inv = 1.f / denom;
+    for (int w = 0; w < W; ++w) {
+      // Match torch: softmax fp32 -> cast bf16 (P saved pre-dropout).
+      const bf16_t pb = __float2bfloat16(ps[l][w] * inv);
+      p_out[((size_t)item * T + l) * W + w] = pb;
+      // Pd in fp32 from the bf16 value (torch drops out the bf16 cast).
+      float pd = __bfloat162float(pb);
+      if (mask != nullptr) {
+        pd *= mask[((size_t)item * T + l) * W + w] ? keep_inv : 0.f;
+      }
+      ps[l][w] = pd;
+    }
+  }
+  __syncthreads();
+
+  // ctx = Pd @ v.
+  for (int cell = tid; cell < T * D; cell += 256) {
+    const int l = cell / D, d = cell % D;
+    float acc = 0.f;
+    const int w0 = max(0, win - l);
+    const int w1 = min(W, T + win - l);
+    for (int w = w0; w < w1; ++w) {
+      acc += ps[l][w] * __bfloat162float(vs[l - win + w][d]);
+    }
+    ctx[base + cell] = __float2bfloat16(acc);
+  }
+}
+
+__global__ __launch_bounds__(256) void battn_train_bwd_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ p_in,
+    const uint8_t* __restrict__ mask, const bf16_t* __restrict__ dout,
+    bf16_t* __restrict__ dq, bf16_t* __restrict__ dk,
+    bf16_t* __restrict__ dv,
+    int BH, int T, int D, int win, float scale, float keep_inv) {
+  __shared__ bf16_t qs[BT_MAXT][BT_MAXD];
+  __shared__ bf16_t ks[BT_MAXT][BT_MAXD];
+  __shared__ bf16_t vs[BT_MAXT][BT_MAXD];
+  __shared__ bf16_t dos[BT_MAXT][BT_MAXD];
+  __shared__ float dss[BT_MAXT][BT_MAXW];   // dS
+  __shared__ float pds[BT_MAXT][BT_MAXW];   // Pd
+
+  const int item = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int W = 2 * win + 1;
+  const size_t base = (size_t)item * T * D;
+
+  for (int idx = tid; idx < T * D; idx += 256) {
+    qs[idx / D][idx % D] = q[base + idx];
+    ks[idx / D][idx % D] = k[base + idx];
+    vs[idx / D][idx % D] = v[base + idx];
+    dos[idx / D][idx % D] = dout[base + idx];
+  }
+  __syncthreads();
+
+  // dPd = dO @ v^T (band) and Pd staged.
+  for (int cell = tid; cell < T * W; cell += 256) {
+    const int l = cell / W, w = cell % W;
+    const int kc = l - win + w;
+    float dpd = 0.f, pd = 0.f;
+    if (kc >= 0 && kc < T) {
+      float acc = 0.f;
+      for (int d = 0; d < D; ++d) {
+        acc += __bfloat162float(dos[l][d]) * __bfloat162float(vs[kc][d]);
+      }
+      dpd = acc;
+      pd = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
+      if (mask != nullptr) {
+        pd *= mask[((size_t)item * T + l) * W + w] ? keep_inv : 0.f;
+      }
+    }
+    pds[l][w] = pd;
+    dss[l][w] = dpd;  // holds dPd for now
+  }
+  __syncthreads();
+
+  // dS = P * (dP - rowsum(dP * P)); one thread per row.
+  for (int l = tid; l < T; l += 256) {
+    float dp_row[BT_MAXW];
+    float r = 0.f;
+    for (int w = 0; w < W; ++w) {
+      const int kc = l - win + w;
+      float p_ = 0.f, dp_ = 0.f;
+      if (kc >= 0 && kc < T) {
+        p_ = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
+        dp_ = dss[l][w];
+        if (mask != nullptr) {
+          dp_ *= mask[((size_t)item * T + l) * W + w] ? keep_inv : 0.f;
+        }
+      }
+      dp_row[w] = dp_;
+      r += dp_ * p_;
+    }
+    for (int w = 0; w < W; ++w) {
+      const int kc = l - win + w;
+      float p_ = 0.f;
+      if (kc >= 0 && kc < T) {
+        p_ = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
+      }
+      dss[l][w] = p_ * (dp_row[w] - r);
+    }
+  }
+  __syncthreads();
+
+  // dq[l,d] = scale * sum_w dS[l,w] * k[kc,d]
+  // dk[kc,d] = scale * sum_l dS[l,w(l,kc)] * q[l,d]   (w = kc - l + win)
+  // dv[kc,d] = sum_l Pd[l,w] * dO[l,d]
+  for (int cell = tid; cell < T * D; cell += 256) {
+    const int l = cell / D, d = cell % D;
+    float acc_q = 0.f;
+    const int w0 = max(0, win - l);
+    const int w1 = min(W, T + win - l);
+    for (int w = w0; w < w1; ++w) {
+      acc_q += dss[l][w] * __bfloat162float(ks[l - win + w][d]);
+    }
+    dq[base + cell] = __float2bfloat16(acc_q * scale);
+
+    // For column kc = l: contributing rows l2 with |l2 - kc| <= win.
+    const int kc = l;
+    float acc_k = 0.f, acc_v = 0.f;
+    const int l2_0 = max(0, kc - win);
+    const int l2_1 = min(T, kc + win + 1);
+    for (int l2 = l2_0; l2 < l2_1; ++l2) {
+      const int w = kc - l2 + win;
+      acc_k += dss[l2][w] * __bfloat162float(qs[l2][d]);
+      acc_v += pds[l2][w] * __bfloat162float(dos[l2][d]);
+    }
+    dk[base + cell] = __float2bfloat16(acc_k * scale);
+    dv[base + cell] = __float2bfloat16(acc_v);
+  }
+}
+
+}  // namespace
+
+#ifndef DC_SAN_MAIN
+
+std::vector<at::Tensor> banded_attn_train_fwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor mask,
+    int64_t win, double p_drop) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16, "q bf16 cuda");
+  auto qc = q.contiguous();
+  auto kc_ = k.contiguous();
+  auto vc = v.contiguous();
+  const int B = qc.size(0), H = qc.size(1), T = qc.size(2),
+            D = qc.size(3);
+  const int W = 2 * (int)win + 1;
+  TORCH_CHECK(T <= BT_MAXT && D <= BT_MAXD && W <= BT_MAXW,
+              "shape exceeds kernel limits");
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  at::Tensor mc;
+  const uint8_t* mptr = nullptr;
+  if (has_mask) {
+    mc = mask.contiguous();
+    TORCH_CHECK(mc.dtype() == at::kBool || mc.dtype() == at::kByte,
+                "mask bool/byte");
+    mptr = (const uint8_t*)mc.data_ptr();
+  }
+  auto ctx = at::empty_like(qc);
+  auto p = at::empty({B, H, T, W}, qc.options());
+  const float keep_inv =
+      p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 1.0f;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(battn_train_fwd_kernel, dim3(B * H), dim3(256), 0,
+                     stream, (const bf16_t*)qc.data_ptr(),
+                     (const bf16_t*)kc_.data_ptr(),
+                     (const bf16_t*)vc.data_ptr(), mptr,
+                     (bf16_t*)ctx.data_ptr(), (bf16_t*)p.data_ptr(),
+                     B * H, T, D, (int)win,
+                     (float)(1.0 / std::sqrt((double)D)), keep_inv);
+  return {ctx, p};
+}
+
+std::vector<at::Tensor> banded_attn_train_bwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor p,
+    at::Tensor mask, at::Tensor dout, int64_t win, double p_drop) {
+  auto qc = q.contiguous();
+  auto kc_ = k.contiguous();
+  auto vc = v.contiguous();
+  auto pc = p.contiguous();
+  auto dc = dout.contiguous();
+  const int B = qc.size(0), H = qc.size(1), T = qc.size(2),
+            D = qc.size(3);
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  at::Tensor mc;
+  const uint8_t* mptr = nullptr;
+  if (has_mask) {
+    mc = mask.contiguous();
+    mptr = (const uint8_t*)mc.data_ptr();
+  }
+  auto dq = at::empty_like(qc);
+  auto dk = at::empty_like(qc);
+  auto dv = at::empty_like(qc);
+  const float keep_inv =
+      p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 1.0f;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(battn_train_bwd_kernel, dim3(B * H), dim3(256), 0,
+                     stream, (const bf16_t*)qc.data_ptr(),
+                     (const bf16_t*)kc_.data_ptr(),
+                     (const bf16_t*)vc.data_ptr(),
+                     (const bf16_t*)pc.data_ptr(), mptr,
+                     (const bf16_t*)dc.data_ptr(),
+                     (bf16_t*)dq.data_ptr(), (bf16_t*)dk.data_ptr(),
+                     (bf16_t*)dv.data_ptr(), B * H, T, D, (int)win,
+                     (float)(1.0 / std::sqrt((double)D)), keep_inv);
+  return {dq, dk, dv};
+}
+
+#endif  // DC_SAN_MAIN

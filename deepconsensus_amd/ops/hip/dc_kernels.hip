@@ -208,6 +208,12 @@ at::Tensor ffn_ablate(at::Tensor x, at::Tensor w1, at::Tensor w2,
                       at::Tensor b2, double alpha, int64_t mode);
 at::Tensor fused_ffn_v4(at::Tensor x, at::Tensor w1, at::Tensor w2,
                         at::Tensor b2, double alpha);
+std::vector<at::Tensor> banded_attn_train_fwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor mask,
+    int64_t win, double p_drop);
+std::vector<at::Tensor> banded_attn_train_bwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor p,
+    at::Tensor mask, at::Tensor dout, int64_t win, double p_drop);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
@@ -234,6 +240,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Alignment-loss wavefront DP backward (K13 VJP)");
   m.def("alignment_metric_counts", &alignment_metric_counts,
         "Device AlignmentMetric (K14): affine NW + backtrace counts");
+  m.def("banded_attn_train_fwd", &banded_attn_train_fwd,
+        "Training banded attention forward (saves band softmax P)");
+  m.def("banded_attn_train_bwd", &banded_attn_train_bwd,
+        "Training banded attention backward (band-local dS -> dq,dk,dv)");
   m.def("fused_ffn_v4", &fused_ffn_v4,
         "Fused FFN v4: B1 on 16x16x32 MFMAs (4 independent chains)");
   m.def("ffn_ablate", &ffn_ablate,
