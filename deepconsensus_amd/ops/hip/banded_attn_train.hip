@@ -29,6 +29,19 @@
 namespace {
 
 using bf16_t = __hip_bfloat16;
+typedef __bf16 btx8 __attribute__((ext_vector_type(8)));
+
+// 8-wide bf16 dot-product step (G13: never scalar bf16 LDS reads).
+__device__ __forceinline__ float dot8(const bf16_t* a, const bf16_t* b) {
+  const btx8 av = *reinterpret_cast<const btx8*>(a);
+  const btx8 bv = *reinterpret_cast<const btx8*>(b);
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    acc += (float)av[j] * (float)bv[j];
+  }
+  return acc;
+}
 
 constexpr int BT_MAXT = 104;   // max window length
 constexpr int BT_MAXW = 25;    // max band width (2*12+1)
@@ -49,11 +62,13 @@ __global__ __launch_bounds__(256) void battn_train_fwd_kernel(
   const int W = 2 * win + 1;
   const size_t base = (size_t)item * T * D;
 
-  for (int idx = tid; idx < T * D; idx += 256) {
-    const int t = idx / D, d = idx % D;
-    qs[t][d] = q[base + idx];
-    ks[t][d] = k[base + idx];
-    vs[t][d] = v[base + idx];
+  for (int idx = tid; idx < T * BT_MAXD; idx += 256) {
+    const int t = idx / BT_MAXD, d = idx % BT_MAXD;
+    const bf16_t z = __float2bfloat16(0.f);
+    const bool in = d < D;
+    qs[t][d] = in ? q[base + (size_t)t * D + d] : z;
+    ks[t][d] = in ? k[base + (size_t)t * D + d] : z;
+    vs[t][d] = in ? v[base + (size_t)t * D + d] : z;
   }
   __syncthreads();
 
@@ -64,8 +79,8 @@ __global__ __launch_bounds__(256) void battn_train_fwd_kernel(
     float s = -1e30f;
     if (kc >= 0 && kc < T) {
       float acc = 0.f;
-      for (int d = 0; d < D; ++d) {
-        acc += __bfloat162float(qs[l][d]) * __bfloat162float(ks[kc][d]);
+      for (int d = 0; d < BT_MAXD; d += 8) {
+        acc += dot8(&qs[l][d], &ks[kc][d]);
       }
       s = acc * scale;
     }
@@ -98,16 +113,27 @@ __global__ __launch_bounds__(256) void battn_train_fwd_kernel(
   }
   __syncthreads();
 
-  // ctx = Pd @ v.
-  for (int cell = tid; cell < T * D; cell += 256) {
-    const int l = cell / D, d = cell % D;
-    float acc = 0.f;
+  // ctx = Pd @ v: each thread an 8-wide d-granule of one row.
+  const int ng = T * (BT_MAXD / 8);
+  for (int g = tid; g < ng; g += 256) {
+    const int l = g / (BT_MAXD / 8), d0 = (g % (BT_MAXD / 8)) * 8;
+    if (d0 >= D) continue;
+    float acc[8] = {};
     const int w0 = max(0, win - l);
     const int w1 = min(W, T + win - l);
     for (int w = w0; w < w1; ++w) {
-      acc += ps[l][w] * __bfloat162float(vs[l - win + w][d]);
+      const float pw = ps[l][w];
+      const btx8 vv = *reinterpret_cast<const btx8*>(
+          &vs[l - win + w][d0]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += pw * (float)vv[j];
     }
-    ctx[base + cell] = __float2bfloat16(acc);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (d0 + j < D) {
+        ctx[base + (size_t)l * D + d0 + j] = __float2bfloat16(acc[j]);
+      }
+    }
   }
 }
 
@@ -130,11 +156,14 @@ __global__ __launch_bounds__(256) void battn_train_bwd_kernel(
   const int W = 2 * win + 1;
   const size_t base = (size_t)item * T * D;
 
-  for (int idx = tid; idx < T * D; idx += 256) {
-    qs[idx / D][idx % D] = q[base + idx];
-    ks[idx / D][idx % D] = k[base + idx];
-    vs[idx / D][idx % D] = v[base + idx];
-    dos[idx / D][idx % D] = dout[base + idx];
+  for (int idx = tid; idx < T * BT_MAXD; idx += 256) {
+    const int t = idx / BT_MAXD, d = idx % BT_MAXD;
+    const bf16_t z = __float2bfloat16(0.f);
+    const bool in = d < D;
+    qs[t][d] = in ? q[base + (size_t)t * D + d] : z;
+    ks[t][d] = in ? k[base + (size_t)t * D + d] : z;
+    vs[t][d] = in ? v[base + (size_t)t * D + d] : z;
+    dos[t][d] = in ? dout[base + (size_t)t * D + d] : z;
   }
   __syncthreads();
 
@@ -145,8 +174,8 @@ __global__ __launch_bounds__(256) void battn_train_bwd_kernel(
     float dpd = 0.f, pd = 0.f;
     if (kc >= 0 && kc < T) {
       float acc = 0.f;
-      for (int d = 0; d < D; ++d) {
-        acc += __bfloat162float(dos[l][d]) * __bfloat162float(vs[kc][d]);
+      for (int d = 0; d < BT_MAXD; d += 8) {
+        acc += dot8(&dos[l][d], &vs[kc][d]);
       }
       dpd = acc;
       pd = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
@@ -190,28 +219,45 @@ __global__ __launch_bounds__(256) void battn_train_bwd_kernel(
   // dq[l,d] = scale * sum_w dS[l,w] * k[kc,d]
   // dk[kc,d] = scale * sum_l dS[l,w(l,kc)] * q[l,d]   (w = kc - l + win)
   // dv[kc,d] = sum_l Pd[l,w] * dO[l,d]
-  for (int cell = tid; cell < T * D; cell += 256) {
-    const int l = cell / D, d = cell % D;
-    float acc_q = 0.f;
+  const int ng = T * (BT_MAXD / 8);
+  for (int g = tid; g < ng; g += 256) {
+    const int l = g / (BT_MAXD / 8), d0 = (g % (BT_MAXD / 8)) * 8;
+    if (d0 >= D) continue;
+    float aq[8] = {}, ak[8] = {}, av[8] = {};
     const int w0 = max(0, win - l);
     const int w1 = min(W, T + win - l);
     for (int w = w0; w < w1; ++w) {
-      acc_q += dss[l][w] * __bfloat162float(ks[l - win + w][d]);
+      const float dsw = dss[l][w];
+      const btx8 kv = *reinterpret_cast<const btx8*>(
+          &ks[l - win + w][d0]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) aq[j] += dsw * (float)kv[j];
     }
-    dq[base + cell] = __float2bfloat16(acc_q * scale);
-
-    // For column kc = l: contributing rows l2 with |l2 - kc| <= win.
+    // Column kc = l: rows l2 within the band of kc.
     const int kc = l;
-    float acc_k = 0.f, acc_v = 0.f;
     const int l2_0 = max(0, kc - win);
     const int l2_1 = min(T, kc + win + 1);
     for (int l2 = l2_0; l2 < l2_1; ++l2) {
       const int w = kc - l2 + win;
-      acc_k += dss[l2][w] * __bfloat162float(qs[l2][d]);
-      acc_v += pds[l2][w] * __bfloat162float(dos[l2][d]);
+      const float dsw = dss[l2][w];
+      const float pdw = pds[l2][w];
+      const btx8 qv = *reinterpret_cast<const btx8*>(&qs[l2][d0]);
+      const btx8 dov = *reinterpret_cast<const btx8*>(&dos[l2][d0]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        ak[j] += dsw * (float)qv[j];
+        av[j] += pdw * (float)dov[j];
+      }
     }
-    dk[base + cell] = __float2bfloat16(acc_k * scale);
-    dv[base + cell] = __float2bfloat16(acc_v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (d0 + j < D) {
+        const size_t off = base + (size_t)l * D + d0 + j;
+        dq[off] = __float2bfloat16(aq[j] * scale);
+        dk[off] = __float2bfloat16(ak[j] * scale);
+        dv[off] = __float2bfloat16(av[j]);
+      }
+    }
   }
 }
 

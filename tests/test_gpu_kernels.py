@@ -733,7 +733,7 @@ def test_banded_attn_train_dropout_mask_semantics():
 def test_model_attention_fused_training_path():
     """SelfAttention.forward(training=True) routes through the fused op
     on GPU bf16 and its grads match the torch path (dropout 0)."""
-    from deepconsensus_amd.models.model import SelfAttention
+    from deepconsensus_amd.models.model import BandedSelfAttention as SelfAttention
 
     torch.manual_seed(5)
     attn = SelfAttention(280, 2, 0.0, 12, 100).cuda()
