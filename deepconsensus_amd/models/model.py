@@ -35,8 +35,19 @@ _BATTN_AVAILABLE = None
 
 
 def _battn_train_available() -> bool:
+    """Fused training attention is OPT-IN (DC_ATTN_TRAIN=1): measured
+    0.62x the torch chain at batch 4096 (2,059 us fwd + 4,655 us bwd vs
+    4,188 us torch fwd+bwd; gpurun_out/r2_attnb.log) — the band-only
+    VALU kernels are occupancy-bound at 1 block/CU and lose to
+    hipBLASLt's full-matrix path. Kept for parity/iteration; see
+    profiles/r02_perf_journal.md."""
     global _BATTN_AVAILABLE
     if _BATTN_AVAILABLE is None:
+        import os
+
+        if os.environ.get("DC_ATTN_TRAIN") != "1":
+            _BATTN_AVAILABLE = False
+            return False
         try:
             from deepconsensus_amd import ops as dc_ops
 

@@ -730,11 +730,15 @@ def test_banded_attn_train_dropout_mask_semantics():
     assert err < 0.05, f"dropout fwd max err {err}"
 
 
-def test_model_attention_fused_training_path():
+def test_model_attention_fused_training_path(monkeypatch):
     """SelfAttention.forward(training=True) routes through the fused op
-    on GPU bf16 and its grads match the torch path (dropout 0)."""
+    (DC_ATTN_TRAIN=1) on GPU bf16 and its grads match the torch path
+    (dropout 0)."""
+    import deepconsensus_amd.models.model as model_mod
     from deepconsensus_amd.models.model import BandedSelfAttention as SelfAttention
 
+    monkeypatch.setenv("DC_ATTN_TRAIN", "1")
+    monkeypatch.setattr(model_mod, "_BATTN_AVAILABLE", None)
     torch.manual_seed(5)
     attn = SelfAttention(280, 2, 0.0, 12, 100).cuda()
     x = torch.randn(16, 100, 280, device="cuda")
