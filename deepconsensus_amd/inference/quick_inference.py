@@ -353,6 +353,33 @@ class Timelog:
                 f.write(",".join(str(r[c]) for c in cols) + "\n")
 
 
+def _stitch_batch_to_text(args):
+    """Process-pool stitcher: sorts + stitches a batch of predictions
+    into FASTQ text, returning (text, outcome_counts). Runs in the
+    worker pool so the GIL-heavy stitching never blocks the serial
+    model loop or the prefetch threads."""
+    predictions, max_length, min_quality, min_length = args
+    counter = stitch_utils.OutcomeCounter()
+    parts = []
+    predictions = sorted(
+        predictions, key=lambda dc: (dc.molecule_name, dc.window_pos)
+    )
+    for zmw, preds in itertools.groupby(
+        predictions, lambda p: p.molecule_name
+    ):
+        fastq_string = stitch_utils.stitch_to_fastq(
+            molecule_name=zmw,
+            predictions=list(preds),
+            max_length=max_length,
+            min_quality=min_quality,
+            min_length=min_length,
+            outcome_counter=counter,
+        )
+        if fastq_string:
+            parts.append(fastq_string)
+    return "".join(parts), dataclasses.asdict(counter)
+
+
 def _write_outputs(
     predictions: List[stitch_utils.DCModelOutput],
     output_writer,
@@ -541,21 +568,37 @@ def run(
         if options.end_after_stage == DebugStage.RUN_MODEL:
             return
 
-        def write_task(preds=preds, batch_name=batch_name,
-                       n_examples=n_examples, n_subreads=n_subreads,
-                       n_inputs=len(inputs)):
-            before = time.time()
+        before = time.time()
+        if pool is not None and bam_out is None:
+            # Stitch in the PROCESS pool (no GIL contention with the
+            # prefetch threads), write the returned text on the ordered
+            # writer thread. BAM output keeps the in-process path (the
+            # writer needs record objects).
+            fut = pool.submit(
+                _stitch_batch_to_text,
+                (preds, options.max_length, options.min_quality,
+                 options.min_length),
+            )
+
+            def write_task(fut=fut, batch_name=batch_name,
+                           n_examples=n_examples,
+                           n_subreads=n_subreads, n_inputs=len(inputs),
+                           before=before):
+                text, counts = fut.result()
+                for k, v in counts.items():
+                    setattr(outcome_counter, k,
+                            getattr(outcome_counter, k) + v)
+                if text:
+                    output_writer.write(text)
+                timelog.add("stitch_and_write_fastq", batch_name,
+                            before, n_examples, n_subreads, n_inputs)
+
+            writer_futs.append(writer_pool.submit(write_task))
+        else:
             _write_outputs(preds, output_writer, bam_out, options,
                            outcome_counter)
             timelog.add("stitch_and_write_fastq", batch_name, before,
-                        n_examples, n_subreads, n_inputs)
-
-        # Stitch + write overlap the next batch's model execution on a
-        # single ordered writer thread (the reference serializes
-        # preprocess -> predict -> stitch per batch; SURVEY section 7
-        # hard part 4 calls out the pipelined design as the throughput
-        # lever the reference never built).
-        writer_pool.submit(write_task)
+                        n_examples, n_subreads, len(inputs))
 
     # Pipelined loop: up to `prefetch_batches` ZMW batches preprocess in
     # the worker pool while the current batch runs the model.
@@ -565,6 +608,7 @@ def run(
     n_fed = 0
     n_batches = 0
     writer_pool = concurrent.futures.ThreadPoolExecutor(1)
+    writer_futs: List[concurrent.futures.Future] = []
     with concurrent.futures.ThreadPoolExecutor(lookahead) as prefetcher:
 
         def feed_one() -> bool:
@@ -613,6 +657,8 @@ def run(
                 pass
 
     writer_pool.shutdown(wait=True)
+    for f in writer_futs:
+        f.result()  # surface stitch/write errors instead of truncating
     if pool is not None:
         pool.shutdown()
     if bam_out is not None:
