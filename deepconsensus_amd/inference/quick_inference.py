@@ -577,7 +577,9 @@ def run(
             return
 
         before = time.time()
-        if stitch_pool is not None and bam_out is None:
+        stitch_mode = os.environ.get("DC_STITCH_MODE", "pool")
+        if (stitch_mode == "pool" and stitch_pool is not None
+                and bam_out is None):
             # Stitch in the PROCESS pool (no GIL contention with the
             # prefetch threads), write the returned text on the ordered
             # writer thread. BAM output keeps the in-process path (the
