@@ -238,14 +238,23 @@ static void run_ln_head() {
 }
 
 int main() {
+  setvbuf(stdout, nullptr, _IONBF, 0);
+  setvbuf(stderr, nullptr, _IONBF, 0);
+  fprintf(stderr, "[san] start\n");
   int n = 0;
   CHECK(hipGetDeviceCount(&n));
   printf("devices: %d\n", n);
+  fprintf(stderr, "[san] run_embed_gather\n");
   run_embed_gather();
+  fprintf(stderr, "[san] run_alignment_dp\n");
   run_alignment_dp();
+  fprintf(stderr, "[san] run_ffn_v3\n");
   run_ffn_v3();
+  fprintf(stderr, "[san] run_fused_linear\n");
   run_fused_linear();
+  fprintf(stderr, "[san] run_attn\n");
   run_attn();
+  fprintf(stderr, "[san] run_ln_head\n");
   run_ln_head();
   printf("ALL KERNELS PASSED UNDER DEVICE ASAN\n");
   return 0;
