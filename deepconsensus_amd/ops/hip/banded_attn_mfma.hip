@@ -52,6 +52,19 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
+// Un-hoistable lane id (see fused_ffn_v3.hip): at 256 VGPRs the
+// allocator spills lane-derived loop invariants, and each in-loop
+// scratch reload's compiler vmcnt(0) drains the cross-item K/V
+// prefetch — recomputing per item keeps those values loop-local.
+__device__ __forceinline__ int am_lane_recompute() {
+  int l;
+  asm volatile(
+      "v_mbcnt_lo_u32_b32 %0, -1, 0\n\t"
+      "v_mbcnt_hi_u32_b32 %0, -1, %0"
+      : "=v"(l));
+  return l;
+}
+
 template <bool PREFETCH>
 __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
     const bf16* __restrict__ qkv, bf16* __restrict__ out,
@@ -166,6 +179,9 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
   const int BH = B * H;
   if (PREFETCH && blockIdx.x < BH) issue_kv(blockIdx.x);
   for (int item = blockIdx.x; item < BH; item += gridDim.x) {
+    const int ln_ = am_lane_recompute();
+    const int c = ln_ & 31;    // shadow the entry copies: loop-local,
+    const int hi = ln_ >> 5;   // dead at the backedge, nothing to spill
     const bf16* base =
         qkv + (size_t)(item / H) * L * RS + (size_t)(item % H) * D;
     __syncthreads();  // prior item's LDS reads done before re-staging
