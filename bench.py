@@ -98,9 +98,8 @@ def pipeline_main(args):
 
             torch.manual_seed(1234)
             options = qi.InferenceOptions(
-                batch_size=args.batch_size if args.batch_size < 16384
-                else 4096,
-                batch_zmws=50, cpus=cpus, min_quality=0,
+                batch_size=args.batch_size,
+                batch_zmws=100, cpus=cpus, min_quality=0,
                 skip_windows_above=0,
             )
             t0 = time.perf_counter()
@@ -132,7 +131,7 @@ def pipeline_main(args):
                     "--subreads_to_ccs", sub, "--ccs_bam", ccs,
                     "--checkpoint", "random",
                     "--output", os.path.join(td, f"out_{i}.fastq"),
-                    "--batch_size", "4096", "--batch_zmws", "50",
+                    "--batch_size", "8192", "--batch_zmws", "100",
                     "--cpus", str(per_shard_cpus), "--min_quality", "0",
                     "--skip_windows_above", "0",
                     "--shard", f"{i}/{shards}",
