@@ -16,7 +16,7 @@ with tempfile.TemporaryDirectory() as td:
     for i, mode in enumerate(order):
         os.environ["DC_STITCH_MODE"] = mode
         torch.manual_seed(1234)
-        options = qi.InferenceOptions(batch_size=4096, batch_zmws=50,
+        options = qi.InferenceOptions(batch_size=16384, batch_zmws=100,
                                       cpus=min(os.cpu_count() or 4, 16),
                                       min_quality=0, skip_windows_above=0)
         t0 = time.perf_counter()
