@@ -379,6 +379,48 @@ class InferenceRunner:
         return x
 
     @torch.no_grad()
+    def forward_windows_graphed(self, host_rows: torch.Tensor):
+        """Fixed-shape hipGraph-captured serving step (native path).
+
+        Captures H2D-target + embed/encoder/LN-head once per shape and
+        replays it thereafter (the same capture bench.py uses; the
+        whole-pipeline model stage ran eager and measured ~2x slower
+        per window than the graphed bench step). Caller pads the batch
+        to a fixed shape; returns device (bases, quals) views.
+        Falls back to the eager path if capture fails.
+        """
+        if not self.native:
+            return self.forward_windows(host_rows)
+        shape = tuple(host_rows.shape)
+        if getattr(self, "_graph_shape", None) != shape:
+            try:
+                g_in = torch.empty(
+                    shape, dtype=host_rows.dtype, device=self.device
+                )
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(2):
+                        self.forward_windows(g_in)
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    g_bases, g_quals = self.forward_windows(g_in)
+                torch.cuda.synchronize()
+                self._graph = graph
+                self._graph_in = g_in
+                self._graph_out = (g_bases, g_quals)
+                self._graph_shape = shape
+            except Exception:  # pragma: no cover - capture unsupported
+                self._graph_shape = None
+                self._graph = None
+                return self.forward_windows(host_rows)
+        self._graph_in.copy_(host_rows, non_blocking=True)
+        self._graph.replay()
+        return self._graph_out
+
+    @torch.no_grad()
     def forward_windows(
         self, rows: torch.Tensor, want_probs: bool = False
     ):

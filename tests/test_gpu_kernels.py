@@ -760,3 +760,33 @@ def test_model_attention_fused_training_path(monkeypatch):
     )
     assert e_out < 0.05, e_out
     assert e_g < 0.08, e_g
+
+
+def test_forward_windows_graphed_matches_eager():
+    """hipGraph-captured serving step == eager native path, including a
+    zero-padded partial tail."""
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=False)
+    torch.manual_seed(9)
+    runner = InferenceRunner(params, get_model(params), device="cuda")
+    assert runner.native
+    rng = np.random.default_rng(2)
+    B = 512
+    mp, L = params.max_passes, params.max_length
+    rows = np.zeros((B, params.total_rows, L), np.int16)
+    rows[:, 0:mp] = rng.integers(0, 5, size=(B, mp, L))
+    rows[:, mp:3 * mp] = rng.integers(0, 60, size=(B, 2 * mp, L))
+    rows[:, 3 * mp:4 * mp] = rng.integers(1, 3, size=(B, mp, L))
+    rows[:, 4 * mp] = rng.integers(0, 5, size=(B, L))
+    rows[:, -4:] = rng.integers(3, 10, size=(B, 4, 1))
+    host = torch.from_numpy(rows).pin_memory()
+    b_e, q_e = runner.forward_windows(host)
+    b_g, q_g = runner.forward_windows_graphed(host)
+    assert torch.equal(b_g.cpu(), b_e.cpu())
+    assert torch.equal(q_g.cpu(), q_e.cpu())
+    # replay again with fresh data (graph reuse)
+    rows2 = np.ascontiguousarray(rows[::-1])
+    host2 = torch.from_numpy(rows2).pin_memory()
+    b_e2, _ = runner.forward_windows(host2)
+    b_g2, _ = runner.forward_windows_graphed(host2)
+    assert torch.equal(b_g2.cpu(), b_e2.cpu())
