@@ -47,3 +47,19 @@ def test_bench_torchrun_world2_gloo():
     assert r["n_gpus"] == 2 and r["config"]["parallelism"] == "dp2"
     # value is the whole-job aggregate across ranks.
     assert r["config"]["global_batch"] == 64  # 32 per rank on CPU
+
+
+def test_bench_pipeline_mode_schema():
+    """--pipeline emits the JSON contract with per-stage seconds."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--pipeline", "--pipeline-zmws", "4",
+         "--pipeline-length", "800"],
+        cwd=REPO, capture_output=True, text=True, timeout=900,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    r = _last_json_line(out.stdout)
+    assert r["metric"] == "zmw_per_sec_pipeline"
+    assert r["value"] > 0 and r["higher_is_better"] is True
+    assert r["config"]["mode"] == "whole_pipeline"
+    assert r["config"]["reads_written"] == 4
+    assert "run_model" in r["config"]["stage_seconds"]

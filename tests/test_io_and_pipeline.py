@@ -798,3 +798,35 @@ def test_quick_inference_mixed_skip_and_model_windows(tmp_path):
     # Window 0 adopted the CCS sequence verbatim (skip path).
     assert recs[0].sequence[:100] == seq[:100]
     assert len(recs[0].sequence) >= 100
+
+
+def test_quick_inference_stitch_modes_equivalent(tmp_path, monkeypatch):
+    """DC_STITCH_MODE=pool produces byte-identical FASTQ and outcome
+    counts to the serial default."""
+    from deepconsensus_amd.inference import quick_inference as qi
+
+    sub, ccs = make_test_bams(tmp_path, n_zmws=4, length=150, seed=3)
+    outs = {}
+    counters = {}
+    for mode in ("serial", "pool"):
+        monkeypatch.setenv("DC_STITCH_MODE", mode)
+        import torch
+
+        torch.manual_seed(7)
+        out = str(tmp_path / f"out_{mode}.fastq")
+        options = qi.InferenceOptions(
+            batch_size=64, batch_zmws=2, cpus=2, min_quality=0,
+            skip_windows_above=0,
+        )
+        c = qi.run(subreads_to_ccs=sub, ccs_bam=ccs, checkpoint="random",
+                   output=out, options=options)
+        outs[mode] = open(out).read()
+        counters[mode] = dataclasses_asdict(c)
+    assert outs["serial"] == outs["pool"]
+    assert counters["serial"] == counters["pool"]
+
+
+def dataclasses_asdict(c):
+    import dataclasses
+
+    return dataclasses.asdict(c)
