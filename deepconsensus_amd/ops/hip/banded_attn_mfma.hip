@@ -65,10 +65,13 @@ __device__ __forceinline__ int am_lane_recompute() {
   return l;
 }
 
-template <bool PREFETCH>
+template <bool PREFETCH, bool SAVE_P = false>
 __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
     const bf16* __restrict__ qkv, bf16* __restrict__ out,
-    int B, int L, int H, int win, float scale) {
+    int B, int L, int H, int win, float scale,
+    bf16* __restrict__ p_out = nullptr,
+    const uint8_t* __restrict__ drop_mask = nullptr,
+    float keep_inv = 1.0f) {
   constexpr int D = AM_D;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -264,6 +267,33 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
 #pragma unroll
     for (int i = 0; i < 32; ++i) st[i] *= inv;
 
+    if (SAVE_P) {
+      // Training-forward mode: persist the band softmax (pre-dropout,
+      // bf16 like the torch path's cast) as P[item][qrow][w], w = kl-c
+      // in [0, 2win], then apply the caller's dropout band mask to the
+      // PV input registers.
+      const int W = 2 * win + 1;
+      const int qr = l0w + c;
+      if (qr < L) {
+        bf16* prow = p_out + ((size_t)item * L + qr) * W;
+        const uint8_t* mrow =
+            drop_mask ? drop_mask + ((size_t)item * L + qr) * W : nullptr;
+#pragma unroll
+        for (int i = 0; i < 32; ++i) {
+          const int kl =
+              32 * (i >> 4) + (i & 3) + 8 * ((i & 15) >> 2) + 4 * hi;
+          const int w = kl - c;
+          if (w >= 0 && w < W) {
+            const bf16 pb = __float2bfloat16(st[i]);
+            prow[w] = pb;
+            float pd = __bfloat162float(pb);
+            if (mrow) pd *= mrow[w] ? keep_inv : 0.f;
+            st[i] = pd;
+          }
+        }
+      }
+    }
+
     // ---- Repack P to A-fragments: 4 k-steps x 4 dwords (T12 pattern).
     // Lane pair (l, l+32) holds interleaved keys; after cvt_pk + two
     // permlane32_swaps per k-step, lane l's fragment covers its 8 contiguous
@@ -315,6 +345,41 @@ __global__ __launch_bounds__(256, 2) void banded_attn_mfma_kernel(
 }  // namespace
 
 #ifndef DC_SAN_MAIN
+
+std::vector<at::Tensor> banded_attn_mfma_train_fwd(
+    at::Tensor qkv, int64_t H, int64_t win, double scale,
+    at::Tensor drop_mask, double p_drop) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == at::kBFloat16,
+              "qkv must be bf16 on device");
+  auto q = qkv.contiguous();
+  const int B = q.size(0), L = q.size(1);
+  const int D = q.size(2) / (3 * H);
+  TORCH_CHECK(D == AM_D && L <= AM_L && L >= 32,
+              "banded_attn_mfma requires D=140, 32<=L<=104");
+  TORCH_CHECK(win <= 12 && win >= 1, "win must be in [1, 12]");
+  const int W = 2 * (int)win + 1;
+  auto out = at::empty({B, L, H * D}, q.options());
+  auto p = at::zeros({B * (int)H, L, W}, q.options());
+  const bool has_mask = drop_mask.defined() && drop_mask.numel() > 0;
+  at::Tensor mc;
+  const uint8_t* mptr = nullptr;
+  if (has_mask) {
+    mc = drop_mask.contiguous();
+    mptr = (const uint8_t*)mc.data_ptr();
+  }
+  const float keep_inv =
+      p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 1.0f;
+  dim3 grid(std::min(B * (int)H, 512));
+  dim3 block(256);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL((banded_attn_mfma_kernel<true, true>), grid, block,
+                     0, stream, reinterpret_cast<bf16*>(q.data_ptr()),
+                     reinterpret_cast<bf16*>(out.data_ptr()), B, L,
+                     (int)H, (int)win, (float)scale,
+                     reinterpret_cast<bf16*>(p.data_ptr()), mptr,
+                     keep_inv);
+  return {out, p};
+}
 
 at::Tensor banded_attn_mfma(at::Tensor qkv, int64_t H, int64_t win,
                             double scale) {

@@ -337,3 +337,195 @@ std::vector<at::Tensor> banded_attn_train_bwd(
 }
 
 #endif  // DC_SAN_MAIN
+
+// ---------------------------------------------------------------------------
+// bwd v2: packed layouts + occupancy diet. Pairs with the MFMA serving
+// forward's SAVE_P variant (banded_attn_mfma.hip): qkv/dout arrive in
+// the serving layout ([B,L,3HD] / [B,L,HD]) so no transposes or
+// contiguous() copies sit on the train path, and dq/dk/dv write one
+// packed dqkv whose grad flows through a single fused QKV Linear.
+// LDS holds only V + dO tiles (+ band dS fp32 / Pd bf16): K and Q rows
+// stream from L2 in the 8-wide inner loops, cutting LDS from 140 KB to
+// ~75 KB = 2 blocks/CU (the v1 kernel's 1-block occupancy was the
+// measured 0.62x loss vs the torch chain).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+__global__ __launch_bounds__(256, 2) void battn_bwd2_kernel(
+    const bf16_t* __restrict__ qkv,   // [B, T, 3*H*D]
+    const bf16_t* __restrict__ p_in,  // [B*H, T, W]
+    const uint8_t* __restrict__ mask, // [B*H, T, W] or null
+    const bf16_t* __restrict__ dout,  // [B, T, H*D]
+    bf16_t* __restrict__ dqkv,        // [B, T, 3*H*D]
+    int BH, int H, int T, int D, int win, float scale, float keep_inv) {
+  __shared__ bf16_t vs[BT_MAXT][BT_MAXD];
+  __shared__ bf16_t dos[BT_MAXT][BT_MAXD];
+  __shared__ float dss[BT_MAXT][BT_MAXW];
+  __shared__ bf16_t pds[BT_MAXT][BT_MAXW];
+
+  const int item = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int W = 2 * win + 1;
+  const int h = item % H;
+  const int b = item / H;
+  const size_t RS = (size_t)3 * H * D;   // qkv row stride
+  const size_t OS = (size_t)H * D;       // dout/dqkv-head row stride
+  const bf16_t* qb = qkv + (size_t)b * T * RS + (size_t)h * D;
+  const bf16_t* kb = qb + OS;
+  const bf16_t* vb = qb + 2 * OS;
+  const bf16_t* db = dout + (size_t)b * T * OS + (size_t)h * D;
+  bf16_t* dqb = dqkv + (size_t)b * T * RS + (size_t)h * D;
+  bf16_t* dkb = dqb + OS;
+  bf16_t* dvb = dqb + 2 * OS;
+
+  for (int idx = tid; idx < T * BT_MAXD; idx += 256) {
+    const int t = idx / BT_MAXD, d = idx % BT_MAXD;
+    const bf16_t z = __float2bfloat16(0.f);
+    const bool in = d < D;
+    vs[t][d] = in ? vb[(size_t)t * RS + d] : z;
+    dos[t][d] = in ? db[(size_t)t * OS + d] : z;
+  }
+  __syncthreads();
+
+  // dPd = dO @ v^T (band); Pd staged bf16.
+  for (int cell = tid; cell < T * W; cell += 256) {
+    const int l = cell / W, w = cell % W;
+    const int kc = l - win + w;
+    float dpd = 0.f, pd = 0.f;
+    if (kc >= 0 && kc < T) {
+      float acc = 0.f;
+      for (int d = 0; d < BT_MAXD; d += 8) {
+        acc += dot8(&dos[l][d], &vs[kc][d]);
+      }
+      dpd = acc;
+      pd = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
+      if (mask != nullptr) {
+        pd *= mask[((size_t)item * T + l) * W + w] ? keep_inv : 0.f;
+      }
+    }
+    pds[l][w] = __float2bfloat16(pd);
+    dss[l][w] = dpd;
+  }
+  __syncthreads();
+
+  // dS = P * (dP - rowsum(dP * P)); one thread per row.
+  for (int l = tid; l < T; l += 256) {
+    float dp_row[BT_MAXW];
+    float r = 0.f;
+    for (int w = 0; w < W; ++w) {
+      const int kc = l - win + w;
+      float p_ = 0.f, dp_ = 0.f;
+      if (kc >= 0 && kc < T) {
+        p_ = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
+        dp_ = dss[l][w];
+        if (mask != nullptr) {
+          dp_ *= mask[((size_t)item * T + l) * W + w] ? keep_inv : 0.f;
+        }
+      }
+      dp_row[w] = dp_;
+      r += dp_ * p_;
+    }
+    for (int w = 0; w < W; ++w) {
+      const int kc = l - win + w;
+      float p_ = 0.f;
+      if (kc >= 0 && kc < T) {
+        p_ = __bfloat162float(p_in[((size_t)item * T + l) * W + w]);
+      }
+      dss[l][w] = p_ * (dp_row[w] - r);
+    }
+  }
+  __syncthreads();
+
+  // Outputs: 8-wide d-granules; K and Q rows stream from L2.
+  const int ng = T * (BT_MAXD / 8);
+  for (int g = tid; g < ng; g += 256) {
+    const int l = g / (BT_MAXD / 8), d0 = (g % (BT_MAXD / 8)) * 8;
+    if (d0 >= D) continue;
+    const bool full8 = d0 + 8 <= D;
+    float aq[8] = {}, ak[8] = {}, av[8] = {};
+    const int w0 = max(0, win - l);
+    const int w1 = min(W, T + win - l);
+    for (int w = w0; w < w1; ++w) {
+      const float dsw = dss[l][w];
+      const bf16_t* kr = kb + (size_t)(l - win + w) * RS + d0;
+      if (full8) {
+        const btx8 kv = *reinterpret_cast<const btx8*>(kr);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) aq[j] += dsw * (float)kv[j];
+      } else {
+        for (int j = 0; j + d0 < D; ++j) {
+          aq[j] += dsw * __bfloat162float(kr[j]);
+        }
+      }
+    }
+    const int kc = l;
+    const int l2_0 = max(0, kc - win);
+    const int l2_1 = min(T, kc + win + 1);
+    for (int l2 = l2_0; l2 < l2_1; ++l2) {
+      const int w = kc - l2 + win;
+      const float dsw = dss[l2][w];
+      const float pdw = __bfloat162float(pds[l2][w]);
+      const bf16_t* qr = qb + (size_t)l2 * RS + d0;
+      const btx8 dov = *reinterpret_cast<const btx8*>(&dos[l2][d0]);
+      if (full8) {
+        const btx8 qv = *reinterpret_cast<const btx8*>(qr);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          ak[j] += dsw * (float)qv[j];
+          av[j] += pdw * (float)dov[j];
+        }
+      } else {
+        for (int j = 0; j + d0 < D; ++j) {
+          ak[j] += dsw * __bfloat162float(qr[j]);
+          av[j] += pdw * (float)dov[j];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (d0 + j < D) {
+        const size_t off = (size_t)l * RS + d0 + j;
+        dqb[off] = __float2bfloat16(aq[j] * scale);
+        dkb[off] = __float2bfloat16(ak[j] * scale);
+        dvb[off] = __float2bfloat16(av[j]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+#ifndef DC_SAN_MAIN
+
+at::Tensor banded_attn_train_bwd2(
+    at::Tensor qkv, at::Tensor p, at::Tensor mask, at::Tensor dout,
+    int64_t H, int64_t win, double p_drop) {
+  auto qc = qkv.contiguous();
+  auto pc = p.contiguous();
+  auto dc = dout.contiguous();
+  const int B = qc.size(0), T = qc.size(1);
+  const int D = qc.size(2) / (3 * (int)H);
+  TORCH_CHECK(T <= BT_MAXT && D <= BT_MAXD, "shape exceeds limits");
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  at::Tensor mc;
+  const uint8_t* mptr = nullptr;
+  if (has_mask) {
+    mc = mask.contiguous();
+    mptr = (const uint8_t*)mc.data_ptr();
+  }
+  auto dqkv = at::empty_like(qc);
+  const float keep_inv =
+      p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 1.0f;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(battn_bwd2_kernel, dim3(B * (int)H), dim3(256), 0,
+                     stream, (const bf16_t*)qc.data_ptr(),
+                     (const bf16_t*)pc.data_ptr(), mptr,
+                     (const bf16_t*)dc.data_ptr(),
+                     (bf16_t*)dqkv.data_ptr(), B * (int)H, (int)H, T, D,
+                     (int)win, (float)(1.0 / std::sqrt((double)D)),
+                     keep_inv);
+  return dqkv;
+}
+
+#endif  // DC_SAN_MAIN

@@ -214,6 +214,12 @@ std::vector<at::Tensor> banded_attn_train_fwd(
 std::vector<at::Tensor> banded_attn_train_bwd(
     at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor p,
     at::Tensor mask, at::Tensor dout, int64_t win, double p_drop);
+std::vector<at::Tensor> banded_attn_mfma_train_fwd(
+    at::Tensor qkv, int64_t H, int64_t win, double scale,
+    at::Tensor drop_mask, double p_drop);
+at::Tensor banded_attn_train_bwd2(
+    at::Tensor qkv, at::Tensor p, at::Tensor mask, at::Tensor dout,
+    int64_t H, int64_t win, double p_drop);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
@@ -242,6 +248,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Device AlignmentMetric (K14): affine NW + backtrace counts");
   m.def("banded_attn_train_fwd", &banded_attn_train_fwd,
         "Training banded attention forward (saves band softmax P)");
+  m.def("banded_attn_mfma_train_fwd", &banded_attn_mfma_train_fwd,
+        "MFMA training attention forward (serving kernel + band-P save "
+        "+ fused dropout)");
+  m.def("banded_attn_train_bwd2", &banded_attn_train_bwd2,
+        "Training attention backward v2 (packed qkv layout, 2 blocks/"
+        "CU) -> packed dqkv");
   m.def("banded_attn_train_bwd", &banded_attn_train_bwd,
         "Training banded attention backward (band-local dS -> dq,dk,dv)");
   m.def("fused_ffn_v4", &fused_ffn_v4,

@@ -43,10 +43,23 @@ def fused_seg():
     ctx = _BandedAttnTrain.apply(qq, kk, vv, mask, win, 0.1)
     ctx.backward(g)
 
+from deepconsensus_amd.models.model import _BandedAttnTrainPacked
+
+qkv_p = (torch.randn(B, T, 3 * H * D, device="cuda") * 0.3).to(torch.bfloat16)
+g_p = torch.randn(B, T, H * D, device="cuda").to(torch.bfloat16)
+
+def packed_seg():
+    qq = qkv_p.detach().requires_grad_()
+    mask = torch.rand(B * H, T, 2 * win + 1, device="cuda") >= 0.1
+    ctx = _BandedAttnTrainPacked.apply(qq, mask, H, win, 0.1)
+    ctx.backward(g_p)
+
 t_torch = timeit(torch_seg)
 t_fused = timeit(fused_seg)
+t_packed = timeit(packed_seg)
 print(f"torch chain fwd+bwd: {t_torch:.0f} us")
-print(f"fused HIP fwd+bwd:   {t_fused:.0f} us  ({t_torch/t_fused:.2f}x)")
+print(f"fused HIP v1 fwd+bwd: {t_fused:.0f} us  ({t_torch/t_fused:.2f}x)")
+print(f"fused HIP v2 packed:  {t_packed:.0f} us  ({t_torch/t_packed:.2f}x)")
 # kernels alone
 mask = torch.rand(B, H, T, 2 * win + 1, device="cuda") >= 0.1
 tf = timeit(lambda: ext.banded_attn_train_fwd(q, k, v, mask, win, 0.1))

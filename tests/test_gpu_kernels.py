@@ -790,3 +790,44 @@ def test_forward_windows_graphed_matches_eager():
     b_e2, _ = runner.forward_windows(host2)
     b_g2, _ = runner.forward_windows_graphed(host2)
     assert torch.equal(b_g2.cpu(), b_e2.cpu())
+
+
+def test_banded_attn_packed_train_matches_autograd():
+    """Packed fused training attention v2 (MFMA fwd + bwd2): ctx and
+    dqkv match torch autograd through the same math (dropout 0)."""
+    from deepconsensus_amd.models.model import _BandedAttnTrainPacked
+
+    torch.manual_seed(21)
+    B, H, T, D, win = 16, 2, 100, 140, 12
+    qkv = (torch.randn(B, T, 3 * H * D, device="cuda") * 0.3).to(
+        torch.bfloat16
+    ).requires_grad_()
+
+    def torch_ref(qkv):
+        q, k, v = (
+            qkv.view(B, T, 3, H, D).permute(2, 0, 3, 1, 4).unbind(0)
+        )
+        scale = D ** -0.5
+        logits = torch.matmul(q * scale, k.transpose(-1, -2))
+        i = torch.arange(T, device="cuda")
+        band = (i[:, None] - i[None, :]).abs() <= win
+        logits = logits.masked_fill(~band, -1e9)
+        w = torch.softmax(logits.float(), dim=-1).to(qkv.dtype)
+        ctx = torch.matmul(w, v)  # [B,H,T,D]
+        return ctx.transpose(1, 2).reshape(B, T, H * D)
+
+    ref = torch_ref(qkv)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    dref = qkv.grad.clone()
+    qkv.grad = None
+
+    qkv2 = qkv.detach().requires_grad_()
+    out = _BandedAttnTrainPacked.apply(qkv2, None, H, win, 0.0)
+    e_out = (out.float() - ref.float()).abs().max().item()
+    assert e_out < 0.03, e_out
+    out.backward(g)
+    dgot = qkv2.grad
+    scale_ref = dref.float().abs().max().item() + 1e-6
+    e_g = (dgot.float() - dref.float()).abs().max().item() / scale_ref
+    assert e_g < 0.06, e_g
