@@ -119,7 +119,7 @@ class _BandedAttnTrainPacked(torch.autograd.Function):
         ext = dc_ops.get_ext(required=True)
         qkv, p, mask_t = ctx.saved_tensors
         num_heads, win, p_drop = ctx.meta
-        dqkv = ext.banded_attn_train_bwd2(
+        dqkv = ext.banded_attn_bwd_mfma(
             qkv, p, mask_t, dout.contiguous(), num_heads, win, p_drop
         )
         return dqkv, None, None, None, None

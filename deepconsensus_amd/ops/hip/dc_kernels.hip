@@ -220,6 +220,9 @@ std::vector<at::Tensor> banded_attn_mfma_train_fwd(
 at::Tensor banded_attn_train_bwd2(
     at::Tensor qkv, at::Tensor p, at::Tensor mask, at::Tensor dout,
     int64_t H, int64_t win, double p_drop);
+at::Tensor banded_attn_bwd_mfma(
+    at::Tensor qkv, at::Tensor p, at::Tensor mask, at::Tensor dout,
+    int64_t H, int64_t win, double p_drop);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
@@ -251,6 +254,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("banded_attn_mfma_train_fwd", &banded_attn_mfma_train_fwd,
         "MFMA training attention forward (serving kernel + band-P save "
         "+ fused dropout)");
+  m.def("banded_attn_bwd_mfma", &banded_attn_bwd_mfma,
+        "MFMA training attention backward (band_scores/band_apply "
+        "primitives; packed dqkv)");
   m.def("banded_attn_train_bwd2", &banded_attn_train_bwd2,
         "Training attention backward v2 (packed qkv layout, 2 blocks/"
         "CU) -> packed dqkv");

@@ -66,3 +66,15 @@ tf = timeit(lambda: ext.banded_attn_train_fwd(q, k, v, mask, win, 0.1))
 out, p = ext.banded_attn_train_fwd(q, k, v, mask, win, 0.1)
 tb = timeit(lambda: ext.banded_attn_train_bwd(q, k, v, p, mask, g, win, 0.1))
 print(f"kernel fwd: {tf:.0f} us   kernel bwd: {tb:.0f} us")
+qkv_c = qkv_p.contiguous()
+mask2 = torch.rand(B * H, T, 2 * win + 1, device="cuda") >= 0.1
+tfm = timeit(lambda: ext.banded_attn_mfma_train_fwd(
+    qkv_c, H, win, D ** -0.5, mask2, 0.1))
+out2, p2 = ext.banded_attn_mfma_train_fwd(qkv_c, H, win, D ** -0.5,
+                                          mask2, 0.1)
+tbm = timeit(lambda: ext.banded_attn_bwd_mfma(
+    qkv_c, p2, mask2, g_p, H, win, 0.1))
+tb2 = timeit(lambda: ext.banded_attn_train_bwd2(
+    qkv_c, p2, mask2, g_p, H, win, 0.1))
+print(f"MFMA fwd: {tfm:.0f} us  MFMA bwd: {tbm:.0f} us  "
+      f"(VALU bwd2: {tb2:.0f} us)")

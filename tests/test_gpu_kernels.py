@@ -792,6 +792,61 @@ def test_forward_windows_graphed_matches_eager():
     assert torch.equal(b_g2.cpu(), b_e2.cpu())
 
 
+def test_banded_attn_bwd_mfma_matches_autograd_with_dropout():
+    """MFMA backward with a fixed dropout mask == torch autograd run
+    with the same mask (full packed chain)."""
+    from deepconsensus_amd import ops as dc_ops
+
+    ext = dc_ops.get_ext(required=True)
+    torch.manual_seed(33)
+    B, H, T, D, win, p_drop = 12, 2, 100, 140, 12, 0.25
+    W = 2 * win + 1
+    qkv = (torch.randn(B, T, 3 * H * D, device="cuda") * 0.3).to(
+        torch.bfloat16
+    ).requires_grad_()
+    mask = torch.rand(B * H, T, W, device="cuda") >= p_drop
+    i = torch.arange(T, device="cuda")
+    band = (i[:, None] - i[None, :]).abs() <= win
+    full_mask = torch.zeros(B, H, T, T, device="cuda",
+                            dtype=torch.bool)
+    mb = mask.view(B, H, T, W)
+    for wi in range(W):
+        kc = i - win + wi
+        valid = (kc >= 0) & (kc < T)
+        full_mask[:, :, valid, kc[valid]] = mb[:, :, valid, wi]
+
+    def torch_ref(qkv):
+        q, k, v = (
+            qkv.view(B, T, 3, H, D).permute(2, 0, 3, 1, 4).unbind(0)
+        )
+        scale = D ** -0.5
+        logits = torch.matmul(q * scale, k.transpose(-1, -2))
+        logits = logits.masked_fill(~band, -1e9)
+        w = torch.softmax(logits.float(), dim=-1).to(qkv.dtype)
+        wd = torch.where(full_mask, w.float() / (1 - p_drop), 0.0).to(
+            qkv.dtype
+        )
+        ctx = torch.matmul(wd, v)
+        return ctx.transpose(1, 2).reshape(B, T, H * D)
+
+    ref = torch_ref(qkv)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    dref = qkv.grad.clone()
+
+    out, p = ext.banded_attn_mfma_train_fwd(
+        qkv.detach(), H, win, D ** -0.5, mask, p_drop
+    )
+    e_out = (out.float() - ref.float()).abs().max().item()
+    assert e_out < 0.05, e_out
+    dqkv = ext.banded_attn_bwd_mfma(
+        qkv.detach(), p, mask, g, H, win, p_drop
+    )
+    scale_ref = dref.float().abs().max().item() + 1e-6
+    e_g = (dqkv.float() - dref.float()).abs().max().item() / scale_ref
+    assert e_g < 0.08, e_g
+
+
 def test_banded_attn_packed_train_matches_autograd():
     """Packed fused training attention v2 (MFMA fwd + bwd2): ctx and
     dqkv match torch autograd through the same math (dropout 0)."""
