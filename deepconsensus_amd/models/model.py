@@ -35,17 +35,17 @@ _BATTN_AVAILABLE = None
 
 
 def _battn_train_available() -> bool:
-    """Fused training attention is OPT-IN (DC_ATTN_TRAIN=1): measured
-    0.62x the torch chain at batch 4096 (2,059 us fwd + 4,655 us bwd vs
-    4,188 us torch fwd+bwd; gpurun_out/r2_attnb.log) — the band-only
-    VALU kernels are occupancy-bound at 1 block/CU and lose to
-    hipBLASLt's full-matrix path. Kept for parity/iteration; see
-    profiles/r02_perf_journal.md."""
+    """Fused training attention v2 (MFMA forward with band-P save +
+    MFMA band-primitive backward) measures 2.13x the torch chain at
+    batch 4096 (404 + 1,510 us vs 4,180 us fwd+bwd;
+    profiles/r02_perf_journal.md) — DEFAULT ON. DC_ATTN_TRAIN=0 falls
+    back to the torch chain. (The earlier VALU pair, kept for the
+    sanitizer/oracle tests, measured 0.62x.)"""
     global _BATTN_AVAILABLE
     if _BATTN_AVAILABLE is None:
         import os
 
-        if os.environ.get("DC_ATTN_TRAIN") != "1":
+        if os.environ.get("DC_ATTN_TRAIN", "1") == "0":
             _BATTN_AVAILABLE = False
             return False
         try:
@@ -53,7 +53,9 @@ def _battn_train_available() -> bool:
 
             ext = dc_ops.get_ext()
             _BATTN_AVAILABLE = bool(
-                ext is not None and hasattr(ext, "banded_attn_train_fwd")
+                ext is not None
+                and hasattr(ext, "banded_attn_mfma_train_fwd")
+                and hasattr(ext, "banded_attn_bwd_mfma")
             )
         except Exception:  # pragma: no cover
             _BATTN_AVAILABLE = False
