@@ -41,6 +41,9 @@ namespace san_ln {
 namespace san_attn {
 #include "../hip/banded_attn_mfma.hip"
 }
+namespace san_abwd {
+#include "../hip/banded_attn_bwd_mfma.hip"
+}
 using bf16 = __hip_bfloat16;
 
 #define CHECK(x)                                                        \
@@ -212,6 +215,30 @@ static void run_attn() {
   printf("banded_attn_mfma OK under ASAN\n");
 }
 
+static void run_attn_bwd() {
+  const int B = 256, L = 100, H = 2, D = 140, win = 12;
+  const int W = 2 * win + 1;
+  std::vector<unsigned short> qkv((size_t)B * L * 3 * H * D),
+      pvec((size_t)B * H * L * W), dov((size_t)B * L * H * D);
+  for (auto& v : qkv) v = f2bf(frand() * 0.3f);
+  for (auto& v : pvec) v = f2bf(0.04f);
+  for (auto& v : dov) v = f2bf(frand() * 0.3f);
+  std::vector<unsigned char> msk((size_t)B * H * L * W, 1);
+  auto d_qkv = to_dev(qkv);
+  auto d_p = to_dev(pvec);
+  auto d_m = to_dev(msk);
+  auto d_do = to_dev(dov);
+  auto d_dq = dalloc<unsigned short>((size_t)B * L * 3 * H * D);
+  const int BH = B * H;
+  hipLaunchKernelGGL(san_abwd::battn_bwd_mfma_kernel,
+                     dim3(BH < 1024 ? BH : 1024), dim3(256), 0, 0,
+                     (const bf16*)d_qkv, (const bf16*)d_p, d_m,
+                     (const bf16*)d_do, (bf16*)d_dq, B, L, H, win,
+                     0.084515f, 1.111f);
+  CHECK(hipDeviceSynchronize());
+  printf("banded_attn_bwd_mfma OK under ASAN\n");
+}
+
 static void run_ln_head() {
   const int N = 40960, H = 280;
   std::vector<unsigned short> x((size_t)N * H);
@@ -250,6 +277,8 @@ int main() {
   run_alignment_dp();
   fprintf(stderr, "[san] run_attn\n");
   run_attn();
+  fprintf(stderr, "[san] run_attn_bwd\n");
+  run_attn_bwd();
   fprintf(stderr, "[san] run_ln_head\n");
   run_ln_head();
   printf("ALL NON-DMA KERNELS PASSED UNDER DEVICE ASAN\n");
