@@ -217,8 +217,7 @@ class BandedSelfAttention(nn.Module):
         b, t, _ = x.shape
         h, d = self.num_heads, self.head_dim
         if (
-            training
-            and not need_weights
+            not need_weights
             and self.attn_win_size is not None
             and x.is_cuda
             and x.dtype in (torch.bfloat16, torch.float32)
@@ -240,17 +239,17 @@ class BandedSelfAttention(nn.Module):
             )
             qkv = F.linear(x, w_qkv)  # [B, T, 3*H*D], bf16 (autocast)
             drop_mask = None
-            if self.dropout > 0:
+            p_drop = float(self.dropout) if training else 0.0
+            if p_drop > 0:
                 drop_mask = (
                     torch.rand(
                         b * h, t, 2 * self.attn_win_size + 1,
                         device=x.device,
                     )
-                    >= self.dropout
+                    >= p_drop
                 )
             ctx = _BandedAttnTrainPacked.apply(
-                qkv, drop_mask, h, self.attn_win_size,
-                float(self.dropout),
+                qkv, drop_mask, h, self.attn_win_size, p_drop,
             )
             return self.out_proj(ctx), None
         q = self.q_proj(x).view(b, t, h, d).transpose(1, 2)  # [B,H,T,D]
