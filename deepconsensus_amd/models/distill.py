@@ -83,6 +83,7 @@ def train_model(
     device: str = "cpu",
     eval_every: int = 3000,
     limit_steps: int = 0,
+    use_bf16: bool = False,
 ) -> dict:
     rank, world = comm.init_distributed()
     main = rank == 0
@@ -131,13 +132,26 @@ def train_model(
             schedule.apply(optimizer, step)
             rows, label = _prepare_batch(batch, device)
             reducer.zero_()
-            with torch.no_grad():
-                t_logits = teacher.encode(rows, training=False)["logits"]
-            s_out = student.encode(rows, training=True)
+            # bf16 autocast: teacher + student forwards take the fused
+            # MFMA banded-attention path on GPU (model.py); losses stay
+            # fp32.
+            if use_bf16 and rows.is_cuda:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    with torch.no_grad():
+                        t_logits = teacher.encode(
+                            rows, training=False
+                        )["logits"]
+                    s_out = student.encode(rows, training=True)
+            else:
+                with torch.no_grad():
+                    t_logits = teacher.encode(
+                        rows, training=False
+                    )["logits"]
+                s_out = student.encode(rows, training=True)
             s_logits = s_out["logits"]
             probs = torch.softmax(s_logits.float(), -1)
             l_student = align_loss(label, probs) / global_batch
-            l_distill = distill_loss(t_logits, s_logits)
+            l_distill = distill_loss(t_logits.float(), s_logits.float())
             loss = student_alpha * l_student + distill_alpha * l_distill
             loss.backward()
             reducer.reduce()
@@ -174,6 +188,9 @@ def main(argv: Optional[List[str]] = None) -> None:
     ap.add_argument("--epochs", type=int, default=None)
     ap.add_argument("--limit_steps", type=int, default=0)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--bf16", action="store_true",
+                    help="autocast forwards to bf16 (fused MFMA "
+                    "attention path on GPU)")
     args = ap.parse_args(argv)
     params = cfg.get_config(args.params)
     if args.train_path:
