@@ -205,7 +205,8 @@ def main(argv: Optional[List[str]] = None) -> None:
         "cuda" if torch.cuda.is_available() else "cpu"
     )
     train_model(args.out_dir, args.teacher_checkpoint, params,
-                device=device, limit_steps=args.limit_steps)
+                device=device, limit_steps=args.limit_steps,
+                use_bf16=getattr(args, "bf16", False))
 
 
 if __name__ == "__main__":
