@@ -163,3 +163,30 @@ def test_embed_grad_kernel_matches_cpu_fallback():
     g_gpu = grads_on("cuda:0")
     for a, b in zip(g_cpu, g_gpu):
         torch.testing.assert_close(a, b, atol=2e-3, rtol=1e-3)
+
+
+def test_distill_bf16_e2e_gpu(tmp_path):
+    """Distillation loop on cuda with bf16 autocast: teacher + student
+    forwards run the fused MFMA attention path; checkpoint written."""
+    from deepconsensus_amd.models import distill as distill_lib
+    from deepconsensus_amd.models import train as train_lib
+
+    train_file, _ = make_training_data(tmp_path)
+    params = _tiny_params(train_file)
+    teacher_dir = str(tmp_path / "teacher")
+    train_lib.train_model(
+        teacher_dir, params, device="cuda", eval_every=100,
+        limit_steps=1,
+    )
+    sparams = _tiny_params(train_file)
+    sparams.num_hidden_layers = max(params.num_hidden_layers - 1, 1)
+    from deepconsensus_amd.models.config import modify_params
+
+    modify_params(sparams)
+    out = str(tmp_path / "student")
+    summary = distill_lib.train_model(
+        out, teacher_dir, sparams, device="cuda", limit_steps=2,
+        use_bf16=True,
+    )
+    assert summary["steps"] >= 2
+    assert glob.glob(os.path.join(out, "checkpoint-*.pt"))
