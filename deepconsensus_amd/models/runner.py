@@ -187,6 +187,24 @@ class InferenceRunner:
             if model.add_pos_encoding
             else None
         )
+        # fused_condense (K3+K4): weight image [Npad, 568] (row n =
+        # condenser.weight[n], LDS row stride) + fp32 pos table fused into
+        # the epilogue. Production width only; DC_FUSED_CONDENSE=0 falls
+        # back to hipBLASLt + separate add.
+        import os as _os
+
+        self.cond_img = None
+        self.pos_f32 = None
+        cw = model.condenser.weight.detach()
+        if cw.shape[1] == 560 and _os.environ.get("DC_FUSED_CONDENSE") != "0":
+            npad = -(-cw.shape[0] // 64) * 64
+            img = torch.zeros(npad, 568, dtype=bf16)
+            img[: cw.shape[0], :560] = cw.to(bf16)
+            self.cond_img = img.contiguous().to(dev)
+            if model.add_pos_encoding:
+                self.pos_f32 = (
+                    model.pos_encoding.detach().float().contiguous().to(dev)
+                )
         # Packed per-layer weights (fast rezero path); generic fallback keeps
         # a bf16 deepcopy of the layer stack for non-rezero configs.
         self.rezero_fast = all(
@@ -320,11 +338,24 @@ class InferenceRunner:
                 self.row_vocab, self.chunk_cnt, self.chunk_entries,
             )  # [B, L, concat] bf16
         b, l, _ = emb.shape
-        with trace.range("K3_condenser"):
-            x = emb.reshape(b * l, -1) @ self.cond_wt  # [B*L, H]
-        x = x.view(b, l, -1)
-        if self.pos is not None:
-            x = x + self.pos[:l]
+        if self.cond_img is not None:
+            with trace.range("K3_condenser"):
+                h = self.cond_wt.shape[1]
+                pos = (
+                    self.pos_f32
+                    if self.pos_f32 is not None
+                    else emb.new_empty(0, dtype=torch.float32)
+                )
+                x = self.ext.fused_condense(
+                    emb.reshape(b * l, -1), self.cond_img, pos, h, l
+                )
+            x = x.view(b, l, -1)
+        else:
+            with trace.range("K3_condenser"):
+                x = emb.reshape(b * l, -1) @ self.cond_wt  # [B*L, H]
+            x = x.view(b, l, -1)
+            if self.pos is not None:
+                x = x + self.pos[:l]
         if self.rezero_fast:
             h = x.shape[-1]
             flat = x.reshape(b * l, h)

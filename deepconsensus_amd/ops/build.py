@@ -27,6 +27,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "fused_ffn_v4.hip"),
     os.path.join(_OPS_DIR, "hip", "banded_attn_train.hip"),
     os.path.join(_OPS_DIR, "hip", "banded_attn_bwd_mfma.hip"),
+    os.path.join(_OPS_DIR, "hip", "fused_condense.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

@@ -223,8 +223,13 @@ at::Tensor banded_attn_train_bwd2(
 at::Tensor banded_attn_bwd_mfma(
     at::Tensor qkv, at::Tensor p, at::Tensor mask, at::Tensor dout,
     int64_t H, int64_t win, double p_drop);
+at::Tensor fused_condense(at::Tensor x, at::Tensor w, at::Tensor pos,
+                          int64_t n_out, int64_t seq_len);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_condense", &fused_condense,
+        "Condenser GEMM [M,560]x[560,N] + fused position-encoding add "
+        "(K3+K4)");
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
         "Fused final LayerNorm + head + softmax + QV (K10+K11+K12)");
   m.def("banded_attn", &banded_attn,

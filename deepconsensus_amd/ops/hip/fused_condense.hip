@@ -1,0 +1,228 @@
+// Fused condenser GEMM for gfx950: out = x @ W^T + pos  (K3 + K4).
+//
+// The production model's no-bias condenser Linear (networks.py:426-434,
+// 509-516: [B*L, 560] x [560, 280]) plus the sinusoidal position-encoding
+// add (networks.py:203-205) fused into the epilogue — the last library
+// kernel (hipBLASLt) and the last standalone elementwise op on the serving
+// path, replaced by one MFMA kernel reusing the fused_linear structure.
+//
+// Shape specifics vs fused_linear (K = 280 there, 560 here):
+//  * BM = 256 with 8 waves, each wave OWNS one 32-row group and walks both
+//    32-column halves of the 64-col weight chunk (halves weight refetch
+//    vs BM=128: 2.3 GB/call at M = 1.6 M rows).
+//  * A-fragments: 35 bf16x8 granules per lane (140 VGPRs) — so one block
+//    per CU (LDS: 64x568 weight image 72.7 KB + 256x72 out image 36.9 KB).
+//  * The 35-step dependent MFMA chain is split into two independent
+//    accumulators (even/odd granules) merged at the epilogue, since at
+//    2 waves/SIMD there is less latency cover than fused_linear's 4.
+//  * Epilogue adds pos[(m0+row) % L, col] (fp32 table) before the bf16
+//    round — the separate "+ pos" elementwise kernel disappears.
+//
+// W arrives host-padded to [Npad, 568] (LDS row stride; Npad mult of 64)
+// with row n = condenser.weight[n, :560] — see runner.py.
+
+#ifndef DC_SAN_MAIN
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#endif
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int BM = 256;
+constexpr int KIN = 560;         // condenser input width
+constexpr int NG = 35;           // 16-elem K granules (560 / 16)
+constexpr int NC = 64;           // output-column chunk
+constexpr int W_STRIDE = 568;    // weight/x LDS row stride
+constexpr int O_STRIDE = 72;     // output chunk image stride
+
+// Unhoistable lane id (volatile v_mbcnt): keeps lane-derived LDS addresses
+// loop-local so nothing spills around the chunk loop (see fused_ffn_v3 —
+// a spilled address's scratch reload carries a vmcnt(0) that drains the
+// in-flight glds queue).
+__device__ __forceinline__ int lane_recompute() {
+  int l;
+  asm volatile(
+      "v_mbcnt_lo_u32_b32 %0, -1, 0\n\t"
+      "v_mbcnt_hi_u32_b32 %0, -1, %0"
+      : "=v"(l));
+  return l;
+}
+
+__global__ __launch_bounds__(512, 1) void fused_condense_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ pos, bf16* __restrict__ out,
+    int M, int N, int Npad, int L) {
+  __shared__ __attribute__((aligned(16))) bf16 w_lds[NC * W_STRIDE];
+  __shared__ __attribute__((aligned(16))) bf16 o_lds[BM * O_STRIDE];
+
+  const int tid = threadIdx.x;
+  const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
+  const int m0 = blockIdx.x * BM;
+
+  // Stage x through w_lds 64 rows at a time; the wave whose 32-row group
+  // sits in that quarter pulls its A-fragments. 70 granules of 8 elems
+  // per 560-wide row.
+  bf16x8 af[NG];
+  for (int quarter = 0; quarter < 4; ++quarter) {
+    __syncthreads();
+    for (int idx = tid; idx < 64 * 70; idx += 512) {
+      const int r = idx / 70, q8 = idx % 70;
+      const int row = 64 * quarter + r;
+      uint4 v = {};
+      if (m0 + row < M) {
+        v = *reinterpret_cast<const uint4*>(
+            x + (size_t)(m0 + row) * KIN + 8 * q8);
+      }
+      *reinterpret_cast<uint4*>(&w_lds[r * W_STRIDE + 8 * q8]) = v;
+    }
+    __syncthreads();
+    if (wave / 2 == quarter) {
+      const int ln = lane_recompute();
+      const int r_local = 32 * (wave & 1) + (ln & 31);
+      const int hi = ln >> 5;
+#pragma unroll
+      for (int s = 0; s < NG; ++s) {
+        af[s] = *reinterpret_cast<const bf16x8*>(
+            &w_lds[r_local * W_STRIDE + 16 * s + 8 * hi]);
+      }
+    }
+  }
+  __syncthreads();
+
+  // Weight chunk = 64 rows x 568 elems = 71 KiB-units; glds stream, each
+  // wave copying KiB-units wave, wave+8, ... (per-lane 16-B granules).
+  auto issue_w = [&](int chunk) {
+    const int ln = lane_recompute();
+    const bf16* src = w + (size_t)chunk * NC * W_STRIDE;
+#pragma unroll
+    for (int i = 0; i < 9; ++i) {
+      const int ck = wave + i * 8;
+      if (ck < NC * W_STRIDE * 2 / 1024) {
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned*)(
+                src + ck * 512 + ln * 8),
+            (__attribute__((address_space(3))) unsigned*)(
+                &w_lds[ck * 512]),
+            16, 0, 0);
+      }
+    }
+  };
+
+  issue_w(0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int nchunk = Npad / NC;
+  for (int chunk = 0; chunk < nchunk; ++chunk) {
+    const int ln = lane_recompute();
+    const int c = ln & 31;
+    const int hi = ln >> 5;
+#pragma unroll
+    for (int ch = 0; ch < 2; ++ch) {        // 32-col halves of the chunk
+      const int colt = 32 * ch;
+      const int ncol = chunk * NC + colt + c;
+      f32x16 acc0 = {}, acc1 = {};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int s = 0; s < NG; s += 2) {     // two independent chains
+        const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+            &w_lds[(colt + c) * W_STRIDE + 16 * s + 8 * hi]);
+        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], b0, acc0,
+                                                       0, 0, 0);
+        if (s + 1 < NG) {
+          const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
+              &w_lds[(colt + c) * W_STRIDE + 16 * (s + 1) + 8 * hi]);
+          acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s + 1], b1,
+                                                         acc1, 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * wave;
+        float v = acc0[r] + acc1[r];
+        if (pos != nullptr && ncol < N && m0 + row < M) {
+          v += pos[(size_t)((m0 + row) % L) * N + ncol];
+        }
+        o_lds[row * O_STRIDE + colt + c] = __float2bfloat16(v);
+      }
+    }
+    __syncthreads();  // o_lds complete; w_lds consumed
+    if (chunk + 1 < nchunk) issue_w(chunk + 1);
+
+    // Coalesced copy-out of this 64-col chunk, overlapping the glds fetch.
+    const int n0 = chunk * NC;
+    for (int idx = tid; idx < BM * (NC / 8); idx += 512) {
+      const int row = idx / (NC / 8), c8 = idx % (NC / 8);
+      if (m0 + row >= M) continue;
+      const int col = n0 + 8 * c8;
+      if (col >= N) continue;
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          &o_lds[row * O_STRIDE + 8 * c8]);
+      if (col + 8 <= N) {
+        *reinterpret_cast<bf16x8*>(
+            out + (size_t)(m0 + row) * N + col) = v;
+      } else {
+        const unsigned short* vs =
+            reinterpret_cast<const unsigned short*>(&v);
+        for (int j = 0; j < 8 && col + j < N; ++j) {
+          reinterpret_cast<unsigned short*>(
+              out)[(size_t)(m0 + row) * N + col + j] = vs[j];
+        }
+      }
+    }
+    if (chunk + 1 < nchunk) {
+      asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+}
+
+}  // namespace
+
+#ifndef DC_SAN_MAIN
+
+at::Tensor fused_condense(at::Tensor x, at::Tensor w, at::Tensor pos,
+                          int64_t n_out, int64_t seq_len) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16,
+              "x must be bf16 on device");
+  auto xc = x.contiguous();
+  const int K = xc.size(-1);
+  const int M = xc.numel() / K;
+  TORCH_CHECK(K == KIN, "fused_condense requires width-560 input");
+  const int Npad = w.size(0);
+  TORCH_CHECK(w.size(1) == W_STRIDE && Npad % NC == 0 &&
+                  w.dtype() == at::kBFloat16,
+              "w must be bf16 [Npad (mult of 64), 568]");
+  const int N = (int)n_out;
+  TORCH_CHECK(N <= Npad, "n_out exceeds padded weight rows");
+  const int L = (int)seq_len;
+  const float* pos_ptr = nullptr;
+  at::Tensor pc;
+  if (pos.defined() && pos.numel() > 0) {
+    pc = pos.contiguous();
+    TORCH_CHECK(pc.dtype() == at::kFloat && pc.size(-1) == N &&
+                    pc.numel() >= (int64_t)L * N && L > 0,
+                "pos must be fp32 [>=L, N]");
+    pos_ptr = pc.data_ptr<float>();
+  }
+  auto out = at::empty({M, N}, xc.options());
+  dim3 grid((M + BM - 1) / BM);
+  dim3 block(512);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fused_condense_kernel, grid, block, 0, stream,
+                     reinterpret_cast<bf16*>(xc.data_ptr()),
+                     reinterpret_cast<bf16*>(w.data_ptr()), pos_ptr,
+                     reinterpret_cast<bf16*>(out.data_ptr()), M, N, Npad,
+                     L > 0 ? L : 1);
+  return out;
+}
+
+#endif  // DC_SAN_MAIN

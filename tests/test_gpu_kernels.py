@@ -886,3 +886,38 @@ def test_banded_attn_packed_train_matches_autograd():
     scale_ref = dref.float().abs().max().item() + 1e-6
     e_g = (dgot.float() - dref.float()).abs().max().item() / scale_ref
     assert e_g < 0.06, e_g
+
+
+def test_fused_condense_matches_torch(setup):
+    """fused_condense (K3+K4) vs fp32 matmul + position add."""
+    params, model, runner, rows = setup
+    torch.manual_seed(41)
+    for M, L in ((1600, 100), (300, 77)):
+        x = (torch.randn(M, 560, device="cuda") * 0.5).to(torch.bfloat16)
+        w = torch.randn(280, 560, device="cuda") * 0.05
+        w_img = torch.zeros(320, 568, device="cuda")
+        w_img[:280, :560] = w
+        w_img = w_img.to(torch.bfloat16).contiguous()
+        pos = torch.randn(L, 280, device="cuda") * 0.3
+        out = runner.ext.fused_condense(x, w_img, pos, 280, L).float()
+        idx = torch.arange(M, device="cuda") % L
+        ref = x.float() @ w.t() + pos[idx]
+        assert out.shape == (M, 280)
+        assert (out - ref).abs().max().item() < 0.05
+        # No-pos variant.
+        empty = torch.empty(0, device="cuda", dtype=torch.float32)
+        out2 = runner.ext.fused_condense(x, w_img, empty, 280, L).float()
+        assert (out2 - x.float() @ w.t()).abs().max().item() < 0.05
+
+
+def test_runner_condense_fallback_agrees(setup, monkeypatch):
+    """Runner output with DC_FUSED_CONDENSE=0 matches the fused default."""
+    params, model, runner, rows = setup
+    assert runner.cond_img is not None
+    b1, q1 = runner.forward_windows(rows)
+    monkeypatch.setenv("DC_FUSED_CONDENSE", "0")
+    runner2 = InferenceRunner(params, model, device="cuda:0")
+    assert runner2.cond_img is None
+    b2, q2 = runner2.forward_windows(rows)
+    assert (b1 == b2).float().mean().item() > 0.999
+    assert (q1.float() - q2.float()).abs().mean().item() < 0.5
