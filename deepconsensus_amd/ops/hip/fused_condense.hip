@@ -66,36 +66,6 @@ __global__ __launch_bounds__(512, 1) void fused_condense_kernel(
   const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
   const int m0 = blockIdx.x * BM;
 
-  // Stage x through w_lds 64 rows at a time; the wave whose 32-row group
-  // sits in that quarter pulls its A-fragments. 70 granules of 8 elems
-  // per 560-wide row.
-  bf16x8 af[NG];
-  for (int quarter = 0; quarter < 4; ++quarter) {
-    __syncthreads();
-    for (int idx = tid; idx < 64 * 70; idx += 512) {
-      const int r = idx / 70, q8 = idx % 70;
-      const int row = 64 * quarter + r;
-      uint4 v = {};
-      if (m0 + row < M) {
-        v = *reinterpret_cast<const uint4*>(
-            x + (size_t)(m0 + row) * KIN + 8 * q8);
-      }
-      *reinterpret_cast<uint4*>(&w_lds[r * W_STRIDE + 8 * q8]) = v;
-    }
-    __syncthreads();
-    if (wave / 2 == quarter) {
-      const int ln = lane_recompute();
-      const int r_local = 32 * (wave & 1) + (ln & 31);
-      const int hi = ln >> 5;
-#pragma unroll
-      for (int s = 0; s < NG; ++s) {
-        af[s] = *reinterpret_cast<const bf16x8*>(
-            &w_lds[r_local * W_STRIDE + 16 * s + 8 * hi]);
-      }
-    }
-  }
-  __syncthreads();
-
   // Weight chunk = 64 rows x 568 elems = 71 KiB-units; glds stream, each
   // wave copying KiB-units wave, wave+8, ... (per-lane 16-B granules).
   auto issue_w = [&](int chunk) {
@@ -116,6 +86,27 @@ __global__ __launch_bounds__(512, 1) void fused_condense_kernel(
   };
 
   issue_w(0);
+
+  // A-fragments straight from global: lane (c, hi) of wave w owns row
+  // m0 + 32w + c, granules 16s + 8hi — the wave collectively reads its
+  // contiguous 32-row x 1120-B region, so every DRAM sector is consumed
+  // exactly once; no LDS bounce, no staging barriers (the 4-pass
+  // stage-through-w_lds prologue this replaces cost ~1/3 of the kernel).
+  bf16x8 af[NG];
+  {
+    const int ln = lane_recompute();
+    const int row = m0 + 32 * wave + (ln & 31);
+    if (row < M) {
+      const bf16* xr = x + (size_t)row * KIN + 8 * (ln >> 5);
+#pragma unroll
+      for (int s = 0; s < NG; ++s) {
+        af[s] = *reinterpret_cast<const bf16x8*>(xr + 16 * s);
+      }
+    } else {
+#pragma unroll
+      for (int s = 0; s < NG; ++s) af[s] = bf16x8{};
+    }
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
