@@ -3,7 +3,11 @@
 Production serving shape: M = 16384*100 rows, 560 -> 280, pos [100, 280].
 Run on a GPU box:  python scripts/condense_bench.py
 """
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
