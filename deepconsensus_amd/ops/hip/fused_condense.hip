@@ -38,10 +38,9 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 constexpr int BM = 256;
 constexpr int KIN = 560;         // condenser input width
 constexpr int NG = 35;           // 16-elem K granules (560 / 16)
-constexpr int NC = 32;           // output-column chunk (double-buffered)
-constexpr int W_STRIDE = 568;    // weight LDS row stride
-constexpr int O_STRIDE = 40;     // output chunk image stride
-constexpr int WCHUNK_B = NC * W_STRIDE * 2;  // 36,352 B = 35.5 KiB units
+constexpr int NC = 64;           // output-column chunk
+constexpr int W_STRIDE = 568;    // weight/x LDS row stride
+constexpr int O_STRIDE = 72;     // output chunk image stride
 
 // Unhoistable lane id (volatile v_mbcnt): keeps lane-derived LDS addresses
 // loop-local so nothing spills around the chunk loop (see fused_ffn_v3 —
@@ -60,44 +59,39 @@ __global__ __launch_bounds__(512, 1) void fused_condense_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     const float* __restrict__ pos, bf16* __restrict__ out,
     int M, int N, int Npad, int L) {
-  // Two weight buffers (36.4 KB each) + output image (20.5 KB) = 93 KB:
-  // chunk n+1 streams by glds while chunk n computes — the NC=64 variant
-  // could not double-buffer (2x72.7 + 36.9 > 160 KB) and measured its
-  // weight DMA fully serialized with the MFMAs.
-  __shared__ __attribute__((aligned(16))) bf16 w_lds[2][NC * W_STRIDE];
+  __shared__ __attribute__((aligned(16))) bf16 w_lds[NC * W_STRIDE];
   __shared__ __attribute__((aligned(16))) bf16 o_lds[BM * O_STRIDE];
 
   const int tid = threadIdx.x;
   const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
   const int m0 = blockIdx.x * BM;
 
-  // Weight chunk = 32 rows x 568 elems = 35.5 KiB units; glds stream,
-  // each wave copying units wave, wave+8, ... (per-lane 16-B granules);
-  // the trailing half unit is issued by lanes 0-31 only.
-  auto issue_w = [&](int chunk, int buf) {
+  // Weight chunk = 64 rows x 568 elems = 71 KiB-units; glds stream, each
+  // wave copying KiB-units wave, wave+8, ... (per-lane 16-B granules).
+  auto issue_w = [&](int chunk) {
     const int ln = lane_recompute();
     const bf16* src = w + (size_t)chunk * NC * W_STRIDE;
 #pragma unroll
-    for (int i = 0; i < 5; ++i) {
+    for (int i = 0; i < 9; ++i) {
       const int ck = wave + i * 8;
-      if (ck * 1024 + ln * 16 < WCHUNK_B) {
+      if (ck < NC * W_STRIDE * 2 / 1024) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned*)(
                 src + ck * 512 + ln * 8),
             (__attribute__((address_space(3))) unsigned*)(
-                &w_lds[buf][ck * 512]),
+                &w_lds[ck * 512]),
             16, 0, 0);
       }
     }
   };
 
-  issue_w(0, 0);
+  issue_w(0);
 
   // A-fragments straight from global: lane (c, hi) of wave w owns row
   // m0 + 32w + c, granules 16s + 8hi — the wave collectively reads its
   // contiguous 32-row x 1120-B region, so every DRAM sector is consumed
   // exactly once; no LDS bounce, no staging barriers (the 4-pass
-  // stage-through-LDS prologue this replaces cost ~1/3 of the kernel).
+  // stage-through-w_lds prologue this replaces cost ~1/3 of the kernel).
   bf16x8 af[NG];
   {
     const int ln = lane_recompute();
@@ -121,38 +115,40 @@ __global__ __launch_bounds__(512, 1) void fused_condense_kernel(
     const int ln = lane_recompute();
     const int c = ln & 31;
     const int hi = ln >> 5;
-    const int buf = chunk & 1;
-    // Next chunk's DMA in flight while this chunk computes.
-    if (chunk + 1 < nchunk) issue_w(chunk + 1, buf ^ 1);
-    const int ncol = chunk * NC + c;
-    f32x16 acc0 = {}, acc1 = {};
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < NG; s += 2) {     // two independent chains
-      const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
-          &w_lds[buf][c * W_STRIDE + 16 * s + 8 * hi]);
-      acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], b0, acc0,
-                                                     0, 0, 0);
-      if (s + 1 < NG) {
-        const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
-            &w_lds[buf][c * W_STRIDE + 16 * (s + 1) + 8 * hi]);
-        acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s + 1], b1,
-                                                       acc1, 0, 0, 0);
+    for (int ch = 0; ch < 2; ++ch) {        // 32-col halves of the chunk
+      const int colt = 32 * ch;
+      const int ncol = chunk * NC + colt + c;
+      f32x16 acc0 = {}, acc1 = {};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int s = 0; s < NG; s += 2) {     // two independent chains
+        const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+            &w_lds[(colt + c) * W_STRIDE + 16 * s + 8 * hi]);
+        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s], b0, acc0,
+                                                       0, 0, 0);
+        if (s + 1 < NG) {
+          const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
+              &w_lds[(colt + c) * W_STRIDE + 16 * (s + 1) + 8 * hi]);
+          acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[s + 1], b1,
+                                                         acc1, 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * wave;
+        float v = acc0[r] + acc1[r];
+        if (pos != nullptr && ncol < N && m0 + row < M) {
+          v += pos[(size_t)((m0 + row) % L) * N + ncol];
+        }
+        o_lds[row * O_STRIDE + colt + c] = __float2bfloat16(v);
       }
     }
-    __builtin_amdgcn_s_setprio(0);
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * wave;
-      float v = acc0[r] + acc1[r];
-      if (pos != nullptr && ncol < N && m0 + row < M) {
-        v += pos[(size_t)((m0 + row) % L) * N + ncol];
-      }
-      o_lds[row * O_STRIDE + c] = __float2bfloat16(v);
-    }
-    __syncthreads();  // o_lds complete; w_lds[buf] consumed
+    __syncthreads();  // o_lds complete; w_lds consumed
+    if (chunk + 1 < nchunk) issue_w(chunk + 1);
 
-    // Coalesced copy-out of this 32-col chunk, overlapping the glds.
+    // Coalesced copy-out of this 64-col chunk, overlapping the glds fetch.
     const int n0 = chunk * NC;
     for (int idx = tid; idx < BM * (NC / 8); idx += 512) {
       const int row = idx / (NC / 8), c8 = idx % (NC / 8);
@@ -173,8 +169,6 @@ __global__ __launch_bounds__(512, 1) void fused_condense_kernel(
         }
       }
     }
-    // Drain the next chunk's glds before computing from it; the barrier
-    // also fences o_lds reuse across waves.
     if (chunk + 1 < nchunk) {
       asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     }
