@@ -179,9 +179,20 @@ def test_encode_native_equals_condensed_input_at_alpha0():
     emb = r.ext.embed_gather(rows.contiguous(), r.table_flat, r.row_shift,
                              r.row_vocab, r.chunk_cnt, r.chunk_entries)
     b, l, _ = emb.shape
-    want = (emb.reshape(b * l, -1) @ r.cond_wt).view(b, l, -1)
-    if r.pos is not None:
-        want = want + r.pos[:l]
+    if r.cond_img is not None:
+        # Same kernel as encode_native so the identity stays bitwise.
+        pos = (
+            r.pos_f32
+            if r.pos_f32 is not None
+            else emb.new_empty(0, dtype=torch.float32)
+        )
+        want = r.ext.fused_condense(
+            emb.reshape(b * l, -1), r.cond_img, pos, r.cond_wt.shape[1], l
+        ).view(b, l, -1)
+    else:
+        want = (emb.reshape(b * l, -1) @ r.cond_wt).view(b, l, -1)
+        if r.pos is not None:
+            want = want + r.pos[:l]
     bad = (x != want).any(dim=-1)
     assert not bad.any(), (
         f"alpha=0 encoder is not identity for {int(bad.sum())} of {b * l} "
