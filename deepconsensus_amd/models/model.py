@@ -62,6 +62,83 @@ def _battn_train_available() -> bool:
     return _BATTN_AVAILABLE
 
 
+_FFN_TRAIN_AVAILABLE = None
+
+
+def _ffn_train_available() -> bool:
+    """Fused training FFN (ffn_train.hip: v3-structure forward with
+    in-register relu + hash dropout, dgrad fusing both transposed GEMMs
+    with the mask recovered from hd>0) — DEFAULT ON for the production
+    280/2048 shape. DC_FFN_TRAIN=0 falls back to the torch chain."""
+    global _FFN_TRAIN_AVAILABLE
+    if _FFN_TRAIN_AVAILABLE is None:
+        import os
+
+        if os.environ.get("DC_FFN_TRAIN", "1") == "0":
+            _FFN_TRAIN_AVAILABLE = False
+            return False
+        try:
+            from deepconsensus_amd import ops as dc_ops
+
+            ext = dc_ops.get_ext()
+            _FFN_TRAIN_AVAILABLE = bool(
+                ext is not None
+                and hasattr(ext, "ffn_train_fwd")
+                and hasattr(ext, "ffn_train_dgrad")
+            )
+        except Exception:  # pragma: no cover
+            _FFN_TRAIN_AVAILABLE = False
+    return _FFN_TRAIN_AVAILABLE
+
+
+class _FFNTrainFused(torch.autograd.Function):
+    """Training FFN on the HIP pair (ops/hip/ffn_train.hip): forward
+    keeps the 2048-wide intermediate in registers (relu + counter-hash
+    dropout applied there, hd persisted once for the wgrads), dgrad
+    fuses dhd = dy@W2^T -> mask -> dx = dh_pre@W1^T the same way. The
+    split-K wgrads (x^T@dh_pre, hd^T@dy) and bias sums stay on
+    hipBLASLt/torch. x is the flattened bf16 [M, 280] input."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2, p_drop, seed):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        bf16 = torch.bfloat16
+        w1b = w1.detach().to(bf16)
+        w1_img = torch.zeros(2048, 296, dtype=bf16, device=x.device)
+        w1_img[:, :280] = w1b
+        w1_img[:, 287] = b1.detach().to(bf16)
+        w2b = w2.detach().to(bf16)
+        w2_img = torch.zeros(320, 2048, dtype=bf16, device=x.device)
+        w2_img[:280] = w2b
+        y, hd = ext.ffn_train_fwd(
+            x, w1_img, w2_img, b2.detach().float(), p_drop, seed
+        )
+        ctx.save_for_backward(x, hd, w1b, w2b)
+        ctx.p_drop = p_drop
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        x, hd, w1b, w2b = ctx.saved_tensors
+        bf16 = torch.bfloat16
+        dyc = dy.contiguous()
+        w2t_img = torch.zeros(2048, 296, dtype=bf16, device=dy.device)
+        w2t_img[:, :280] = w2b.t()
+        w1t_img = torch.zeros(320, 2048, dtype=bf16, device=dy.device)
+        w1t_img[:280] = w1b.t()
+        dx, dh = ext.ffn_train_dgrad(dyc, hd, w2t_img, w1t_img, ctx.p_drop)
+        dw2 = (dyc.t() @ hd).float()
+        dw1 = (dh.t() @ x).float()
+        db2 = dyc.sum(0).float()
+        db1 = dh.sum(0).float()
+        return dx, dw1, db1, dw2, db2, None, None
+
+
 class _BandedAttnTrain(torch.autograd.Function):
     """HIP banded attention for the training path (K5-K7 on device):
     band-only compute + fused softmax/dropout, band-local backward
@@ -287,6 +364,28 @@ class FeedForward(nn.Module):
         self.dropout = dropout
 
     def forward(self, x: torch.Tensor, training: bool) -> torch.Tensor:
+        if (
+            training
+            and x.is_cuda
+            and torch.is_autocast_enabled()
+            and self.filter_layer.in_features == 280
+            and self.filter_layer.out_features == 2048
+            and _ffn_train_available()
+        ):
+            shape = x.shape
+            xb = x.to(torch.bfloat16).reshape(-1, shape[-1])
+            seed = int(torch.randint(0, 2 ** 31 - 1, ()).item())
+            p = float(self.dropout) if self.dropout > 0 else 0.0
+            y = _FFNTrainFused.apply(
+                xb,
+                self.filter_layer.weight,
+                self.filter_layer.bias,
+                self.output_layer.weight,
+                self.output_layer.bias,
+                p,
+                seed,
+            )
+            return y.view(shape)
         y = F.relu(self.filter_layer(x))
         if training and self.dropout > 0:
             y = F.dropout(y, p=self.dropout, training=True)

@@ -28,6 +28,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "banded_attn_train.hip"),
     os.path.join(_OPS_DIR, "hip", "banded_attn_bwd_mfma.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_condense.hip"),
+    os.path.join(_OPS_DIR, "hip", "ffn_train.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

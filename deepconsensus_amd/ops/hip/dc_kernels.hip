@@ -225,11 +225,23 @@ at::Tensor banded_attn_bwd_mfma(
     int64_t H, int64_t win, double p_drop);
 at::Tensor fused_condense(at::Tensor x, at::Tensor w, at::Tensor pos,
                           int64_t n_out, int64_t seq_len);
+std::vector<at::Tensor> ffn_train_fwd(at::Tensor x, at::Tensor w1,
+                                      at::Tensor w2, at::Tensor b2,
+                                      double p_drop, int64_t seed);
+std::vector<at::Tensor> ffn_train_dgrad(at::Tensor dy, at::Tensor hd,
+                                        at::Tensor w2t, at::Tensor w1t,
+                                        double p_drop);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_condense", &fused_condense,
         "Condenser GEMM [M,560]x[560,N] + fused position-encoding add "
         "(K3+K4)");
+  m.def("ffn_train_fwd", &ffn_train_fwd,
+        "Training FFN forward: relu + hash dropout in-register, returns "
+        "(y, hd)");
+  m.def("ffn_train_dgrad", &ffn_train_dgrad,
+        "Training FFN dgrad: both transposed GEMMs fused, mask from hd>0, "
+        "returns (dx, dh_pre)");
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
         "Fused final LayerNorm + head + softmax + QV (K10+K11+K12)");
   m.def("banded_attn", &banded_attn,

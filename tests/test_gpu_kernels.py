@@ -924,3 +924,138 @@ def test_runner_condense_fallback_agrees(setup, monkeypatch):
     # near-ties are expected at ~1e-3 rate, same as native-vs-torch.
     assert (b1 == b2).float().mean().item() > 0.99
     assert (q1.float() - q2.float()).abs().mean().item() < 1.0
+
+
+def _ffn_imgs(w1, b1, w2, dev):
+    bf16 = torch.bfloat16
+    w1_img = torch.zeros(2048, 296, dtype=bf16, device=dev)
+    w1_img[:, :280] = w1.to(bf16)
+    w1_img[:, 287] = b1.to(bf16)
+    w2_img = torch.zeros(320, 2048, dtype=bf16, device=dev)
+    w2_img[:280] = w2.to(bf16)
+    return w1_img, w2_img
+
+
+def test_ffn_train_fwd_matches_torch(setup):
+    """ffn_train_fwd at p=0 vs the fp32 torch chain (y and hd)."""
+    params, model, runner, rows = setup
+    torch.manual_seed(11)
+    M = 2048 + 96
+    x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+    w1 = torch.randn(2048, 280, device="cuda") * 0.05
+    b1 = torch.randn(2048, device="cuda") * 0.1
+    w2 = torch.randn(280, 2048, device="cuda") * 0.02
+    b2 = torch.randn(280, device="cuda") * 0.1
+    w1_img, w2_img = _ffn_imgs(w1, b1, w2, "cuda")
+    y, hd = runner.ext.ffn_train_fwd(x, w1_img, w2_img, b2, 0.0, 123)
+    h_ref = torch.relu(x.float() @ w1.t() + b1)
+    y_ref = h_ref @ w2.t() + b2
+    assert (hd.float() - h_ref).abs().max().item() < 0.05
+    assert (y.float() - y_ref).abs().max().item() < 0.25
+    # hd is bitwise re-derivable: second call identical.
+    y2, hd2 = runner.ext.ffn_train_fwd(x, w1_img, w2_img, b2, 0.0, 123)
+    assert torch.equal(y, y2) and torch.equal(hd, hd2)
+
+
+def test_ffn_train_dropout_semantics(setup):
+    """p=0.5: keep-rate ~ 1-p, y == hd@W2+b2, determinism per seed."""
+    params, model, runner, rows = setup
+    torch.manual_seed(12)
+    M = 4096
+    x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+    w1 = torch.randn(2048, 280, device="cuda") * 0.05
+    b1 = torch.zeros(2048, device="cuda")
+    w2 = torch.randn(280, 2048, device="cuda") * 0.02
+    b2 = torch.zeros(280, device="cuda")
+    w1_img, w2_img = _ffn_imgs(w1, b1, w2, "cuda")
+    y, hd = runner.ext.ffn_train_fwd(x, w1_img, w2_img, b2, 0.5, 99)
+    h_ref = torch.relu(x.float() @ w1.t() + b1)
+    pos = h_ref > 1e-3
+    keep = (hd.float()[pos] != 0).float().mean().item()
+    assert 0.45 < keep < 0.55, keep
+    # Kept elements are h * 1/(1-p); dropped are exactly zero.
+    kept_mask = hd.float() != 0
+    scaled = (hd.float() - 2.0 * h_ref).abs() * kept_mask
+    assert scaled.max().item() < 0.25
+    y_ref = hd.float() @ w2.t() + b2
+    assert (y.float() - y_ref).abs().max().item() < 0.25
+    y2, hd2 = runner.ext.ffn_train_fwd(x, w1_img, w2_img, b2, 0.5, 99)
+    assert torch.equal(hd, hd2)
+    _, hd3 = runner.ext.ffn_train_fwd(x, w1_img, w2_img, b2, 0.5, 100)
+    assert not torch.equal(hd, hd3)
+
+
+def test_ffn_train_function_grads_match_autograd(setup):
+    """_FFNTrainFused end-to-end grads vs the torch chain with the SAME
+    mask (derived from hd>0), p=0.3."""
+    from deepconsensus_amd.models.model import _FFNTrainFused
+
+    params, model, runner, rows = setup
+    torch.manual_seed(13)
+    M, p = 4096, 0.3
+    x = (torch.randn(M, 280, device="cuda") * 0.5).to(torch.bfloat16)
+    x_f = x.clone().requires_grad_(True)
+    w1 = (torch.randn(2048, 280, device="cuda") * 0.05).requires_grad_(True)
+    b1 = (torch.randn(2048, device="cuda") * 0.1).requires_grad_(True)
+    w2 = (torch.randn(280, 2048, device="cuda") * 0.02).requires_grad_(True)
+    b2 = (torch.randn(280, device="cuda") * 0.1).requires_grad_(True)
+    y = _FFNTrainFused.apply(x_f, w1, b1, w2, b2, p, 77)
+    dy = torch.randn_like(y.float()).to(torch.bfloat16)
+    y.backward(dy)
+    # Torch reference with the mask the kernel actually drew.
+    with torch.no_grad():
+        w1_img, w2_img = _ffn_imgs(w1, b1, w2, "cuda")
+        _, hd = runner.ext.ffn_train_fwd(
+            x, w1_img, w2_img, b2.detach().float(), p, 77
+        )
+        mask = (hd > 0).float()
+    x_r = x.clone().float().requires_grad_(True)
+    w1_r = w1.detach().clone().requires_grad_(True)
+    b1_r = b1.detach().clone().requires_grad_(True)
+    w2_r = w2.detach().clone().requires_grad_(True)
+    b2_r = b2.detach().clone().requires_grad_(True)
+    h_r = torch.relu(x_r @ w1_r.t() + b1_r)
+    hd_r = h_r * mask / (1 - p)
+    y_r = hd_r @ w2_r.t() + b2_r
+    y_r.backward(dy.float())
+
+    def rel(a, b):
+        return (a.float() - b.float()).abs().max().item() / (
+            b.float().abs().max().item() + 1e-9
+        )
+
+    assert (y.float() - y_r).abs().max().item() < 0.3
+    assert rel(x_f.grad, x_r.grad) < 0.06
+    assert rel(w1.grad, w1_r.grad) < 0.06
+    assert rel(w2.grad, w2_r.grad) < 0.06
+    assert rel(b1.grad, b1_r.grad) < 0.06
+    assert rel(b2.grad, b2_r.grad) < 0.06
+
+
+def test_model_ffn_fused_training_path(monkeypatch):
+    """FeedForward.forward(training=True) routes through the fused pair
+    under autocast and its grads match the torch path (dropout 0)."""
+    import deepconsensus_amd.models.model as model_mod
+    from deepconsensus_amd.models.model import FeedForward
+
+    monkeypatch.setenv("DC_FFN_TRAIN", "1")
+    monkeypatch.setattr(model_mod, "_FFN_TRAIN_AVAILABLE", None)
+    torch.manual_seed(15)
+    ffn = FeedForward(280, 2048, 0.0).cuda()
+    x = torch.randn(8, 100, 280, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y_fused = ffn(x, training=True)
+    y_fused.float().sum().backward()
+    g_fused = ffn.filter_layer.weight.grad.clone()
+    ffn.zero_grad()
+    monkeypatch.setattr(model_mod, "_FFN_TRAIN_AVAILABLE", False)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y_ref = ffn(x, training=True)
+    y_ref.float().sum().backward()
+    g_ref = ffn.filter_layer.weight.grad.clone()
+    e_out = (y_fused.float() - y_ref.float()).abs().max().item()
+    e_g = (g_fused - g_ref).abs().max().item() / (
+        g_ref.abs().max().item() + 1e-9
+    )
+    assert e_out < 0.3, e_out
+    assert e_g < 0.08, e_g
