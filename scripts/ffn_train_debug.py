@@ -1,0 +1,64 @@
+"""Isolate the ffn_train_dgrad mismatch: compare dx and dh_pre piecewise."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from deepconsensus_amd import ops as dc_ops
+
+
+def main():
+    ext = dc_ops.get_ext(required=True)
+    torch.manual_seed(3)
+    bf16 = torch.bfloat16
+    M, p = 512, 0.0
+    x = (torch.randn(M, 280, device="cuda") * 0.5).to(bf16)
+    w1 = torch.randn(2048, 280, device="cuda") * 0.05
+    b1 = torch.randn(2048, device="cuda") * 0.1
+    w2 = torch.randn(280, 2048, device="cuda") * 0.02
+    b2 = torch.zeros(280, device="cuda")
+    w1_img = torch.zeros(2048, 296, dtype=bf16, device="cuda")
+    w1_img[:, :280] = w1.to(bf16)
+    w1_img[:, 287] = b1.to(bf16)
+    w2_img = torch.zeros(320, 2048, dtype=bf16, device="cuda")
+    w2_img[:280] = w2.to(bf16)
+    y, hd = ext.ffn_train_fwd(x, w1_img, w2_img, b2, p, 7)
+
+    dy = (torch.randn(M, 280, device="cuda") * 0.5).to(bf16)
+    w2t_img = torch.zeros(2048, 296, dtype=bf16, device="cuda")
+    w2t_img[:, :280] = w2.to(bf16).t()
+    w1t_img = torch.zeros(320, 2048, dtype=bf16, device="cuda")
+    w1t_img[:280] = w1.to(bf16).t()
+    dx, dh = ext.ffn_train_dgrad(dy, hd, w2t_img, w1t_img, p)
+
+    dhd_ref = dy.float() @ w2  # [M, 2048]
+    mask = (hd.float() > 0).float()
+    dh_ref = dhd_ref * mask
+    dx_ref = dh_ref @ w1
+
+    e_dh = (dh.float() - dh_ref).abs()
+    e_dx = (dx.float() - dx_ref).abs()
+    print("dh err max/mean:", e_dh.max().item(), e_dh.mean().item(),
+          " ref scale:", dh_ref.abs().max().item())
+    print("dx err max/mean:", e_dx.max().item(), e_dx.mean().item(),
+          " ref scale:", dx_ref.abs().max().item())
+    # Where is dh wrong? column histogram of bad entries.
+    bad = (e_dh > 0.05).nonzero()
+    if bad.numel():
+        print("bad dh entries:", bad.shape[0], "first:", bad[:8].tolist())
+        cols = bad[:, 1]
+        print("col range:", int(cols.min()), int(cols.max()),
+              "col%64 hist:", torch.bincount(cols % 64, minlength=64)[:32].tolist())
+        rows = bad[:, 0]
+        print("row range:", int(rows.min()), int(rows.max()),
+              "row%256 min/max:", int((rows % 256).min()), int((rows % 256).max()))
+    # dhd WITHOUT mask: run dgrad against an all-positive hd.
+    dx2, dh2 = ext.ffn_train_dgrad(dy, torch.ones_like(hd), w2t_img, w1t_img, p)
+    e2 = (dh2.float() - dhd_ref).abs()
+    print("unmasked dhd err max/mean:", e2.max().item(), e2.mean().item())
+
+
+if __name__ == "__main__":
+    main()
