@@ -58,6 +58,36 @@ def main():
     dx2, dh2 = ext.ffn_train_dgrad(dy, torch.ones_like(hd), w2t_img, w1t_img, p)
     e2 = (dh2.float() - dhd_ref).abs()
     print("unmasked dhd err max/mean:", e2.max().item(), e2.mean().item())
+    n2 = torch.isnan(dh2.float())
+    print("dh2 NaN count:", int(n2.sum()), "of", dh2.numel())
+    if n2.any():
+        locs = n2.nonzero()[:6].tolist()
+        print("NaN locs:", locs)
+    # sample values at a known-bad column
+    print("dh2[0,18], ref:", dh2.float()[0, 18].item(), dhd_ref[0, 18].item())
+    print("dh2[0,50], ref:", dh2.float()[0, 50].item(), dhd_ref[0, 50].item())
+    print("dh2[0,17], ref:", dh2.float()[0, 17].item(), dhd_ref[0, 17].item())
+    # Probe 1: zero B1 weights -> dh must be exactly 0.
+    _, dh0 = ext.ffn_train_dgrad(dy, torch.ones_like(hd),
+                                 torch.zeros_like(w2t_img), w1t_img, p)
+    z = dh0.float()
+    print("zero-w2t: dh nonzero:", int((z != 0).sum()), "NaN:",
+          int(torch.isnan(z).sum()))
+    if (z != 0).any():
+        bl = (z != 0).nonzero()[:6].tolist()
+        print("  nonzero locs:", bl, "vals:",
+              [z[tuple(i)].item() for i in bl])
+    # Probe 2: zero B2 weights -> dx must be exactly 0 (dh unaffected).
+    dxz, _ = ext.ffn_train_dgrad(dy, torch.ones_like(hd), w2t_img,
+                                 torch.zeros_like(w1t_img), p)
+    zz = dxz.float()
+    print("zero-w1t: dx nonzero:", int((zz != 0).sum()), "NaN:",
+          int(torch.isnan(zz).sum()))
+    # Probe 3: is it deterministic?
+    dx3, dh3 = ext.ffn_train_dgrad(dy, torch.ones_like(hd), w2t_img,
+                                   w1t_img, p)
+    print("dgrad deterministic:", bool(torch.equal(dh2, dh3)),
+          bool(torch.equal(dx2, dx3)))
 
 
 if __name__ == "__main__":
