@@ -231,6 +231,10 @@ std::vector<at::Tensor> ffn_train_fwd(at::Tensor x, at::Tensor w1,
 std::vector<at::Tensor> ffn_train_dgrad(at::Tensor dy, at::Tensor hd,
                                         at::Tensor w2t, at::Tensor w1t,
                                         double p_drop);
+std::vector<at::Tensor> ffn_train_dgrad_nomask(at::Tensor dy, at::Tensor hd,
+                                               at::Tensor w2t,
+                                               at::Tensor w1t,
+                                               double p_drop);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_condense", &fused_condense,
@@ -242,6 +246,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ffn_train_dgrad", &ffn_train_dgrad,
         "Training FFN dgrad: both transposed GEMMs fused, mask from hd>0, "
         "returns (dx, dh_pre)");
+  m.def("ffn_train_dgrad_nomask", &ffn_train_dgrad_nomask,
+        "Bisection probe: dgrad without the mask (dh == dy@W2^T)");
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
         "Fused final LayerNorm + head + softmax + QV (K10+K11+K12)");
   m.def("banded_attn", &banded_attn,

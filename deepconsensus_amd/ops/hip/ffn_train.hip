@@ -97,7 +97,8 @@ __device__ __forceinline__ float rand01(unsigned seed, unsigned m,
   return (float)(z >> 8) * (1.0f / 16777216.0f);
 }
 
-// MODE 0 = forward (relu + dropout, +b2), 1 = dgrad (mask from hread).
+// MODE 0 = forward (relu + dropout, +b2), 1 = dgrad (mask from hread),
+// 2 = dgrad WITHOUT the mask (bisection probe: dh == plain dy@W2^T).
 template <int MODE>
 __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w1,
@@ -370,9 +371,26 @@ std::vector<at::Tensor> ffn_train_fwd(at::Tensor x, at::Tensor w1,
   return {y, hd};
 }
 
+std::vector<at::Tensor> ffn_train_dgrad_impl(at::Tensor dy, at::Tensor hd,
+                                             at::Tensor w2t, at::Tensor w1t,
+                                             double p_drop, bool nomask);
+
 std::vector<at::Tensor> ffn_train_dgrad(at::Tensor dy, at::Tensor hd,
                                         at::Tensor w2t, at::Tensor w1t,
                                         double p_drop) {
+  return ffn_train_dgrad_impl(dy, hd, w2t, w1t, p_drop, false);
+}
+
+std::vector<at::Tensor> ffn_train_dgrad_nomask(at::Tensor dy, at::Tensor hd,
+                                               at::Tensor w2t,
+                                               at::Tensor w1t,
+                                               double p_drop) {
+  return ffn_train_dgrad_impl(dy, hd, w2t, w1t, p_drop, true);
+}
+
+std::vector<at::Tensor> ffn_train_dgrad_impl(at::Tensor dy, at::Tensor hd,
+                                             at::Tensor w2t, at::Tensor w1t,
+                                             double p_drop, bool nomask) {
   TORCH_CHECK(dy.is_cuda() && dy.dtype() == at::kBFloat16,
               "dy must be bf16 on device");
   auto dyc = dy.contiguous();
@@ -394,14 +412,19 @@ std::vector<at::Tensor> ffn_train_dgrad(at::Tensor dy, at::Tensor hd,
   const float p = (float)p_drop;
   dim3 grid((M + BM - 1) / BM);
   hipStream_t stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(ffn_train_kernel<1>, grid, dim3(512), 0, stream,
-                     reinterpret_cast<bf16*>(dyc.data_ptr()),
-                     reinterpret_cast<bf16*>(w2t.data_ptr()),
-                     reinterpret_cast<bf16*>(w1t.data_ptr()), nullptr,
-                     reinterpret_cast<bf16*>(dx.data_ptr()),
-                     reinterpret_cast<bf16*>(dh.data_ptr()),
-                     reinterpret_cast<bf16*>(hdc.data_ptr()), M, p,
-                     p < 1.f ? 1.f / (1.f - p) : 0.f, 0ull);
+  auto launch = [&](auto mode_t) {
+    hipLaunchKernelGGL((ffn_train_kernel<decltype(mode_t)::value>), grid,
+                       dim3(512), 0, stream,
+                       reinterpret_cast<bf16*>(dyc.data_ptr()),
+                       reinterpret_cast<bf16*>(w2t.data_ptr()),
+                       reinterpret_cast<bf16*>(w1t.data_ptr()), nullptr,
+                       reinterpret_cast<bf16*>(dx.data_ptr()),
+                       reinterpret_cast<bf16*>(dh.data_ptr()),
+                       reinterpret_cast<bf16*>(hdc.data_ptr()), M, p,
+                       p < 1.f ? 1.f / (1.f - p) : 0.f, 0ull);
+  };
+  if (nomask) launch(std::integral_constant<int, 2>{});
+  else launch(std::integral_constant<int, 1>{});
   return {dx, dh};
 }
 

@@ -83,6 +83,14 @@ def main():
     zz = dxz.float()
     print("zero-w1t: dx nonzero:", int((zz != 0).sum()), "NaN:",
           int(torch.isnan(zz).sum()))
+    # Probe 2b: MODE 2 (no mask at all) vs plain dhd.
+    dx4, dh4 = ext.ffn_train_dgrad_nomask(dy, hd, w2t_img, w1t_img, p)
+    e4 = (dh4.float() - dhd_ref).abs()
+    print("MODE2 nomask dh err max/mean:", e4.max().item(), e4.mean().item(),
+          "NaN:", int(torch.isnan(dh4.float()).sum()))
+    dx4b, dh4b = ext.ffn_train_dgrad_nomask(dy, hd, w2t_img, w1t_img, p)
+    print("MODE2 deterministic:", bool(torch.equal(dh4, dh4b)),
+          bool(torch.equal(dx4, dx4b)))
     # Probe 3: is it deterministic?
     dx3, dh3 = ext.ffn_train_dgrad(dy, torch.ones_like(hd), w2t_img,
                                    w1t_img, p)
