@@ -98,5 +98,53 @@ def main():
           bool(torch.equal(dx2, dx3)))
 
 
+def probes2():
+    """Decisive probes: same images as the WORKING forward, template 2."""
+    ext = dc_ops.get_ext(required=True)
+    torch.manual_seed(4)
+    bf16 = torch.bfloat16
+    M = 512
+    x = (torch.randn(M, 280, device="cuda") * 0.5).to(bf16)
+    w1 = torch.randn(2048, 280, device="cuda") * 0.05
+    b1 = torch.randn(2048, device="cuda") * 0.1
+    w2 = torch.randn(280, 2048, device="cuda") * 0.02
+    w1_img = torch.zeros(2048, 296, dtype=bf16, device="cuda")
+    w1_img[:, :280] = w1.to(bf16)
+    w1_img[:, 287] = b1.to(bf16)
+    w2_img = torch.zeros(320, 2048, dtype=bf16, device="cuda")
+    w2_img[:280] = w2.to(bf16)
+    ones = torch.ones(M, 2048, dtype=bf16, device="cuda")
+    # A: MODE 2 with the EXACT images the working forward uses.
+    _, dhA = ext.ffn_train_dgrad_nomask(x, ones, w1_img, w2_img, 0.0)
+    refA = x.float() @ w1.t() + b1
+    eA = (dhA.float() - refA).abs()
+    print("A(M512, fwd-images, MODE2) err max:", eA.max().item(),
+          "NaN:", int(torch.isnan(dhA.float()).sum()))
+    _, dhA2 = ext.ffn_train_dgrad_nomask(x, ones, w1_img, w2_img, 0.0)
+    print("A deterministic:", bool(torch.equal(dhA, dhA2)))
+    # B: single block (M=256).
+    _, dhB = ext.ffn_train_dgrad_nomask(x[:256], ones[:256], w1_img,
+                                        w2_img, 0.0)
+    eB = (dhB.float() - refA[:256]).abs()
+    print("B(M256) err max:", eB.max().item(),
+          "NaN:", int(torch.isnan(dhB.float()).sum()))
+    # C: MODE 0 with the dgrad transpose images (relu'd, but checks
+    # whether the IMAGE data triggers anything in the working template).
+    w2t_img = torch.zeros(2048, 296, dtype=bf16, device="cuda")
+    w2t_img[:, :280] = w2.to(bf16).t()
+    w1t_img = torch.zeros(320, 2048, dtype=bf16, device="cuda")
+    w1t_img[:280] = w1.to(bf16).t()
+    zb = torch.zeros(280, device="cuda")
+    yC, hC = ext.ffn_train_fwd(x, w2t_img, w1t_img, zb, 0.0, 5)
+    refC = torch.relu(x.float() @ w2)
+    eC = (hC.float() - refC).abs()
+    print("C(MODE0, dgrad-images) err max:", eC.max().item(),
+          "NaN:", int(torch.isnan(hC.float()).sum()))
+    yC2, hC2 = ext.ffn_train_fwd(x, w2t_img, w1t_img, zb, 0.0, 5)
+    print("C deterministic:", bool(torch.equal(hC, hC2)))
+
+
 if __name__ == "__main__":
     main()
+    print("---- probes2 ----")
+    probes2()
