@@ -148,17 +148,27 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
     }
   }
 
+  // The glds LDS destination is wave-uniform and reaches the hardware
+  // through M0. Force it into an SGPR with an explicit readfirstlane:
+  // under register pressure (the dgrad instantiations) the allocator
+  // otherwise materializes it in a VGPR, SPILLS it, and rebuilds M0 from
+  // a scratch reload under a partial exec mask — when the reloading lane
+  // differs from the spilling lane, M0 is garbage and the weight DMA
+  // deposits at arbitrary LDS offsets (measured: nondeterministic NaN
+  // corruption in MODE 1/2 while MODE 0 happened to allocate cleanly).
   auto issue_w1 = [&](int chunk, int buf) {
     const bf16* src = w1 + (size_t)chunk * W1_ELEMS;
-    bf16* dst = &smem[OFF_W1 + buf * W1_ELEMS];
 #pragma unroll
     for (int i = 0; i < 5; ++i) {
       const int ck = wave + i * 8;
-      if (ck < W1_CHUNKS) glds16(src + ck * 512 + lane * 8, dst + ck * 512);
+      if (ck < W1_CHUNKS) {
+        const int dst_off = __builtin_amdgcn_readfirstlane(
+            OFF_W1 + buf * W1_ELEMS + ck * 512);
+        glds16(src + ck * 512 + lane * 8, &smem[dst_off]);
+      }
     }
   };
   auto issue_w2 = [&](int chunk, int buf) {
-    bf16* dst0 = &smem[OFF_W2 + buf * W2_ELEMS];
 #pragma unroll
     for (int i = 0; i < 6; ++i) {
       const int ck = wave + i * 8;
@@ -166,8 +176,10 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
       if (g < W2_GRAN) {
         const int row = g / 9, sub = g % 9;
         const int k8 = sub == 8 ? 0 : sub;
+        const int dst_off = __builtin_amdgcn_readfirstlane(
+            OFF_W2 + buf * W2_ELEMS + ck * 512);
         glds16(w2 + (size_t)row * NHID + chunk * NC + 8 * k8,
-               dst0 + ck * 512);
+               &smem[dst_off]);
       }
     }
   };
@@ -176,6 +188,16 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
   issue_w2(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
+
+  // Wave-uniform hd bases (m0, wave are SGPRs): the compiler keeps these
+  // in scalar registers, so only a 32-bit per-lane offset is ever live
+  // in the chunk loop — the 64-bit per-lane base this replaces was what
+  // the allocator spilled (and in the dgrad instantiation its reloads
+  // fed nondeterministic corruption; see profiles/r02_perf_journal.md).
+  bf16* const hsave_w = hsave + (size_t)(m0 + 32 * wave) * NHID;
+  const bf16* const hread_w =
+      (MODE == 1) ? hread + (size_t)(m0 + 32 * wave) * NHID : nullptr;
+  const int mrem = M - m0 - 32 * wave;  // lane c valid iff c < mrem
 
   f32x16 oacc[9] = {};
 
@@ -254,22 +276,30 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
       // that drain the weight DMA queue (the v3 de-spill lesson).
       if (MODE == 1) {
         const int ln2 = lane_recompute();
-        const int mrow = m0 + 32 * wave + (ln2 & 31);
-        if (mrow < M) {
-          const unsigned hoff = (unsigned)mrow * NHID + 64 * chunk
+        if ((ln2 & 31) < mrem) {
+          const unsigned hoff = (unsigned)(ln2 & 31) * NHID + 64 * chunk
                                 + 32 * t + 8 * (ln2 >> 5);
-          // dh_pre = dhd * (hd>0 ? 1/(1-p) : 0), applied post-repack.
-          const bf16x8 h0 = *reinterpret_cast<const bf16x8*>(hread + hoff);
-          const bf16x8 h1 =
-              *reinterpret_cast<const bf16x8*>(hread + hoff + 16);
+          // dh_pre = dhd * (hd>0 ? 1/(1-p) : 0), applied post-repack;
+          // one h fragment live at a time.
+          {
+            const bf16x8 h0 =
+                *reinterpret_cast<const bf16x8*>(hread_w + hoff);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            pa[0][j] = (float)h0[j] > 0.f
-                           ? (__bf16)((float)pa[0][j] * inv_keep)
-                           : (__bf16)0.f;
-            pa[1][j] = (float)h1[j] > 0.f
-                           ? (__bf16)((float)pa[1][j] * inv_keep)
-                           : (__bf16)0.f;
+            for (int j = 0; j < 8; ++j) {
+              pa[0][j] = (float)h0[j] > 0.f
+                             ? (__bf16)((float)pa[0][j] * inv_keep)
+                             : (__bf16)0.f;
+            }
+          }
+          {
+            const bf16x8 h1 =
+                *reinterpret_cast<const bf16x8*>(hread_w + hoff + 16);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              pa[1][j] = (float)h1[j] > 0.f
+                             ? (__bf16)((float)pa[1][j] * inv_keep)
+                             : (__bf16)0.f;
+            }
           }
         } else {
 #pragma unroll
@@ -298,12 +328,11 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
       // anyway; storing here keeps the address math out of their span).
       {
         const int ln2 = lane_recompute();
-        const int mrow = m0 + 32 * wave + (ln2 & 31);
-        if (mrow < M) {
-          const unsigned hoff = (unsigned)mrow * NHID + 64 * chunk
+        if ((ln2 & 31) < mrem) {
+          const unsigned hoff = (unsigned)(ln2 & 31) * NHID + 64 * chunk
                                 + 32 * t + 8 * (ln2 >> 5);
-          *reinterpret_cast<bf16x8*>(hsave + hoff) = pa[0];
-          *reinterpret_cast<bf16x8*>(hsave + hoff + 16) = pa[1];
+          *reinterpret_cast<bf16x8*>(hsave_w + hoff) = pa[0];
+          *reinterpret_cast<bf16x8*>(hsave_w + hoff + 16) = pa[1];
         }
       }
     }
