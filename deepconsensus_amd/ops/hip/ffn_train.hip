@@ -156,7 +156,11 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
   // differs from the spilling lane, M0 is garbage and the weight DMA
   // deposits at arbitrary LDS offsets (measured: nondeterministic NaN
   // corruption in MODE 1/2 while MODE 0 happened to allocate cleanly).
+  // Lane id is re-derived (volatile v_mbcnt) inside each lambda and in
+  // the epilogue: the kernel-top `lane`/`c`/`hi` otherwise stay live
+  // across the whole chunk loop and are what the allocator spills.
   auto issue_w1 = [&](int chunk, int buf) {
+    const int ln = lane_recompute();
     const bf16* src = w1 + (size_t)chunk * W1_ELEMS;
 #pragma unroll
     for (int i = 0; i < 5; ++i) {
@@ -164,15 +168,16 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
       if (ck < W1_CHUNKS) {
         const int dst_off = __builtin_amdgcn_readfirstlane(
             OFF_W1 + buf * W1_ELEMS + ck * 512);
-        glds16(src + ck * 512 + lane * 8, &smem[dst_off]);
+        glds16(src + ck * 512 + ln * 8, &smem[dst_off]);
       }
     }
   };
   auto issue_w2 = [&](int chunk, int buf) {
+    const int ln = lane_recompute();
 #pragma unroll
     for (int i = 0; i < 6; ++i) {
       const int ck = wave + i * 8;
-      const int g = ck * 64 + lane;
+      const int g = ck * 64 + ln;
       if (g < W2_GRAN) {
         const int row = g / 9, sub = g % 9;
         const int k8 = sub == 8 ? 0 : sub;
@@ -226,11 +231,13 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
                                                       0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_sched_barrier(0);  // fence: keep B1 state local
 
       // Intermediate element (m, hidden) for lane (c, hi) at index r:
       // m = m0 + (r&3) + 8*(r>>2) + 4*hi + 32*wave (the af B-operand
-      // rows), hidden = 64*chunk + 32*t + c.
-      float st[16];
+      // rows), hidden = 64*chunk + 32*t + c. relu/dropout applied
+      // IN-PLACE on acc (a separate st[16] doubled the live range by 16
+      // registers — the difference between spilling and not).
       if (MODE == 0) {
         const int ln2 = lane_recompute();
         const unsigned hidx = 64 * chunk + 32 * t + (ln2 & 31);
@@ -243,19 +250,16 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
             v = rand01((unsigned)seed, mrow, hidx) < p ? 0.f
                                                        : v * inv_keep;
           }
-          st[r] = v;
+          acc[r] = v;
         }
-      } else {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) st[r] = acc[r];
       }
       bf16x8 pa[2];
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        const unsigned x0 = cvt_pk_bf16(st[8 * s + 0], st[8 * s + 1]);
-        const unsigned y0 = cvt_pk_bf16(st[8 * s + 2], st[8 * s + 3]);
-        const unsigned x1 = cvt_pk_bf16(st[8 * s + 4], st[8 * s + 5]);
-        const unsigned y1 = cvt_pk_bf16(st[8 * s + 6], st[8 * s + 7]);
+        const unsigned x0 = cvt_pk_bf16(acc[8 * s + 0], acc[8 * s + 1]);
+        const unsigned y0 = cvt_pk_bf16(acc[8 * s + 2], acc[8 * s + 3]);
+        const unsigned x1 = cvt_pk_bf16(acc[8 * s + 4], acc[8 * s + 5]);
+        const unsigned y1 = cvt_pk_bf16(acc[8 * s + 6], acc[8 * s + 7]);
         const auto rx = __builtin_amdgcn_permlane32_swap(x0, x1, false,
                                                          false);
         const auto ry = __builtin_amdgcn_permlane32_swap(y0, y1, false,
@@ -310,6 +314,21 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
         }
       }
 
+      __builtin_amdgcn_sched_barrier(0);  // fence: mask/repack local
+      // Persist the intermediate before B2 (pa dies at its last B2
+      // use instead of staying live across all 18 MFMAs for a post-B2
+      // store — the difference between 0 and ~8 spilled VGPRs here).
+      {
+        const int ln2 = lane_recompute();
+        if ((ln2 & 31) < mrem) {
+          const unsigned hoff = (unsigned)(ln2 & 31) * NHID + 64 * chunk
+                                + 32 * t + 8 * (ln2 >> 5);
+          *reinterpret_cast<bf16x8*>(hsave_w + hoff) = pa[0];
+          *reinterpret_cast<bf16x8*>(hsave_w + hoff + 16) = pa[1];
+        }
+      }
+
+      __builtin_amdgcn_sched_barrier(0);  // fence: store addrs die here
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
@@ -323,18 +342,6 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
         }
       }
       __builtin_amdgcn_s_setprio(0);
-
-      // Persist the intermediate after B2 (pa is live through the MFMAs
-      // anyway; storing here keeps the address math out of their span).
-      {
-        const int ln2 = lane_recompute();
-        if ((ln2 & 31) < mrem) {
-          const unsigned hoff = (unsigned)(ln2 & 31) * NHID + 64 * chunk
-                                + 32 * t + 8 * (ln2 >> 5);
-          *reinterpret_cast<bf16x8*>(hsave_w + hoff) = pa[0];
-          *reinterpret_cast<bf16x8*>(hsave_w + hoff + 16) = pa[1];
-        }
-      }
     }
 
     if (more) {
@@ -345,14 +352,17 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
 
   // Epilogue: plain y (+b2 in FWD); the ReZero residual/post-dropout
   // stay in the wrapper (they are part of its autograd graph).
+  const int lne = lane_recompute();
+  const int ce = lne & 31;
+  const int hie = lne >> 5;
 #pragma unroll
   for (int ct = 0; ct < 9; ++ct) {
-    const int col = 32 * ct + c;
+    const int col = 32 * ct + ce;
     if (col >= NOUT) continue;
     const float bias = (MODE == 0) ? b2[col] : 0.f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * wave;
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hie + 32 * wave;
       if (m0 + row < M) {
         out[(size_t)(m0 + row) * K1 + col] =
             __float2bfloat16(oacc[ct][r] + bias);
