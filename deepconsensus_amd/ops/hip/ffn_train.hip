@@ -191,7 +191,14 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
 
   issue_w1(0, 0);
   issue_w2(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  // lgkmcnt(0) is load-bearing: the af ds_reads issued in the staging
+  // pass are lgkm-tracked, and a raw s_barrier does NOT drain counters
+  // (__syncthreads would). Without it a wave can pass the barrier with
+  // af reads still in flight while another wave's chunk-1 glds
+  // overwrites the staging region (W2[1]) — a nondeterministic af
+  // corruption measured in the dgrad instantiation (a handful of NaN
+  // elements per million, schedule-dependent).
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   // Wave-uniform hd bases (m0, wave are SGPRs): the compiler keeps these

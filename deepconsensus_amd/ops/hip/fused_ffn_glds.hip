@@ -129,7 +129,7 @@ __global__ __launch_bounds__(512, 2) void fused_ffn_v2_kernel(
   };
 
   issue_w1(0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   f32x16 oacc[5] = {};

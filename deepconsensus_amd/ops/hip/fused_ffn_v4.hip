@@ -187,7 +187,7 @@ __global__ __launch_bounds__(512, 1) void fused_ffn_v4_kernel(
   // W2(1) below, which lands behind this barrier + the loop's waits.
   issue_w1(0, 0);
   issue_w2(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   f32x16 oacc[9] = {};
