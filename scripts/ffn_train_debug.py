@@ -131,12 +131,18 @@ def probes2():
         locs = bad.nonzero()
         print(f"B run{it}: bad={int(bad.sum())} NaN="
               f"{int(torch.isnan(dhB.float()).sum())}")
-        for r, cidx in locs[:8].tolist():
-            v = dhB[r, cidx].view(torch.int16).item() & 0xFFFF
-            print(f"   (m={r}, h={cidx}) chunk={cidx//64} t={(cidx%64)//32}"
-                  f" c={cidx%32} bits=0x{v:04x}"
-                  f" got={dhB.float()[r,cidx].item():.4g}"
-                  f" want={refA[r,cidx].item():.4g}")
+        if it == 0:
+            for r, cidx in locs[:6].tolist():
+                got = dhB.float()[r, cidx].item()
+                # which (m, h) does this value belong to?
+                dm = (refA[:256, cidx].float() - got).abs()
+                bm = int(dm.argmin())
+                dh_ = (refA[r].float() - got).abs()
+                bh = int(dh_.argmin())
+                print(f"   (m={r},h={cidx}) got={got:.4g} "
+                      f"want={refA[r,cidx].item():.4g} | best m={bm} "
+                      f"(d={dm[bm].item():.2g}) best h={bh} "
+                      f"(d={dh_[bh].item():.2g})")
     # C: MODE 0 with the dgrad transpose images (relu'd, but checks
     # whether the IMAGE data triggers anything in the working template).
     w2t_img = torch.zeros(2048, 296, dtype=bf16, device="cuda")
