@@ -239,6 +239,15 @@ __global__ __launch_bounds__(512, 1) void ffn_train_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
       __builtin_amdgcn_sched_barrier(0);  // fence: keep B1 state local
+      // MFMA D-register read hazard: the cvt_pk inline asm below reads
+      // acc; in MODE 0 the ~48 relu VALU ops provide the required
+      // post-MFMA delay naturally, but MODE 1/2 read acc immediately
+      // and measured deterministic corruption in 3 of 4 lanes (the
+      // compiler's hazard recognizer does not cover asm operands
+      // against the 16-pass MFMA writeback). Two s_nop 15 = 32 cycles.
+      if (MODE != 0) {
+        asm volatile("s_nop 15\n\ts_nop 15" ::: "memory");
+      }
 
       // Intermediate element (m, hidden) for lane (c, hi) at index r:
       // m = m0 + (r&3) + 8*(r>>2) + 4*hi + 32*wave (the af B-operand
