@@ -108,12 +108,67 @@ def _forward_blocks(rows_f: torch.Tensor, meta: EmbedMeta, tables):
     return torch.cat(parts, dim=-1)
 
 
+def _gather_maps(meta: EmbedMeta, model, device):
+    """Shape-only chunk maps for the embed_gather kernel (cached) — the
+    table CONTENTS are rebuilt per call in _device_table_flat."""
+    if getattr(meta, "_gmaps", None) is None:
+        from deepconsensus_amd.models.runner import build_gather_tables
+
+        tf, rs, rv, cc, ce = build_gather_tables(model)
+        meta._gmaps = {
+            "flat_len": tf.numel(),
+            "row_shift": rs,
+            "row_vocab": rv,
+            "chunk_cnt": cc,
+            "chunk_entries": ce,
+        }
+        meta._gmaps_dev = {}
+    key = str(device)
+    if key not in meta._gmaps_dev:
+        g = meta._gmaps
+        meta._gmaps_dev[key] = tuple(
+            g[k].to(device)
+            for k in ("row_shift", "row_vocab", "chunk_cnt",
+                      "chunk_entries")
+        )
+    return meta._gmaps_dev[key], meta._gmaps["flat_len"]
+
+
+def _device_table_flat(meta: EmbedMeta, tables, flat_len, device):
+    """table_flat rebuilt from the LIVE fp32 tables (sqrt-width scale,
+    id-0 row zeroed, 8-elem padding — the build_gather_tables layout)."""
+    parts = []
+    for ti, name in enumerate(meta.names):
+        t = tables[ti]
+        w = t.shape[1]
+        s = t.detach().float() * (float(w) ** 0.5)
+        s = torch.cat([torch.zeros_like(s[:1]), s[1:]], dim=0)
+        flat = s.to(torch.bfloat16).reshape(-1)
+        pad = (-flat.numel()) % 8
+        if pad:
+            flat = torch.cat([flat, flat.new_zeros(pad)])
+        parts.append(flat)
+    out = torch.cat(parts)
+    assert out.numel() == flat_len, (out.numel(), flat_len)
+    return out.to(device)
+
+
 class EmbedStackFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, rows_f: torch.Tensor, meta: EmbedMeta, *tables):
+    def forward(ctx, rows_f: torch.Tensor, meta: EmbedMeta, use_gather,
+                model_ref, *tables):
         ctx.meta = meta
         ctx.save_for_backward(rows_f)
         with torch.no_grad():
+            if use_gather:
+                from deepconsensus_amd import ops as dc_ops
+
+                ext = dc_ops.get_ext(required=True)
+                maps, flat_len = _gather_maps(meta, model_ref,
+                                              rows_f.device)
+                tf = _device_table_flat(meta, tables, flat_len,
+                                        rows_f.device)
+                return ext.embed_gather(rows_f.contiguous(), tf, *maps)
             return _forward_blocks(rows_f, meta, tables)
 
     @staticmethod
@@ -154,7 +209,34 @@ class EmbedStackFunction(torch.autograd.Function):
                 g = g.permute(0, 2, 1, 3).reshape(-1, w) * scale
                 mask = (ids != 0).reshape(-1, 1).to(g.dtype)
                 grads[ti].index_add_(0, ids.reshape(-1), g * mask)
-        return (None, None) + tuple(grads)
+        return (None, None, None, None) + tuple(grads)
+
+
+_EMBED_FWD_GATHER = None
+
+
+def _embed_fwd_gather_enabled() -> bool:
+    """HIP embed_gather as the training FORWARD too (DC_EMBED_FWD=0
+    falls back to the torch gather/concat chain). The table contents
+    are rebuilt on-device from the live fp32 tables each call (~10k
+    elements); the chunk maps are shape-only and cached."""
+    global _EMBED_FWD_GATHER
+    if _EMBED_FWD_GATHER is None:
+        import os
+
+        if os.environ.get("DC_EMBED_FWD", "1") == "0":
+            _EMBED_FWD_GATHER = False
+            return False
+        try:
+            from deepconsensus_amd import ops as dc_ops
+
+            ext = dc_ops.get_ext()
+            _EMBED_FWD_GATHER = bool(
+                ext is not None and hasattr(ext, "embed_gather")
+            )
+        except Exception:  # pragma: no cover
+            _EMBED_FWD_GATHER = False
+    return _EMBED_FWD_GATHER
 
 
 def embed_stack(model, rows_f: torch.Tensor) -> torch.Tensor:
@@ -163,4 +245,10 @@ def embed_stack(model, rows_f: torch.Tensor) -> torch.Tensor:
         model._embed_meta = EmbedMeta(model)
     meta = model._embed_meta
     tables = [getattr(model, a).table for a in meta.table_attrs]
-    return EmbedStackFunction.apply(rows_f, meta, *tables)
+    use_gather = (
+        rows_f.is_cuda
+        and torch.is_autocast_enabled()
+        and _embed_fwd_gather_enabled()
+    )
+    return EmbedStackFunction.apply(rows_f, meta, use_gather, model,
+                                    *tables)

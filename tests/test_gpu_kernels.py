@@ -1062,3 +1062,33 @@ def test_model_ffn_fused_training_path(monkeypatch):
     )
     assert e_out < 0.3, e_out
     assert e_g < 0.08, e_g
+
+
+def test_embed_stack_gather_forward_matches_torch(monkeypatch):
+    """Training embed forward via the embed_gather kernel == the torch
+    gather/concat chain (and grads flow through embed_grad unchanged)."""
+    import deepconsensus_amd.models.embed_stack as es
+
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=True)
+    torch.manual_seed(21)
+    model = get_model(params).cuda()
+    rows = _make_rows(params, B=8, seed=11).cuda()
+    monkeypatch.setattr(es, "_EMBED_FWD_GATHER", None)
+    monkeypatch.setenv("DC_EMBED_FWD", "1")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out_g = es.embed_stack(model, rows)
+    assert out_g.dtype == torch.bfloat16
+    loss = out_g.float().sum()
+    loss.backward()
+    g_tab = model.pw_embedding.table.grad.clone()
+    model.zero_grad()
+    monkeypatch.setattr(es, "_EMBED_FWD_GATHER", False)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out_t = es.embed_stack(model, rows)
+    out_t.float().sum().backward()
+    g_ref = model.pw_embedding.table.grad.clone()
+    assert (out_g.float() - out_t.float()).abs().max().item() < 0.05
+    assert (g_tab - g_ref).abs().max().item() < 1e-3 * (
+        g_ref.abs().max().item() + 1.0
+    )
