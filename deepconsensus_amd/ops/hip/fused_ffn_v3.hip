@@ -141,26 +141,38 @@ __global__ __launch_bounds__(512, 1) void fused_ffn_v3_kernel(
   // uniform: out-of-range chunks are simply exec-masked off (a partial
   // last chunk would otherwise deposit past the buffer — glds lane
   // deposits are unconditional at dst + 16*lane).
+  // Lane id re-derived per lambda and the glds LDS destination pinned
+  // to an SGPR via readfirstlane — hardening against two allocator
+  // hazards found in the training clone of this kernel (ffn_train.hip):
+  // kernel-top lane state spilled around the chunk loop, and a spilled
+  // M0 source rebuilding the DMA destination from scratch under a
+  // partial exec mask (= weight deposits at arbitrary LDS offsets).
   auto issue_w1 = [&](int chunk, int buf) {
+    const int ln = lane_recompute();
     const bf16* src = w1 + (size_t)chunk * W1_ELEMS;
-    bf16* dst = &smem[OFF_W1 + buf * W1_ELEMS];
 #pragma unroll
     for (int i = 0; i < 5; ++i) {
       const int ck = wave + i * 8;
-      if (ck < W1_CHUNKS) glds16(src + ck * 512 + lane * 8, dst + ck * 512);
+      if (ck < W1_CHUNKS) {
+        const int dst_off = __builtin_amdgcn_readfirstlane(
+            OFF_W1 + buf * W1_ELEMS + ck * 512);
+        glds16(src + ck * 512 + ln * 8, &smem[dst_off]);
+      }
     }
   };
   auto issue_w2 = [&](int chunk, int buf) {
-    bf16* dst0 = &smem[OFF_W2 + buf * W2_ELEMS];
+    const int ln = lane_recompute();
 #pragma unroll
     for (int i = 0; i < 6; ++i) {
       const int ck = wave + i * 8;
-      const int g = ck * 64 + lane;
+      const int g = ck * 64 + ln;
       if (g < W2_GRAN) {
         const int row = g / 9, sub = g % 9;
         const int k8 = sub == 8 ? 0 : sub;  // pad slot re-loads granule 0
+        const int dst_off = __builtin_amdgcn_readfirstlane(
+            OFF_W2 + buf * W2_ELEMS + ck * 512);
         glds16(w2 + (size_t)row * NHID + chunk * NC + 8 * k8,
-               dst0 + ck * 512);
+               &smem[dst_off]);
       }
     }
   };
@@ -255,14 +267,17 @@ __global__ __launch_bounds__(512, 1) void fused_ffn_v3_kernel(
   }
 
   // ---- Epilogue: b2 + ReZero alpha + residual (fully unrolled). ----
+  const int lne = lane_recompute();
+  const int ce = lne & 31;
+  const int hie = lne >> 5;
 #pragma unroll
   for (int ct = 0; ct < 9; ++ct) {
-    const int col = 32 * ct + c;
+    const int col = 32 * ct + ce;
     if (col >= NOUT) continue;
     const float bias = b2[col];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi + 32 * wave;
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hie + 32 * wave;
       if (m0 + row < M) {
         const size_t off = (size_t)(m0 + row) * K1 + col;
         const float resid = __bfloat162float(x[off]);
