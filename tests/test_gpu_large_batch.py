@@ -224,3 +224,47 @@ def test_forward_windows_large_batch_matches_torch():
         f"native vs torch base agreement {rate:.4f} at B={B_LARGE} "
         "(large-batch corruption)"
     )
+
+
+def test_train_step_bitwise_deterministic_large_batch():
+    """Fixed seeds => two identical fwd+bwd passes produce bitwise-equal
+    losses and grads through the fully-fused bf16 training path
+    (attention pair + ffn_train pair + embed_gather forward). This is
+    the tripwire class that caught the r1 embed_gather corruption and
+    the r2 dgrad allocator hazards."""
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models.model import get_model
+
+    params = cfg.get_config("transformer_learn_values+custom")
+    cfg.modify_params(params, is_training=True)
+
+    def one_pass():
+        torch.manual_seed(123)
+        model = get_model(params).cuda()
+        rows = torch.from_numpy(
+            _rows(params, 2048).astype(np.float32)
+        ).cuda()
+        labels = torch.randint(
+            0, 5, (2048, params.max_length), device="cuda"
+        )
+        torch.manual_seed(77)
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            probs = model(rows, training=True)
+            logp = torch.log(probs.float().clamp_min(1e-9))
+            loss = torch.nn.functional.nll_loss(
+                logp.reshape(-1, 5), labels.reshape(-1)
+            )
+        loss.backward()
+        grads = [
+            p.grad.detach().clone()
+            for p in model.parameters()
+            if p.grad is not None
+        ]
+        return loss.detach().clone(), grads
+
+    l1, g1 = one_pass()
+    l2, g2 = one_pass()
+    assert torch.equal(l1, l2), (l1.item(), l2.item())
+    assert len(g1) == len(g2) and len(g1) > 10
+    for a, b in zip(g1, g2):
+        assert torch.equal(a, b)
