@@ -155,7 +155,12 @@ def test_embed_grad_kernel_matches_cpu_fallback():
 
     def grads_on(device):
         tabs = [t.detach().to(device).requires_grad_(True) for t in tables]
-        out = EmbedStackFunction.apply(x_cpu.to(device), meta, *tabs)
+        # use_gather=False on both sides: this test pins the BACKWARD
+        # kernel against the CPU index_add fallback with an identical
+        # torch forward.
+        out = EmbedStackFunction.apply(
+            x_cpu.to(device), meta, False, model, *tabs
+        )
         out.square().sum().backward()
         return [t.grad.cpu() for t in tabs]
 
