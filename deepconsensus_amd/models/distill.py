@@ -85,6 +85,9 @@ def train_model(
     limit_steps: int = 0,
     use_bf16: bool = False,
 ) -> dict:
+    from deepconsensus_amd.utils.tuned_gemm import enable_tuned_gemms
+
+    enable_tuned_gemms()
     rank, world = comm.init_distributed()
     main = rank == 0
     os.makedirs(out_dir, exist_ok=True)

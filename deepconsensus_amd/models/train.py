@@ -53,6 +53,10 @@ def train_model(
     use_bf16: bool = False,
 ) -> dict:
     """Runs the custom training loop; returns summary metrics."""
+    from deepconsensus_amd.utils.tuned_gemm import enable_tuned_gemms
+
+    enable_tuned_gemms()
+
     rank, world = comm.init_distributed()
     main = rank == 0
     os.makedirs(out_dir, exist_ok=True)

@@ -25,6 +25,9 @@ from deepconsensus_amd.parallel import comm
 
 
 def main():
+    from deepconsensus_amd.utils.tuned_gemm import enable_tuned_gemms
+
+    enable_tuned_gemms()
     ap = argparse.ArgumentParser()
     ap.add_argument("--batch-size", type=int, default=256)
     ap.add_argument("--steps", type=int, default=30)

@@ -115,3 +115,15 @@ def test_train_writes_tensorboard_events(tmp_path):
     erecs = list(tfrecord.read_tfrecords(efiles[0]))
     assert len(erecs) >= 2  # version header + >=1 eval summary
     assert any(b"eval/per_example_accuracy" in r for r in erecs)
+
+
+def test_enable_tuned_gemms_cpu_noop(monkeypatch):
+    """On CPU the tuned-GEMM loader is a no-op returning False."""
+    import deepconsensus_amd.utils.tuned_gemm as tg
+
+    monkeypatch.setattr(tg, "_ENABLED", None)
+    assert tg.enable_tuned_gemms() is False
+    # Disabled via env as well.
+    monkeypatch.setattr(tg, "_ENABLED", None)
+    monkeypatch.setenv("DC_TUNED_GEMM", "0")
+    assert tg.enable_tuned_gemms() is False
