@@ -139,6 +139,60 @@ class _FFNTrainFused(torch.autograd.Function):
         return dx, dw1, db1, dw2, db2, None, None
 
 
+_RESID_DROP_AVAILABLE = None
+
+
+def _resid_drop_available() -> bool:
+    """Fused ReZero residual + post-sublayer dropout (resid_dropout.hip:
+    hash dropout, no mask tensor, dalpha via deterministic block
+    partials). DC_RESID_DROP=0 falls back to the torch chain."""
+    global _RESID_DROP_AVAILABLE
+    if _RESID_DROP_AVAILABLE is None:
+        import os
+
+        if os.environ.get("DC_RESID_DROP", "1") == "0":
+            _RESID_DROP_AVAILABLE = False
+            return False
+        try:
+            from deepconsensus_amd import ops as dc_ops
+
+            ext = dc_ops.get_ext()
+            _RESID_DROP_AVAILABLE = bool(
+                ext is not None and hasattr(ext, "resid_drop_fwd")
+            )
+        except Exception:  # pragma: no cover
+            _RESID_DROP_AVAILABLE = False
+    return _RESID_DROP_AVAILABLE
+
+
+class _ResidDropAdd(torch.autograd.Function):
+    """out = x + alpha * dropout(y): one fused elementwise kernel per
+    direction; the mask is recomputed in backward from (seed, index)."""
+
+    @staticmethod
+    def forward(ctx, x, y, alpha, p_drop, seed):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        a = float(alpha.detach())
+        out = ext.resid_drop_fwd(x, y, a, p_drop, seed)
+        ctx.save_for_backward(y)
+        ctx.meta = (a, p_drop, seed)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        (y,) = ctx.saved_tensors
+        a, p_drop, seed = ctx.meta
+        dc = dout.contiguous()
+        dy, part = ext.resid_drop_bwd(dc, y, a, p_drop, seed)
+        dalpha = part.sum()
+        return dc, dy, dalpha, None, None
+
+
 class _BandedAttnTrain(torch.autograd.Function):
     """HIP banded attention for the training path (K5-K7 on device):
     band-only compute + fused softmax/dropout, band-local backward
@@ -416,6 +470,20 @@ class SublayerWrapper(nn.Module):
     def post(
         self, x: torch.Tensor, y: torch.Tensor, training: bool
     ) -> torch.Tensor:
+        if (
+            training
+            and self.rezero
+            and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and y.dtype == torch.bfloat16
+            and x.numel() % 8 == 0
+            and _resid_drop_available()
+        ):
+            seed = int(torch.randint(0, 2 ** 31 - 1, ()).item())
+            p = float(self.post_dropout) if self.post_dropout > 0 else 0.0
+            return _ResidDropAdd.apply(
+                x.contiguous(), y.contiguous(), self.alpha, p, seed
+            )
         if training and self.post_dropout > 0:
             y = F.dropout(y, p=self.post_dropout, training=True)
         if self.rezero:

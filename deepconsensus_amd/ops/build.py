@@ -29,6 +29,7 @@ _SRCS = [
     os.path.join(_OPS_DIR, "hip", "banded_attn_bwd_mfma.hip"),
     os.path.join(_OPS_DIR, "hip", "fused_condense.hip"),
     os.path.join(_OPS_DIR, "hip", "ffn_train.hip"),
+    os.path.join(_OPS_DIR, "hip", "resid_dropout.hip"),
 ]
 EXT_NAME = "dc_hip_kernels"
 

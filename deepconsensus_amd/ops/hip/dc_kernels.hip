@@ -235,6 +235,11 @@ std::vector<at::Tensor> ffn_train_dgrad_nomask(at::Tensor dy, at::Tensor hd,
                                                at::Tensor w2t,
                                                at::Tensor w1t,
                                                double p_drop);
+at::Tensor resid_drop_fwd(at::Tensor x, at::Tensor y, double alpha,
+                          double p_drop, int64_t seed);
+std::vector<at::Tensor> resid_drop_bwd(at::Tensor dout, at::Tensor y,
+                                       double alpha, double p_drop,
+                                       int64_t seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_condense", &fused_condense,
@@ -248,6 +253,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "returns (dx, dh_pre)");
   m.def("ffn_train_dgrad_nomask", &ffn_train_dgrad_nomask,
         "Bisection probe: dgrad without the mask (dh == dy@W2^T)");
+  m.def("resid_drop_fwd", &resid_drop_fwd,
+        "Fused ReZero residual + hash-dropout forward");
+  m.def("resid_drop_bwd", &resid_drop_bwd,
+        "Fused residual-dropout backward: (dy, dalpha partials)");
   m.def("fused_ln_head_qv", &fused_ln_head_qv,
         "Fused final LayerNorm + head + softmax + QV (K10+K11+K12)");
   m.def("banded_attn", &banded_attn,
