@@ -1153,4 +1153,6 @@ def test_model_resid_drop_training_path(monkeypatch):
     out = w.post(x, y.clone().requires_grad_(True), training=True)
     monkeypatch.setattr(mm, "_RESID_DROP_AVAILABLE", False)
     ref = w.post(x, y.clone().requires_grad_(True), training=True)
-    assert (out.float() - ref.float()).abs().max().item() < 0.02
+    # The torch chain rounds alpha*y to bf16 before the add (double
+    # rounding); the kernel accumulates in fp32 and rounds once.
+    assert (out.float() - ref.float()).abs().max().item() < 0.06
