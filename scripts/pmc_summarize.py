@@ -14,7 +14,8 @@ try:
                      and "name" in c.lower()), None)
     cnt_cands = [c for c in cols if "counter" in c.lower()
                  or c.lower() in ("name", "symbol")]
-    val_col = next((c for c in cols if "value" in c.lower()), None)
+    val_col = "value" if "value" in cols else next(
+        (c for c in cols if "value" in c.lower()), None)
     print("# picked:", name_col, cnt_cands, val_col)
     if name_col is None:
         # fall back: find a text column containing kernel-looking strings
