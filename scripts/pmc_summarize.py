@@ -54,7 +54,8 @@ try:
         act = cs.get("SQ_ACTIVE_INST_ANY", 0)
         mfma = cs.get("SQ_INSTS_MFMA", 0)
         lds = cs.get("SQ_LDS_BANK_CONFLICT", 0)
-        print(f"{kn.split('(')[0][:58]}")
+        short = kn.replace('(anonymous namespace)::', '')
+        print(short.split('(')[0][:58] or short[:58])
         print(f"   wait/wave={wait/wc:.2f} active/wave={act/wc:.2f} "
               f"conflict/active={lds/max(act,1):.3f} "
               f"mfma_insts={mfma:.3g}")
