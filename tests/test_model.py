@@ -199,3 +199,35 @@ def test_embed_stack_fused_backward_matches_autograd():
             g1 = getattr(m1, f"{n}_embedding").table.grad
             g2 = getattr(m2, f"{n}_embedding").table.grad
             torch.testing.assert_close(g1, g2, atol=1e-4, rtol=1e-4)
+
+
+def test_device_table_flat_matches_build_gather_tables():
+    """embed_stack's per-step table rebuild reproduces the canonical
+    build_gather_tables layout bit-for-bit (CPU check — guards layout
+    drift between the training forward and the serving tables)."""
+    import torch
+
+    from deepconsensus_amd.models import config as cfg
+    from deepconsensus_amd.models.embed_stack import (
+        EmbedMeta,
+        _device_table_flat,
+        _gather_maps,
+    )
+    from deepconsensus_amd.models.model import get_model
+
+    for use_bq in (False, True):
+        params = cfg.get_config("transformer_learn_values+custom")
+        params.use_ccs_bq = use_bq
+        cfg.modify_params(params, is_training=True)
+        torch.manual_seed(3)
+        model = get_model(params)
+        meta = EmbedMeta(model)
+        tables = [
+            getattr(model, a).table for a in meta.table_attrs
+        ]
+        from deepconsensus_amd.models.runner import build_gather_tables
+
+        tf_ref, *_ = build_gather_tables(model)
+        _maps, flat_len = _gather_maps(meta, model, "cpu")
+        tf_new = _device_table_flat(meta, tables, flat_len, "cpu")
+        assert torch.equal(tf_ref, tf_new)
