@@ -193,6 +193,101 @@ class _ResidDropAdd(torch.autograd.Function):
         return dc, dy, dalpha, None, None
 
 
+_FLIN_TRAIN_AVAILABLE = None
+
+
+def _flin_train_available() -> bool:
+    """Serving fused_linear kernel as the training QKV/out-proj GEMM
+    (fwd, and dgrad for the 280-wide out-proj). DC_FLIN_TRAIN=0 falls
+    back to torch Linear."""
+    global _FLIN_TRAIN_AVAILABLE
+    if _FLIN_TRAIN_AVAILABLE is None:
+        import os
+
+        if os.environ.get("DC_FLIN_TRAIN", "1") == "0":
+            _FLIN_TRAIN_AVAILABLE = False
+            return False
+        try:
+            from deepconsensus_amd import ops as dc_ops
+
+            ext = dc_ops.get_ext()
+            _FLIN_TRAIN_AVAILABLE = bool(
+                ext is not None and hasattr(ext, "fused_linear")
+            )
+        except Exception:  # pragma: no cover
+            _FLIN_TRAIN_AVAILABLE = False
+    return _FLIN_TRAIN_AVAILABLE
+
+
+def _flin_img(w: torch.Tensor, npad: int) -> torch.Tensor:
+    """fused_linear weight image: [npad, 296] bf16, rows = output cols
+    (the kernel's glds-streamable LDS row layout; fused_linear.hip)."""
+    img = torch.zeros(npad, 296, dtype=torch.bfloat16, device=w.device)
+    img[: w.shape[0], : w.shape[1]] = w.detach().to(torch.bfloat16)
+    return img
+
+
+class _QkvProjFused(torch.autograd.Function):
+    """Training QKV projection on the serving fused_linear kernel
+    (K=280 MFMA GEMM; the library ran this tall-skinny shape at <10% of
+    peak). wgrad stays on the TunableOp-selected hipBLASLt split-K;
+    dgrad (K=840) stays torch."""
+
+    @staticmethod
+    def forward(ctx, x, w_qkv):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        xb = x.to(torch.bfloat16).contiguous()
+        empty = xb.new_empty(0)
+        img = _flin_img(w_qkv, 896)
+        y = ext.fused_linear(xb, img, empty, empty, 840, False, 0.0)
+        ctx.save_for_backward(xb, w_qkv)
+        ctx.x_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        xb, w_qkv = ctx.saved_tensors
+        dyc = dy.contiguous()
+        dx = (dyc @ w_qkv.to(torch.bfloat16)).to(ctx.x_dtype)
+        dw = (dyc.t() @ xb).to(w_qkv.dtype)
+        return dx, dw
+
+
+class _OutProjFused(torch.autograd.Function):
+    """Training output projection (280->280, no bias): forward AND
+    dgrad on fused_linear (both are K=280); wgrad on hipBLASLt."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        xb = x.to(torch.bfloat16).contiguous()
+        empty = xb.new_empty(0)
+        y = ext.fused_linear(xb, _flin_img(w, 320), empty, empty, 280,
+                             False, 0.0)
+        ctx.save_for_backward(xb, w)
+        ctx.x_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from deepconsensus_amd import ops as dc_ops
+
+        ext = dc_ops.get_ext(required=True)
+        xb, w = ctx.saved_tensors
+        dyc = dy.contiguous()
+        empty = dyc.new_empty(0)
+        # dx = dy @ W  ==  fused_linear with the transposed image.
+        wT = w.detach().to(torch.bfloat16).t().contiguous()
+        dx = ext.fused_linear(dyc, _flin_img(wT, 320), empty, empty,
+                              280, False, 0.0).to(ctx.x_dtype)
+        dw = (dyc.t() @ xb).to(w.dtype)
+        return dx, dw
+
+
 class _BandedAttnTrain(torch.autograd.Function):
     """HIP banded attention for the training path (K5-K7 on device):
     band-only compute + fused softmax/dropout, band-local backward
@@ -368,7 +463,12 @@ class BandedSelfAttention(nn.Module):
                 [self.q_proj.weight, self.k_proj.weight,
                  self.v_proj.weight], 0
             )
-            qkv = F.linear(x, w_qkv)  # [B, T, 3*H*D], bf16 (autocast)
+            if self.hidden_size == 280 and _flin_train_available():
+                qkv = _QkvProjFused.apply(
+                    x.reshape(b * t, -1), w_qkv
+                ).view(b, t, -1)
+            else:
+                qkv = F.linear(x, w_qkv)  # [B, T, 3*H*D] (autocast)
             drop_mask = None
             p_drop = float(self.dropout) if training else 0.0
             if p_drop > 0:
@@ -382,6 +482,11 @@ class BandedSelfAttention(nn.Module):
             ctx = _BandedAttnTrainPacked.apply(
                 qkv, drop_mask, h, self.attn_win_size, p_drop,
             )
+            if self.hidden_size == 280 and _flin_train_available():
+                out = _OutProjFused.apply(
+                    ctx.reshape(b * t, -1), self.out_proj.weight
+                ).view(b, t, -1)
+                return out, None
             return self.out_proj(ctx), None
         q = self.q_proj(x).view(b, t, h, d).transpose(1, 2)  # [B,H,T,D]
         k = self.k_proj(x).view(b, t, h, d).transpose(1, 2)
